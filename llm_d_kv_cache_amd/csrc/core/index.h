@@ -1,0 +1,355 @@
+// In-memory KV-block index: requestKey -> pod entries, plus the
+// engineKey -> requestKey bridge used to stitch event-side hash chains.
+//
+// Capability parity with the reference Index interface and InMemoryIndex
+// (pkg/kvcache/kvblock/index.go:120-193, in_memory.go), re-designed for
+// multi-core ingest: the store is sharded by key hash with one mutex, one
+// open hash map and one intrusive LRU list per shard, and pod entries are
+// 12-byte interned records (string_table.h). There is no global lock on the
+// Lookup/Add path; the reference's TOCTOU window between the emptiness check
+// and map removal is closed structurally because a shard's mutex covers the
+// whole check-and-act.
+#pragma once
+
+#include <atomic>
+#include <cstdint>
+#include <list>
+#include <mutex>
+#include <stdexcept>
+#include <unordered_map>
+#include <unordered_set>
+#include <vector>
+
+#include "string_table.h"
+
+namespace kvc {
+
+struct PodEntry {
+  uint32_t pod = 0;
+  uint32_t tier = 0;
+  uint8_t flags = 0;  // bit0: speculative, bit1: has_group
+  int32_t group = 0;
+
+  bool speculative() const { return flags & 1; }
+  bool has_group() const { return flags & 2; }
+  bool operator==(const PodEntry& o) const {
+    return pod == o.pod && tier == o.tier && flags == o.flags && group == o.group;
+  }
+};
+
+enum class KeyType { kEngine = 0, kRequest = 1 };
+
+struct IndexStats {
+  uint64_t admissions = 0;
+  uint64_t evictions = 0;
+  uint64_t lookups = 0;
+  uint64_t hits = 0;
+  uint64_t keys = 0;
+};
+
+struct InMemoryIndexConfig {
+  size_t size = 100000000;   // max request keys (reference default 1e8)
+  size_t pods_per_key = 10;  // max pod entries per key
+  size_t shards = 64;        // power of two
+};
+
+class InMemoryIndex {
+ public:
+  explicit InMemoryIndex(const InMemoryIndexConfig& cfg = {}) : cfg_(cfg) {
+    if (cfg_.shards == 0 || (cfg_.shards & (cfg_.shards - 1)) != 0)
+      throw std::invalid_argument("shards must be a power of two");
+    shards_ = std::vector<Shard>(cfg_.shards);
+    eng_shards_ = std::vector<EngShard>(cfg_.shards);
+    shard_cap_ = std::max<size_t>(1, cfg_.size / cfg_.shards);
+  }
+
+  StringTable& strings() { return strings_; }
+
+  // Lookup keys in order. Filter by pod ids when filter is non-empty.
+  // A key that is present but has no (filtered) pods ends the scan early
+  // (prefix chain is broken); an absent key is simply skipped.
+  std::vector<std::pair<uint64_t, std::vector<PodEntry>>> lookup(
+      const std::vector<uint64_t>& request_keys,
+      const std::unordered_set<uint32_t>& pod_filter) {
+    if (request_keys.empty())
+      throw std::invalid_argument("no request keys provided for lookup");
+    lookups_.fetch_add(1, std::memory_order_relaxed);
+
+    std::vector<std::pair<uint64_t, std::vector<PodEntry>>> out;
+    out.reserve(request_keys.size());
+    bool any_hit = false;
+    for (uint64_t key : request_keys) {
+      Shard& sh = shard(key);
+      std::lock_guard<std::mutex> g(sh.mu);
+      auto it = sh.map.find(key);
+      if (it == sh.map.end()) continue;
+      if (it->second.pods.empty()) break;  // chain break: present but empty
+      sh.touch(it);
+      std::vector<PodEntry> pods;
+      if (pod_filter.empty()) {
+        pods = it->second.pods;
+      } else {
+        for (const auto& e : it->second.pods)
+          if (pod_filter.count(e.pod)) pods.push_back(e);
+      }
+      if (!pods.empty()) {
+        any_hit = true;
+        out.emplace_back(key, std::move(pods));
+      }
+    }
+    if (any_hit) hits_.fetch_add(1, std::memory_order_relaxed);
+    return out;
+  }
+
+  // Add entries under every request key; when engine_keys is non-empty also
+  // record the engine->request bridge, inferring 1:1 / many:1 / 1:many from
+  // the length ratio (both derive from one token count, so they divide).
+  void add(const std::vector<uint64_t>& engine_keys,
+           const std::vector<uint64_t>& request_keys,
+           const std::vector<PodEntry>& entries) {
+    if (request_keys.empty() || entries.empty())
+      throw std::invalid_argument("no keys or entries provided for add");
+
+    if (!engine_keys.empty()) {
+      const size_t ne = engine_keys.size(), nr = request_keys.size();
+      const size_t n = std::max(ne, nr);
+      uint64_t cur_ek = 0;
+      std::vector<uint64_t> rks;
+      bool have = false;
+      for (size_t i = 0; i < n; ++i) {
+        uint64_t ek = engine_keys[i * ne / n];
+        uint64_t rk = request_keys[i * nr / n];
+        if (!have || ek != cur_ek) {
+          if (have) put_engine_mapping(cur_ek, rks);
+          cur_ek = ek;
+          rks.clear();
+          have = true;
+        }
+        rks.push_back(rk);
+      }
+      if (have) put_engine_mapping(cur_ek, rks);
+    }
+
+    for (uint64_t key : request_keys) {
+      Shard& sh = shard(key);
+      std::lock_guard<std::mutex> g(sh.mu);
+      auto it = sh.map.find(key);
+      if (it == sh.map.end()) {
+        it = sh.map.emplace(key, KeyEntry{}).first;
+        sh.lru.push_front(key);
+        it->second.lru_it = sh.lru.begin();
+        if (sh.map.size() > shard_cap_) evict_lru_locked(sh);
+      } else {
+        sh.touch(it);
+      }
+      auto& pods = it->second.pods;
+      for (const auto& e : entries) {
+        // Per-key pod LRU: move-to-front on re-add, bounded capacity.
+        for (size_t i = 0; i < pods.size(); ++i) {
+          if (pods[i] == e) {
+            pods.erase(pods.begin() + i);
+            break;
+          }
+        }
+        pods.insert(pods.begin(), e);
+        if (pods.size() > cfg_.pods_per_key) pods.pop_back();
+      }
+      admissions_.fetch_add(entries.size(), std::memory_order_relaxed);
+    }
+  }
+
+  // Evict the given pod entries from the key. Engine keys resolve through
+  // the bridge (possibly to several request keys); request keys apply
+  // directly. When every resolved request key is gone/empty the bridge
+  // mapping itself is dropped.
+  void evict(uint64_t key, KeyType type, const std::vector<PodEntry>& entries) {
+    if (entries.empty())
+      throw std::invalid_argument("no entries provided for evict");
+    if (type == KeyType::kRequest) {
+      evict_from_request_key(key, entries);
+      return;
+    }
+    std::vector<uint64_t> rks;
+    {
+      EngShard& es = eng_shard(key);
+      std::lock_guard<std::mutex> g(es.mu);
+      auto it = es.map.find(key);
+      if (it == es.map.end()) return;
+      rks = it->second.rks;
+    }
+    for (uint64_t rk : rks) evict_from_request_key(rk, entries);
+    bool all_empty = true;
+    for (uint64_t rk : rks) {
+      Shard& sh = shard(rk);
+      std::lock_guard<std::mutex> g(sh.mu);
+      auto it = sh.map.find(rk);
+      if (it != sh.map.end() && !it->second.pods.empty()) {
+        all_empty = false;
+        break;
+      }
+    }
+    if (all_empty) {
+      EngShard& es = eng_shard(key);
+      std::lock_guard<std::mutex> g(es.mu);
+      auto it = es.map.find(key);
+      if (it != es.map.end()) {
+        es.lru.erase(it->second.lru_it);
+        es.map.erase(it);
+      }
+    }
+  }
+
+  // Last request key of the engine key's chain segment (the one whose chunk
+  // ends where the engine block ends) — what parent-hash resolution needs.
+  // Returns false when the mapping is missing (e.g. already evicted).
+  bool get_request_key(uint64_t engine_key, uint64_t* out) {
+    EngShard& es = eng_shard(engine_key);
+    std::lock_guard<std::mutex> g(es.mu);
+    auto it = es.map.find(engine_key);
+    if (it == es.map.end() || it->second.rks.empty()) return false;
+    es.touch(it);
+    *out = it->second.rks.back();
+    return true;
+  }
+
+  // Drop every entry of the pod, across all tiers. O(N) scan, off the hot
+  // path (backs AllBlocksCleared). The engine bridge is intentionally left
+  // alone: stale mappings resolve to emptied keys that break the prefix
+  // chain correctly and the LRU self-heals.
+  void clear(uint32_t pod_id) {
+    for (auto& sh : shards_) {
+      std::lock_guard<std::mutex> g(sh.mu);
+      for (auto it = sh.map.begin(); it != sh.map.end();) {
+        auto& pods = it->second.pods;
+        size_t before = pods.size();
+        pods.erase(std::remove_if(pods.begin(), pods.end(),
+                                  [pod_id](const PodEntry& e) { return e.pod == pod_id; }),
+                   pods.end());
+        evictions_.fetch_add(before - pods.size(), std::memory_order_relaxed);
+        if (pods.empty()) {
+          sh.lru.erase(it->second.lru_it);
+          it = sh.map.erase(it);
+        } else {
+          ++it;
+        }
+      }
+    }
+  }
+
+  IndexStats stats() const {
+    IndexStats s;
+    s.admissions = admissions_.load(std::memory_order_relaxed);
+    s.evictions = evictions_.load(std::memory_order_relaxed);
+    s.lookups = lookups_.load(std::memory_order_relaxed);
+    s.hits = hits_.load(std::memory_order_relaxed);
+    for (const auto& sh : shards_) {
+      std::lock_guard<std::mutex> g(const_cast<std::mutex&>(sh.mu));
+      s.keys += sh.map.size();
+    }
+    return s;
+  }
+
+ private:
+  struct KeyEntry {
+    std::vector<PodEntry> pods;  // front = most recently added
+    std::list<uint64_t>::iterator lru_it;
+  };
+  struct Shard {
+    std::mutex mu;
+    std::unordered_map<uint64_t, KeyEntry> map;
+    std::list<uint64_t> lru;  // front = most recent
+
+    Shard() = default;
+    Shard(const Shard&) {}
+    void touch(std::unordered_map<uint64_t, KeyEntry>::iterator it) {
+      lru.splice(lru.begin(), lru, it->second.lru_it);
+    }
+  };
+  struct EngEntry {
+    std::vector<uint64_t> rks;
+    std::list<uint64_t>::iterator lru_it;
+  };
+  struct EngShard {
+    std::mutex mu;
+    std::unordered_map<uint64_t, EngEntry> map;
+    std::list<uint64_t> lru;
+
+    EngShard() = default;
+    EngShard(const EngShard&) {}
+    void touch(std::unordered_map<uint64_t, EngEntry>::iterator it) {
+      lru.splice(lru.begin(), lru, it->second.lru_it);
+    }
+  };
+
+  static uint64_t mix(uint64_t k) {
+    // splitmix64 finalizer: decorrelates shard choice from raw block hashes.
+    k += 0x9e3779b97f4a7c15ull;
+    k = (k ^ (k >> 30)) * 0xbf58476d1ce4e5b9ull;
+    k = (k ^ (k >> 27)) * 0x94d049bb133111ebull;
+    return k ^ (k >> 31);
+  }
+  Shard& shard(uint64_t key) { return shards_[mix(key) & (cfg_.shards - 1)]; }
+  EngShard& eng_shard(uint64_t key) {
+    return eng_shards_[mix(key) & (cfg_.shards - 1)];
+  }
+
+  void put_engine_mapping(uint64_t ek, const std::vector<uint64_t>& rks) {
+    EngShard& es = eng_shard(ek);
+    std::lock_guard<std::mutex> g(es.mu);
+    auto it = es.map.find(ek);
+    if (it == es.map.end()) {
+      it = es.map.emplace(ek, EngEntry{}).first;
+      es.lru.push_front(ek);
+      it->second.lru_it = es.lru.begin();
+      if (es.map.size() > shard_cap_) {
+        uint64_t victim = es.lru.back();
+        es.lru.pop_back();
+        es.map.erase(victim);
+      }
+    } else {
+      es.touch(it);
+    }
+    it->second.rks = rks;
+  }
+
+  void evict_from_request_key(uint64_t key, const std::vector<PodEntry>& entries) {
+    Shard& sh = shard(key);
+    std::lock_guard<std::mutex> g(sh.mu);
+    auto it = sh.map.find(key);
+    if (it == sh.map.end()) return;
+    auto& pods = it->second.pods;
+    for (const auto& e : entries) {
+      for (size_t i = 0; i < pods.size(); ++i) {
+        if (pods[i] == e) {
+          pods.erase(pods.begin() + i);
+          evictions_.fetch_add(1, std::memory_order_relaxed);
+          break;
+        }
+      }
+    }
+    if (pods.empty()) {
+      sh.lru.erase(it->second.lru_it);
+      sh.map.erase(it);
+    }
+  }
+
+  void evict_lru_locked(Shard& sh) {
+    if (sh.lru.empty()) return;
+    uint64_t victim = sh.lru.back();
+    sh.lru.pop_back();
+    auto it = sh.map.find(victim);
+    if (it != sh.map.end()) {
+      evictions_.fetch_add(it->second.pods.size(), std::memory_order_relaxed);
+      sh.map.erase(it);
+    }
+  }
+
+  InMemoryIndexConfig cfg_;
+  size_t shard_cap_;
+  std::vector<Shard> shards_;
+  std::vector<EngShard> eng_shards_;
+  StringTable strings_;
+  std::atomic<uint64_t> admissions_{0}, evictions_{0}, lookups_{0}, hits_{0};
+};
+
+}  // namespace kvc

@@ -1,0 +1,183 @@
+"""ZMTP transport: from-scratch ZeroMQ 3.x PUB/SUB over TCP.
+
+Loopback wire tests covering both topologies the reference supports
+(centralized SUB-bind fan-in, pod-discovery SUB-dial fan-out), topic
+filtering, reconnect, and the end-to-end publisher -> subscriber -> pool ->
+index -> score path with no Python in the data path.
+"""
+import threading
+import time
+
+import pytest
+
+from llm_d_kv_cache_amd import ensure_native
+from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+from llm_d_kv_cache_amd.events import (
+    EventPoolConfig,
+    KVEventsPool,
+    SubscriberManager,
+)
+from llm_d_kv_cache_amd.events.publisher import (
+    EventPublisher,
+    block_stored_payload,
+    encode_batch,
+)
+
+k = ensure_native()
+
+MODEL = "m"
+
+
+def wait_for(cond, timeout=5.0, interval=0.01):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        if cond():
+            return True
+        time.sleep(interval)
+    return False
+
+
+def test_pub_bind_sub_dial_roundtrip():
+    pub = k.Publisher("tcp://127.0.0.1:0", bind=True)
+    got = []
+    lock = threading.Lock()
+
+    def on_msg(topic, seq, payload):
+        with lock:
+            got.append((topic, seq, payload))
+
+    sub = k.Subscriber(f"tcp://127.0.0.1:{pub.port}", "kv@", callback=on_msg)
+    try:
+        assert wait_for(lambda: pub.peer_count >= 1)
+        time.sleep(0.1)  # let the subscription land
+        pub.publish("kv@pod@m", 7, b"hello")
+        pub.publish("other@x", 1, b"filtered-out")
+        pub.publish("kv@pod@m", 8, b"world")
+        assert wait_for(lambda: len(got) == 2)
+        with lock:
+            assert got[0] == ("kv@pod@m", 7, b"hello")
+            assert got[1] == ("kv@pod@m", 8, b"world")
+    finally:
+        sub.close()
+        pub.close()
+
+
+def test_sub_bind_pub_dial_roundtrip():
+    # Centralized topology: host binds SUB, engines dial in with PUB.
+    got = []
+    sub = k.Subscriber("tcp://127.0.0.1:0", "", callback=lambda t, s, p: got.append((t, s, p)), bind=True)
+    pub = k.Publisher(f"tcp://127.0.0.1:{sub.port}", bind=False)
+    try:
+        assert wait_for(lambda: pub.peer_count >= 1)
+        time.sleep(0.1)
+        pub.publish("kv@pod@m", 1, b"x")
+        assert wait_for(lambda: len(got) == 1)
+    finally:
+        pub.close()
+        sub.close()
+
+
+def test_subscriber_reconnects():
+    got = []
+    pub = k.Publisher("tcp://127.0.0.1:0", bind=True)
+    port = pub.port
+    sub = k.Subscriber(
+        f"tcp://127.0.0.1:{port}", "", callback=lambda t, s, p: got.append(s),
+        reconnect_ms=200,
+    )
+    try:
+        assert wait_for(lambda: pub.peer_count >= 1)
+        time.sleep(0.05)
+        pub.publish("a", 1, b"x")
+        assert wait_for(lambda: got == [1])
+        pub.close()
+        # new publisher on the same port; subscriber must re-dial
+        pub = k.Publisher(f"tcp://127.0.0.1:{port}", bind=True)
+        assert wait_for(lambda: pub.peer_count >= 1, timeout=10)
+        time.sleep(0.05)
+        pub.publish("b", 2, b"y")
+        assert wait_for(lambda: got == [1, 2])
+    finally:
+        sub.close()
+        pub.close()
+
+
+def test_end_to_end_publish_score():
+    """Dummy engine fleet -> ZMTP -> pool -> index -> Score()."""
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(zmq_endpoint="tcp://127.0.0.1:0"), ix)
+    pool.start()
+    pubs = []
+    try:
+        port = pool.port
+        tokens = list(range(64))
+        for p in range(3):
+            ep = EventPublisher(f"tcp://127.0.0.1:{port}", f"pod-{p}", MODEL, bind=False)
+            pubs.append(ep)
+        time.sleep(0.3)  # handshakes + subscriptions
+        n_blocks = [4, 2, 1]
+        for p, ep in enumerate(pubs):
+            nb = n_blocks[p]
+            ep.publish_events([
+                block_stored_payload(
+                    [100 * p + i for i in range(nb)], None,
+                    tokens[: nb * 16], 16,
+                )
+            ])
+        assert wait_for(lambda: pool.stats().processed == 3)
+        scores = ix.score_tokens(tokens, MODEL)
+        assert scores == {"pod-0": 4.0, "pod-1": 2.0, "pod-2": 1.0}
+    finally:
+        for ep in pubs:
+            ep.close()
+        pool.shutdown()
+
+
+def test_pod_discovery_subscriber_manager():
+    """Fan-out topology: each pod binds PUB, the manager dials each."""
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(discover_pods=True), ix)
+    pool.start()
+    mgr = SubscriberManager(pool, topic_filter="kv@", reconnect_ms=200)
+    pubs = {}
+    try:
+        tokens = list(range(32))
+        for p in range(2):
+            ep = EventPublisher("tcp://127.0.0.1:0", f"pod-{p}", MODEL, bind=True)
+            pubs[f"pod-{p}"] = ep
+            mgr.ensure_subscriber(f"pod-{p}", f"tcp://127.0.0.1:{ep.port}")
+        # idempotent ensure
+        mgr.ensure_subscriber("pod-0", f"tcp://127.0.0.1:{pubs['pod-0'].port}")
+        assert mgr.pods() == ["pod-0", "pod-1"]
+        assert wait_for(lambda: all(p._pub.peer_count >= 1 for p in pubs.values()))
+        time.sleep(0.2)
+        for p, ep in pubs.items():
+            ep.publish_events([
+                block_stored_payload([hash(p) & 0xFFFF], None, tokens[:16], 16)
+            ])
+        assert wait_for(lambda: pool.stats().processed == 2)
+        scores = ix.score_tokens(tokens[:16], MODEL)
+        assert scores == {"pod-0": 1.0, "pod-1": 1.0}
+        mgr.remove_subscriber("pod-1")
+        assert mgr.pods() == ["pod-0"]
+    finally:
+        mgr.shutdown()
+        for ep in pubs.values():
+            ep.close()
+        pool.shutdown()
+
+
+def test_large_payload_long_frames():
+    pub = k.Publisher("tcp://127.0.0.1:0", bind=True)
+    got = []
+    sub = k.Subscriber(f"tcp://127.0.0.1:{pub.port}", "", callback=lambda t, s, p: got.append(p))
+    try:
+        assert wait_for(lambda: pub.peer_count >= 1)
+        time.sleep(0.1)
+        big = bytes(range(256)) * 2048  # 512 KiB: long-frame encoding
+        pub.publish("t", 0, big)
+        assert wait_for(lambda: len(got) == 1)
+        assert got[0] == big
+    finally:
+        sub.close()
+        pub.close()

@@ -1,0 +1,280 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: KV-block offload GB/s on Llama-3-8B KV geometry.
+
+One step = one steady-state offload cycle per GPU: store a fresh 16-block
+set of KV files (GPU -> pinned host -> filesystem via the CDNA4 gather
+kernel + SDMA), load them back (file -> host -> GPU scatter), and unlink
+the previous generation. The headline value is the whole-job aggregate
+offload throughput in GB/s across all ranks (weak scaling: per-GPU work is
+fixed).
+
+Score() and KVEvents-ingest rates — the control-plane north-star metrics —
+are measured in separate bracketed phases and reported as aux fields.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import shutil
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+MODEL = "meta-llama/Llama-3-8B"
+NUM_LAYERS = 32
+KV_HEADS = 8
+HEAD_SIZE = 128
+BLOCK_TOKENS = 16
+BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2  # K+V, bf16 = 128 KiB
+FILES_PER_STEP = 64
+BLOCKS_PER_FILE = 16  # 256-token offload chunks
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def pick_root():
+    for cand in ("/dev/shm", "/tmp"):
+        if os.path.isdir(cand):
+            free = shutil.disk_usage(cand).free
+            if free > 40 * 1024**3:
+                return os.path.join(cand, "kvcache_bench")
+    return os.path.join("/tmp", "kvcache_bench")
+
+
+def bench_control_plane():
+    """Score req/s + p50, ingest events/s (CPU-side, per rank)."""
+    import numpy as np
+
+    from llm_d_kv_cache_amd import ensure_native
+    from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+    from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_stored_payload,
+        encode_batch,
+    )
+
+    k = ensure_native()
+    ix = KVCacheIndexer(IndexerConfig())
+    tokens = np.arange(4096, dtype=np.uint32)
+    keys = ix.compute_block_keys(tokens, MODEL)
+    for p in range(64):
+        ix.index.add([], keys[: 4 * (p % 64 + 1)], [k.PodEntry(f"pod-{p}", "gpu")])
+    lat = []
+    n = 3000
+    t0 = time.perf_counter()
+    for _ in range(n):
+        t1 = time.perf_counter()
+        ix.score_tokens(tokens, MODEL)
+        lat.append(time.perf_counter() - t1)
+    dt = time.perf_counter() - t0
+    lat.sort()
+    score_rps = n / dt
+    score_p50_us = lat[n // 2] * 1e6
+
+    pool = KVEventsPool(EventPoolConfig(concurrency=4), ix)
+    pool._pool.start()
+    payloads = []
+    for p in range(8):
+        topic = f"kv@pod-{p}@{MODEL}"
+        for j in range(1000):
+            payloads.append(
+                (topic,
+                 encode_batch([
+                     block_stored_payload([j * 16 + x for x in range(8)], None,
+                                          list(range(128)), 16)
+                 ]))
+            )
+    t0 = time.perf_counter()
+    for topic, pl in payloads:
+        pool.add_task(topic, 0, pl)
+    pool.drain()
+    dt = time.perf_counter() - t0
+    ingest_eps = len(payloads) / dt
+    pool.shutdown()
+    return {
+        "score_req_s": round(score_rps, 1),
+        "score_p50_us": round(score_p50_us, 1),
+        "ingest_batches_s": round(ingest_eps, 1),
+        "ingest_blocks_s": round(ingest_eps * 8, 1),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--root", type=str, default=None)
+    ap.add_argument("--copy-path", type=str, default="staged",
+                    choices=["staged", "zero_copy"])
+    ap.add_argument("--io-threads", type=int, default=16)
+    ap.add_argument("--device-blocks", type=int, default=2048)
+    args = ap.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        dist = tdist
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+
+    if not torch.cuda.is_available():
+        log("no GPU: control-plane phases only; offload GB/s unmeasured")
+        aux = bench_control_plane()
+        if rank == 0:
+            print(json.dumps({
+                "metric": "kv_block_offload_GBps", "value": None, "unit": "GB/s",
+                "n_gpus": 0, "steps": args.steps, "warmup": args.warmup,
+                "ms_per_step": None, "higher_is_better": True, "scaling": "weak",
+                "vs_baseline": None, "dtype": "bf16", "data": "synthetic",
+                "aux": aux,
+                "config": {"model": MODEL, "note": "cpu-only container"},
+            }))
+        return
+
+    torch.cuda.set_device(local_rank)
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        StorageToGPUHandler,
+        TorchOffloadEngine,
+    )
+
+    root = args.root or pick_root()
+    rank_root = os.path.join(root, f"rank{rank}")
+    shutil.rmtree(rank_root, ignore_errors=True)
+    os.makedirs(rank_root, exist_ok=True)
+
+    # Llama-3-8B canonical KV: one group, 32 layers, (num_blocks, 128 KiB)
+    log(f"allocating KV cache: {args.device_blocks} blocks x {NUM_LAYERS} layers "
+        f"x {BLOCK_BYTES // 1024} KiB = "
+        f"{args.device_blocks * NUM_LAYERS * BLOCK_BYTES / 1e9:.1f} GB")
+    group = [
+        torch.randint(0, 255, (args.device_blocks, BLOCK_BYTES), dtype=torch.uint8,
+                      device="cuda")
+        for _ in range(NUM_LAYERS)
+    ]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=args.io_threads,
+                            gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path=args.copy_path, device=local_rank),
+    )
+    mapper = FileMapper(rank_root, KVCacheLayoutConfig(
+        model=MODEL, tp_size=1, kv_cache_groups=(("full_attention", 16, BLOCK_BYTES),),
+    ))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+
+    blocks_per_step = FILES_PER_STEP * BLOCKS_PER_FILE
+    assert blocks_per_step <= args.device_blocks
+    step_bytes = FILES_PER_STEP * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
+
+    def run_step(step_id):
+        base = step_id * FILES_PER_STEP + 1
+        hashes = list(range(base, base + FILES_PER_STEP))
+        ids = list(range(blocks_per_step))
+        n_jobs = 0
+        # store in 8-file jobs to pipeline the I/O pool
+        for i in range(0, FILES_PER_STEP, 8):
+            store.transfer_async(hashes[i:i + 8], {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
+            n_jobs += 1
+        done = 0
+        while done < n_jobs:
+            done += len(store.get_finished())
+        for i in range(0, FILES_PER_STEP, 8):
+            load.transfer_async(hashes[i:i + 8], {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
+        done = 0
+        while done < n_jobs:
+            done += len(load.get_finished())
+        # steady-state disk management: drop the previous generation
+        prev = (step_id - 1) * FILES_PER_STEP + 1
+        if step_id > 0:
+            for h in range(prev, prev + FILES_PER_STEP):
+                try:
+                    os.unlink(mapper.file_name(h, 0))
+                except OSError:
+                    pass
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for w in range(args.warmup):
+        run_step(-1 - w)  # negative ids: cleaned below
+    log("warmup done")
+
+    barrier()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        run_step(s)
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], device="cuda" if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    moved_bytes = 2 * step_bytes * args.steps  # store + load, per rank
+    total_gbps = moved_bytes * world / elapsed / 1e9
+    stats = eng.stats()
+
+    aux = bench_control_plane() if rank == 0 else None
+
+    shutil.rmtree(rank_root, ignore_errors=True)
+    if rank == 0:
+        print(json.dumps({
+            "metric": "kv_block_offload_GBps",
+            "value": round(total_gbps, 2),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "aux": aux,
+            "config": {
+                "model": MODEL,
+                "global_batch": FILES_PER_STEP * world,
+                "seq_len": FILES_PER_STEP * BLOCKS_PER_FILE * BLOCK_TOKENS,
+                "parallelism": f"dp{world}",
+                "kv_geometry": f"{NUM_LAYERS}L x {KV_HEADS}H x {HEAD_SIZE} bf16, "
+                               f"{BLOCK_TOKENS}-token blocks",
+                "bytes_per_step_per_gpu": step_bytes * 2,
+                "copy_path": args.copy_path,
+                "io_threads": args.io_threads,
+                "root": root,
+                "files_written": stats.files_written,
+                "engine_avg_write_ms": round(stats.avg_write_seconds * 1e3, 3),
+            },
+        }))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,439 @@
+// StorageOffloadEngine: the KV-block offload job orchestrator.
+//
+// Capability parity with the reference StorageOffloadEngine
+// (csrc/storage/storage_offload.cpp): async store/load of GPU KV blocks to
+// content-addressed files, KV-ready event fencing against the serving
+// engine's stream, per-file tasks on the NUMA-pinned I/O pool (loads HIGH
+// priority), skip-if-exists dedupe with atime touch, EMA-driven dynamic
+// write-queue limit with write dropping, job cancellation (preemption), and
+// get_finished() polling.
+//
+// MI355X-first redesign of the copy path: the default "staged" mode runs
+// the CDNA4 gather kernel HBM->HBM into a contiguous device bounce (brief,
+// bandwidth-bound CU usage) and crosses PCIe with hipMemcpyAsync on the
+// SDMA engines — zero CU occupancy while the wire is busy, so serving
+// compute is undisturbed. "zero_copy" mode instead has the kernel write
+// device-mapped pinned host memory directly (the reference's only mode).
+// Constructed against host tensors the same control logic runs on memcpy
+// ("host" mode, CPU-only CI) — an explicit mode, never a silent fallback.
+#pragma once
+
+#include <algorithm>
+#include <list>
+#include <map>
+#include <unordered_map>
+
+#include "common.h"
+#include "file_io.h"
+#include "thread_pool.h"
+
+extern "C" hipError_t kvc_launch_gather(const void* const*, const uint64_t*, int,
+                                        uint64_t, const int32_t*, int, uint8_t*,
+                                        hipStream_t);
+extern "C" hipError_t kvc_launch_scatter(const void* const*, const uint64_t*, int,
+                                         uint64_t, const int32_t*, int,
+                                         const uint8_t*, hipStream_t);
+
+namespace kvo {
+
+constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerFile
+
+enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };
+
+struct GroupDesc {
+  std::vector<void*> layer_ptrs;        // per-layer block-0 base address
+  std::vector<uint64_t> layer_strides;  // bytes between consecutive blocks
+  uint64_t block_bytes = 0;             // bytes per (block, layer)
+};
+
+struct EngineConfig {
+  int io_threads = 16;
+  int gpu_blocks_per_file = 16;
+  double read_preferring_ratio = 0.75;
+  double max_write_queued_seconds = 30.0;
+  bool gpu_mode = false;
+  int device = 0;
+  CopyPath copy_path = CopyPath::kStaged;
+};
+
+struct FileTransfer {
+  int group = 0;
+  std::string path;
+  std::vector<int32_t> block_ids;
+  // Block-slot offset of this span within the file's nominal layout. Stores
+  // emit slot_offset == 0 (head-partial files are simply short); loads may
+  // skip an already-cached head.
+  int slot_offset = 0;
+};
+
+struct FinishedJob {
+  int64_t id = 0;
+  bool success = false;
+  bool dropped = false;
+};
+
+struct EngineStats {
+  uint64_t stores_submitted = 0;
+  uint64_t loads_submitted = 0;
+  uint64_t files_written = 0;
+  uint64_t files_deduped = 0;
+  uint64_t files_read = 0;
+  uint64_t writes_dropped = 0;
+  uint64_t tasks_cancelled = 0;
+  uint64_t errors = 0;
+  double avg_write_seconds = 0;
+  uint64_t bytes_stored = 0;
+  uint64_t bytes_loaded = 0;
+};
+
+class StorageOffloadEngine {
+ public:
+  StorageOffloadEngine(EngineConfig cfg, std::vector<GroupDesc> groups)
+      : cfg_(cfg), groups_(std::move(groups)) {
+    if (cfg_.gpu_blocks_per_file > kMaxBlocksPerFileHost)
+      throw std::invalid_argument("gpu_blocks_per_file exceeds kernel limit (64)");
+    if (cfg_.gpu_mode) {
+      int count = 0;
+      if (hipGetDeviceCount(&count) != hipSuccess || count <= cfg_.device)
+        throw HipError("gpu_mode requested but no HIP device available");
+      KVO_HIP_CHECK(hipSetDevice(cfg_.device));
+    } else {
+      cfg_.copy_path = CopyPath::kHostMemcpy;
+    }
+
+    size_t max_file_bytes = 0;
+    for (auto& g : groups_) {
+      if (g.block_bytes % 16 != 0)
+        throw std::invalid_argument("block_bytes must be a multiple of 16");
+      if (g.layer_ptrs.size() != g.layer_strides.size())
+        throw std::invalid_argument("layer ptr/stride length mismatch");
+      max_file_bytes =
+          std::max(max_file_bytes, static_cast<size_t>(cfg_.gpu_blocks_per_file) *
+                                       g.layer_ptrs.size() * g.block_bytes);
+      if (cfg_.gpu_mode) {
+        void* dp = nullptr;
+        KVO_HIP_CHECK(hipMalloc(&dp, g.layer_ptrs.size() * sizeof(void*)));
+        KVO_HIP_CHECK(hipMemcpy(dp, g.layer_ptrs.data(),
+                                g.layer_ptrs.size() * sizeof(void*),
+                                hipMemcpyHostToDevice));
+        dev_layer_ptrs_.push_back(static_cast<void**>(dp));
+        void* ds = nullptr;
+        KVO_HIP_CHECK(hipMalloc(&ds, g.layer_strides.size() * sizeof(uint64_t)));
+        KVO_HIP_CHECK(hipMemcpy(ds, g.layer_strides.data(),
+                                g.layer_strides.size() * sizeof(uint64_t),
+                                hipMemcpyHostToDevice));
+        dev_layer_strides_.push_back(static_cast<uint64_t*>(ds));
+      }
+    }
+
+    size_t device_staging = cfg_.copy_path == CopyPath::kStaged ? max_file_bytes : 0;
+    pool_ = std::make_unique<IoThreadPool>(
+        cfg_.io_threads, cfg_.gpu_mode, cfg_.device, max_file_bytes, device_staging,
+        cfg_.read_preferring_ratio);
+  }
+
+  ~StorageOffloadEngine() {
+    pool_.reset();  // drain workers before freeing device arrays
+    for (auto p : dev_layer_ptrs_) (void)hipFree(p);
+    for (auto p : dev_layer_strides_) (void)hipFree(p);
+  }
+
+  const EngineConfig& config() const { return cfg_; }
+
+  // ---- store ----------------------------------------------------------------
+
+  int64_t async_store(std::vector<FileTransfer> files, uintptr_t caller_stream) {
+    auto job = new_job(files.size(), /*is_store=*/true);
+
+    // Dynamic write-queue limit (EMA of write duration): when the backlog
+    // exceeds threads * max_queued_seconds / avg_write_seconds, the whole
+    // store is dropped — offload is a cache, dropping is always safe.
+    {
+      std::lock_guard<std::mutex> g(jobs_mu_);
+      double avg = stats_.avg_write_seconds;
+      if (avg > 0) {
+        size_t limit = static_cast<size_t>(
+            pool_->size() * cfg_.max_write_queued_seconds / avg);
+        if (pending_writes_ > limit) {
+          stats_.writes_dropped += files.size();
+          job->remaining = 0;
+          finished_.push_back({job->id, true, /*dropped=*/true});
+          jobs_.erase(job->id);
+          done_cv_.notify_all();
+          return job->id;
+        }
+      }
+      pending_writes_ += files.size();
+    }
+
+    if (cfg_.gpu_mode) {
+      KVO_HIP_CHECK(hipEventCreateWithFlags(&job->kv_ready, hipEventDisableTiming));
+      KVO_HIP_CHECK(hipEventRecord(job->kv_ready,
+                                   reinterpret_cast<hipStream_t>(caller_stream)));
+    }
+    stats_inc([&](EngineStats& s) { s.stores_submitted += files.size(); });
+
+    for (auto& ft : files) {
+      validate(ft);
+      pool_->enqueue(Priority::kNormal, [this, job, ft](WorkerCtx& ctx) {
+        bool ok = true;
+        double t0 = now_s();
+        bool wrote = false;
+        try {
+          if (!job->cancelled.load(std::memory_order_acquire)) {
+            if (file_exists(ft.path)) {
+              touch_atime(ft.path);
+              stats_inc([](EngineStats& s) { s.files_deduped++; });
+            } else {
+              store_one(ctx, *job, ft);
+              wrote = true;
+            }
+          } else {
+            stats_inc([](EngineStats& s) { s.tasks_cancelled++; });
+          }
+        } catch (const std::exception& e) {
+          KVO_LOG_ERROR("store %s failed: %s", ft.path.c_str(), e.what());
+          stats_inc([](EngineStats& s) { s.errors++; });
+          ok = false;
+        }
+        if (wrote) {
+          double dt = now_s() - t0;
+          std::lock_guard<std::mutex> g(jobs_mu_);
+          // EMA alpha=0.05 of write duration drives the queue limit.
+          stats_.avg_write_seconds = stats_.avg_write_seconds == 0
+                                         ? dt
+                                         : 0.95 * stats_.avg_write_seconds + 0.05 * dt;
+        }
+        {
+          std::lock_guard<std::mutex> g(jobs_mu_);
+          pending_writes_--;
+        }
+        task_done(job, ok);
+      });
+    }
+    return job->id;
+  }
+
+  // ---- load -----------------------------------------------------------------
+
+  int64_t async_load(std::vector<FileTransfer> files) {
+    auto job = new_job(files.size(), /*is_store=*/false);
+    stats_inc([&](EngineStats& s) { s.loads_submitted += files.size(); });
+    for (auto& ft : files) {
+      validate(ft);
+      pool_->enqueue(Priority::kHigh, [this, job, ft](WorkerCtx& ctx) {
+        bool ok = true;
+        try {
+          if (!job->cancelled.load(std::memory_order_acquire)) {
+            load_one(ctx, ft);
+          } else {
+            stats_inc([](EngineStats& s) { s.tasks_cancelled++; });
+          }
+        } catch (const std::exception& e) {
+          KVO_LOG_ERROR("load %s failed: %s", ft.path.c_str(), e.what());
+          stats_inc([](EngineStats& s) { s.errors++; });
+          ok = false;
+        }
+        task_done(job, ok);
+      });
+    }
+    return job->id;
+  }
+
+  // ---- completion -----------------------------------------------------------
+
+  std::vector<FinishedJob> get_finished() {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    std::vector<FinishedJob> out;
+    out.swap(finished_);
+    return out;
+  }
+
+  // Cancel a job's queued tasks (preemption) and wait for in-flight ones.
+  // Returns true when every task that DID run succeeded.
+  bool wait_job(int64_t id) {
+    std::shared_ptr<Job> job;
+    {
+      std::lock_guard<std::mutex> g(jobs_mu_);
+      auto it = jobs_.find(id);
+      if (it == jobs_.end()) return true;  // already finished
+      job = it->second;
+    }
+    job->cancelled.store(true, std::memory_order_release);
+    std::unique_lock<std::mutex> g(jobs_mu_);
+    done_cv_.wait(g, [&] { return job->remaining.load() == 0; });
+    return job->success.load();
+  }
+
+  EngineStats stats() {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    return stats_;
+  }
+
+  size_t pending_writes() {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    return pending_writes_;
+  }
+
+ private:
+  struct Job {
+    int64_t id;
+    std::atomic<int> remaining{0};
+    std::atomic<bool> cancelled{false};
+    std::atomic<bool> success{true};
+    hipEvent_t kv_ready = nullptr;
+  };
+
+  void validate(const FileTransfer& ft) {
+    if (ft.group < 0 || ft.group >= static_cast<int>(groups_.size()))
+      throw std::invalid_argument("bad group index");
+    if (ft.block_ids.empty() ||
+        ft.block_ids.size() > static_cast<size_t>(cfg_.gpu_blocks_per_file))
+      throw std::invalid_argument("bad block count for file transfer");
+  }
+
+  std::shared_ptr<Job> new_job(size_t n_tasks, bool) {
+    auto job = std::make_shared<Job>();
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    job->id = next_job_id_++;
+    job->remaining = static_cast<int>(n_tasks);
+    jobs_[job->id] = job;
+    return job;
+  }
+
+  void task_done(const std::shared_ptr<Job>& job, bool ok) {
+    if (!ok) job->success.store(false, std::memory_order_relaxed);
+    if (job->remaining.fetch_sub(1, std::memory_order_acq_rel) == 1) {
+      if (job->kv_ready) {
+        (void)hipEventDestroy(job->kv_ready);
+        job->kv_ready = nullptr;
+      }
+      std::lock_guard<std::mutex> g(jobs_mu_);
+      finished_.push_back({job->id, job->success.load(), false});
+      jobs_.erase(job->id);
+      done_cv_.notify_all();
+    }
+  }
+
+  template <typename F>
+  void stats_inc(F f) {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    f(stats_);
+  }
+
+  size_t file_bytes(const GroupDesc& g, size_t n_blocks) const {
+    return n_blocks * g.layer_ptrs.size() * g.block_bytes;
+  }
+
+  void store_one(WorkerCtx& ctx, Job& job, const FileTransfer& ft) {
+    const GroupDesc& g = groups_[ft.group];
+    const int nb = static_cast<int>(ft.block_ids.size());
+    const int nl = static_cast<int>(g.layer_ptrs.size());
+    const size_t bytes = file_bytes(g, nb);
+
+    if (cfg_.copy_path == CopyPath::kHostMemcpy) {
+      gather_host(g, ft.block_ids, ctx.host_staging->host());
+    } else {
+      // KV-ready fence: the gather must observe the serving engine's
+      // completed KV writes for these blocks.
+      KVO_HIP_CHECK(hipStreamWaitEvent(ctx.stream, job.kv_ready, 0));
+      uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
+                                ? ctx.device_staging->ptr()
+                                : ctx.host_staging->device();
+      hipError_t err = kvc_launch_gather(
+          const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+          dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
+          nb, kernel_dst, ctx.stream);
+      if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+      if (cfg_.copy_path == CopyPath::kStaged) {
+        // PCIe hop on the SDMA engines: zero CU occupancy.
+        KVO_HIP_CHECK(hipMemcpyAsync(ctx.host_staging->host(),
+                                     ctx.device_staging->ptr(), bytes,
+                                     hipMemcpyDeviceToHost, ctx.stream));
+      }
+      KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
+    }
+    write_file_atomic(ft.path, ctx.host_staging->host(), bytes);
+    stats_inc([&](EngineStats& s) {
+      s.files_written++;
+      s.bytes_stored += bytes;
+    });
+  }
+
+  void load_one(WorkerCtx& ctx, const FileTransfer& ft) {
+    const GroupDesc& g = groups_[ft.group];
+    const int nb = static_cast<int>(ft.block_ids.size());
+    const int nl = static_cast<int>(g.layer_ptrs.size());
+    const size_t bytes = file_bytes(g, nb);
+    const uint64_t offset = static_cast<uint64_t>(ft.slot_offset) * nl * g.block_bytes;
+
+    int64_t fsz = file_size(ft.path);
+    if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
+      throw FileIoError("file " + ft.path + " does not cover requested span");
+
+    read_file_range(ft.path, offset, ctx.host_staging->host(), bytes);
+    touch_atime(ft.path);
+
+    if (cfg_.copy_path == CopyPath::kHostMemcpy) {
+      scatter_host(g, ft.block_ids, ctx.host_staging->host());
+    } else {
+      const uint8_t* kernel_src = ctx.host_staging->device();
+      if (cfg_.copy_path == CopyPath::kStaged) {
+        KVO_HIP_CHECK(hipMemcpyAsync(ctx.device_staging->ptr(),
+                                     ctx.host_staging->host(), bytes,
+                                     hipMemcpyHostToDevice, ctx.stream));
+        kernel_src = ctx.device_staging->ptr();
+      }
+      hipError_t err = kvc_launch_scatter(
+          const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+          dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
+          nb, kernel_src, ctx.stream);
+      if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+      KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
+    }
+    stats_inc([&](EngineStats& s) {
+      s.files_read++;
+      s.bytes_loaded += bytes;
+    });
+  }
+
+  void gather_host(const GroupDesc& g, const std::vector<int32_t>& ids,
+                   uint8_t* dst) const {
+    const size_t nl = g.layer_ptrs.size();
+    for (size_t bi = 0; bi < ids.size(); ++bi) {
+      for (size_t l = 0; l < nl; ++l) {
+        const uint8_t* src = static_cast<const uint8_t*>(g.layer_ptrs[l]) +
+                             static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
+        std::memcpy(dst + (bi * nl + l) * g.block_bytes, src, g.block_bytes);
+      }
+    }
+  }
+
+  void scatter_host(const GroupDesc& g, const std::vector<int32_t>& ids,
+                    const uint8_t* src) const {
+    const size_t nl = g.layer_ptrs.size();
+    for (size_t bi = 0; bi < ids.size(); ++bi) {
+      for (size_t l = 0; l < nl; ++l) {
+        uint8_t* dst = static_cast<uint8_t*>(g.layer_ptrs[l]) +
+                       static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
+        std::memcpy(dst, src + (bi * nl + l) * g.block_bytes, g.block_bytes);
+      }
+    }
+  }
+
+  EngineConfig cfg_;
+  std::vector<GroupDesc> groups_;
+  std::vector<void**> dev_layer_ptrs_;
+  std::vector<uint64_t*> dev_layer_strides_;
+  std::unique_ptr<IoThreadPool> pool_;
+
+  std::mutex jobs_mu_;
+  std::condition_variable done_cv_;
+  std::unordered_map<int64_t, std::shared_ptr<Job>> jobs_;
+  std::vector<FinishedJob> finished_;
+  int64_t next_job_id_ = 1;
+  size_t pending_writes_ = 0;
+  EngineStats stats_;
+};
+
+}  // namespace kvo

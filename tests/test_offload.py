@@ -1,0 +1,235 @@
+"""Offload data plane on the host path: the full control logic (transfer
+building, file layout, QoS, dedupe, cancellation) runs CPU-only; the HIP
+copy path is exercised by the gpu-marked twin tests in test_offload_gpu.py.
+"""
+import os
+import time
+
+import pytest
+import torch
+
+from llm_d_kv_cache_amd.offload import (
+    FileMapper,
+    GPUToStorageHandler,
+    KVCacheLayoutConfig,
+    OffloadEngineConfig,
+    SharedStorageOffloadManager,
+    StorageToGPUHandler,
+    TorchOffloadEngine,
+)
+
+NUM_BLOCKS = 64
+NUM_LAYERS = 4
+BLOCK_BYTES = 4096  # 16-token block, small geometry for tests
+BLOCKS_PER_FILE = 8
+
+
+def make_group(seed=0, num_layers=NUM_LAYERS, block_bytes=BLOCK_BYTES):
+    g = torch.Generator().manual_seed(seed)
+    return [
+        torch.randint(0, 255, (NUM_BLOCKS, block_bytes), dtype=torch.uint8, generator=g)
+        for _ in range(num_layers)
+    ]
+
+
+@pytest.fixture
+def setup(tmp_path):
+    group = make_group()
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=4, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="test/model"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    return group, eng, mapper, store, load
+
+
+def wait_finished(handler, n=1, timeout=10.0):
+    out = []
+    deadline = time.time() + timeout
+    while len(out) < n and time.time() < deadline:
+        out.extend(handler.get_finished())
+        time.sleep(0.005)
+    assert len(out) >= n, f"only {len(out)} of {n} jobs finished"
+    return out
+
+
+def test_store_load_roundtrip(setup):
+    group, eng, mapper, store, load = setup
+    hashes = [0xABC123, 0xDEF456]
+    ids = list(range(16))
+    job = store.transfer_async(hashes, {0: ids})
+    res = wait_finished(store)[0]
+    assert res.success and not res.dropped
+    for h in hashes:
+        p = mapper.file_name(h, 0)
+        assert os.path.exists(p)
+        assert os.path.getsize(p) == BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
+
+    orig = [t.clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async(hashes, {0: ids})
+    res = wait_finished(load)[0]
+    assert res.success
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:16], o[:16])
+        assert (t[16:] == 0).all()
+
+
+def test_partial_tail_chunk(setup):
+    group, eng, mapper, store, load = setup
+    # 12 blocks = 1 full file + 4-block head-partial file
+    hashes = [1111, 2222]
+    job = store.transfer_async(hashes, {0: list(range(12))})
+    assert wait_finished(store)[0].success
+    assert os.path.getsize(mapper.file_name(1111, 0)) == 8 * NUM_LAYERS * BLOCK_BYTES
+    assert os.path.getsize(mapper.file_name(2222, 0)) == 4 * NUM_LAYERS * BLOCK_BYTES
+
+    orig = [t.clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async(hashes, {0: list(range(12))})
+    assert wait_finished(load)[0].success
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:12], o[:12])
+
+
+def test_load_with_skip_and_slot_offset(setup):
+    group, eng, mapper, store, load = setup
+    hashes = [7, 8, 9]
+    store.transfer_async(hashes, {0: list(range(24))})
+    assert wait_finished(store)[0].success
+    orig = [t.clone() for t in group]
+    for t in group:
+        t.zero_()
+    # skip the first 10 engine blocks (1 full file + slot 2 of file 1)
+    load.transfer_async(hashes, {0: list(range(10, 24))}, skip_leading_blocks=10)
+    assert wait_finished(load)[0].success
+    for t, o in zip(group, orig):
+        assert torch.equal(t[10:24], o[10:24])
+        assert (t[:10] == 0).all()
+
+
+def test_dedupe_skip_existing(setup):
+    group, eng, mapper, store, load = setup
+    store.transfer_async([42], {0: list(range(8))})
+    assert wait_finished(store)[0].success
+    assert eng.stats().files_written == 1
+    store.transfer_async([42], {0: list(range(8))})
+    assert wait_finished(store)[0].success
+    s = eng.stats()
+    assert s.files_written == 1
+    assert s.files_deduped == 1
+
+
+def test_missing_file_load_fails(setup):
+    group, eng, mapper, store, load = setup
+    load.transfer_async([999999], {0: list(range(8))})
+    res = wait_finished(load)[0]
+    assert not res.success
+
+
+def test_multi_group_hma(tmp_path):
+    # two groups with different geometry (full attention + sliding window)
+    g0 = make_group(seed=1, num_layers=4, block_bytes=4096)
+    g1 = make_group(seed=2, num_layers=2, block_bytes=2048)
+    eng = TorchOffloadEngine(
+        [g0, g1],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=8, copy_path="host"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="hma"))
+    store = GPUToStorageHandler(eng, mapper, [8, 4])
+    load = StorageToGPUHandler(eng, mapper, [8, 4])
+    h = [555]
+    store.transfer_async(h, {0: list(range(8)), 1: list(range(4))})
+    assert wait_finished(store)[0].success
+    p0, p1 = mapper.file_name(555, 0), mapper.file_name(555, 1)
+    assert os.path.exists(p0) and os.path.exists(p1)
+    assert "_g0" in p0 and "_g1" in p1
+    orig0 = [t.clone() for t in g0]
+    orig1 = [t.clone() for t in g1]
+    for t in g0 + g1:
+        t.zero_()
+    load.transfer_async(h, {0: list(range(8)), 1: list(range(4))})
+    assert wait_finished(load)[0].success
+    for t, o in zip(g0, orig0):
+        assert torch.equal(t[:8], o[:8])
+    for t, o in zip(g1, orig1):
+        assert torch.equal(t[:4], o[:4])
+
+
+def test_wait_job_cancels_queued(setup):
+    group, eng, mapper, store, load = setup
+    # flood the 4-thread pool, then cancel the last job: its queued tasks bail
+    jobs = [
+        store.transfer_async([10_000 + i], {0: list(range(8))}) for i in range(50)
+    ]
+    ok = store.wait_job(jobs[-1])
+    assert isinstance(ok, bool)
+    # wait for everything else
+    deadline = time.time() + 10
+    while eng.native.pending_writes > 0 and time.time() < deadline:
+        time.sleep(0.01)
+    s = eng.stats()
+    assert s.files_written + s.tasks_cancelled + s.files_deduped >= 50
+
+
+def test_manager_lookup_prefix(setup):
+    group, eng, mapper, store, load = setup
+    mgr = SharedStorageOffloadManager(mapper, num_groups=1)
+    store.transfer_async([1, 2], {0: list(range(16))})
+    assert wait_finished(store)[0].success
+    assert mgr.lookup([1, 2, 3]) == 2
+    assert mgr.lookup([3, 1, 2]) == 0  # gap at the front stops the scan
+    assert mgr.prepare_store([5, 6]) == [5, 6]
+
+
+def test_file_mapper_layout(tmp_path):
+    cfg = KVCacheLayoutConfig(model="meta-llama/Llama-3-8B", tp_size=2, tp_rank=1)
+    m = FileMapper(str(tmp_path), cfg)
+    p = m.file_name(0xDEADBEEF12345678, group=2)
+    assert p.startswith(str(tmp_path))
+    assert "meta-llama_Llama-3-8B" in p
+    assert "_r1/" in p
+    assert "/dea/db_g2/deadbeef12345678.bin" in p
+    # config changes -> different run dir
+    m2 = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="meta-llama/Llama-3-8B",
+                                                       tp_size=4, tp_rank=1))
+    assert m.run_dir != m2.run_dir
+    # rank does not change the config hash, only the _r suffix
+    m3 = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="meta-llama/Llama-3-8B",
+                                                       tp_size=2, tp_rank=0))
+    assert m.run_dir.rsplit("_r", 1)[0] == m3.run_dir.rsplit("_r", 1)[0]
+    cfgp = m.write_run_config()
+    assert os.path.exists(cfgp)
+
+
+def test_parallel_agnostic_collapses_ranks(tmp_path):
+    a = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="m", tp_size=2, tp_rank=0,
+                                                      parallel_agnostic=True))
+    b = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="m", tp_size=4, tp_rank=3,
+                                                      parallel_agnostic=True))
+    assert a.run_dir == b.run_dir
+
+
+def test_atime_touch_on_dedupe(setup):
+    group, eng, mapper, store, load = setup
+    store.transfer_async([77], {0: list(range(8))})
+    assert wait_finished(store)[0].success
+    p = mapper.file_name(77, 0)
+    old = os.stat(p)
+    os.utime(p, (time.time() - 3600, old.st_mtime))  # age the atime
+    store.transfer_async([77], {0: list(range(8))})
+    assert wait_finished(store)[0].success
+    st = os.stat(p)
+    assert st.st_atime > time.time() - 60  # refreshed
+    assert abs(st.st_mtime - old.st_mtime) < 1e-3  # preserved
+
+
+def test_cpu_tensors_require_host_mode():
+    group = make_group()
+    with pytest.raises(ValueError, match="copy_path='host'"):
+        TorchOffloadEngine([group], OffloadEngineConfig(copy_path="staged"))

@@ -128,12 +128,19 @@ inline std::vector<int> numa_node_cpus(int node) {
 // mode. 4 KiB alignment keeps O_DIRECT file I/O legal on either path.
 class HostStaging {
  public:
-  HostStaging(size_t bytes, bool gpu_mode) : bytes_(bytes), gpu_mode_(gpu_mode) {
+  // mapped=true exposes a device pointer for zero-copy kernel writes;
+  // mapped=false keeps the buffer SDMA-eligible for hipMemcpyAsync (a
+  // mapped buffer makes the runtime treat D2H as device-to-device and run
+  // a blit kernel at ~half SDMA speed while burning CUs — measured on
+  // MI355X, see profiles/).
+  HostStaging(size_t bytes, bool gpu_mode, bool mapped = false)
+      : bytes_(bytes), gpu_mode_(gpu_mode) {
     if (bytes_ == 0) return;
     if (gpu_mode_) {
-      KVO_HIP_CHECK(hipHostMalloc(&ptr_, bytes_,
-                                  hipHostMallocMapped | hipHostMallocPortable));
-      KVO_HIP_CHECK(hipHostGetDevicePointer(&dev_ptr_, ptr_, 0));
+      unsigned flags = hipHostMallocPortable;
+      if (mapped) flags |= hipHostMallocMapped;
+      KVO_HIP_CHECK(hipHostMalloc(&ptr_, bytes_, flags));
+      if (mapped) KVO_HIP_CHECK(hipHostGetDevicePointer(&dev_ptr_, ptr_, 0));
     } else {
       if (posix_memalign(&ptr_, 4096, bytes_) != 0)
         throw std::bad_alloc();

@@ -19,6 +19,8 @@
 #pragma once
 
 #include <algorithm>
+#include <chrono>
+#include <thread>
 #include <list>
 #include <map>
 #include <unordered_map>
@@ -93,9 +95,20 @@ class StorageOffloadEngine {
     if (cfg_.gpu_blocks_per_file > kMaxBlocksPerFileHost)
       throw std::invalid_argument("gpu_blocks_per_file exceeds kernel limit (64)");
     if (cfg_.gpu_mode) {
+      // Transient init failures have been observed right after another
+      // process released the device: retry briefly before giving up.
       int count = 0;
-      if (hipGetDeviceCount(&count) != hipSuccess || count <= cfg_.device)
-        throw HipError("gpu_mode requested but no HIP device available");
+      hipError_t err = hipSuccess;
+      for (int attempt = 0; attempt < 10; ++attempt) {
+        err = hipGetDeviceCount(&count);
+        if (err == hipSuccess && count > cfg_.device) break;
+        std::this_thread::sleep_for(std::chrono::milliseconds(200));
+      }
+      if (err != hipSuccess || count <= cfg_.device)
+        throw HipError(std::string("gpu_mode requested but device ") +
+                       std::to_string(cfg_.device) + " unavailable: " +
+                       hipGetErrorString(err) + " (count=" +
+                       std::to_string(count) + ")");
       KVO_HIP_CHECK(hipSetDevice(cfg_.device));
     } else {
       cfg_.copy_path = CopyPath::kHostMemcpy;
@@ -127,9 +140,10 @@ class StorageOffloadEngine {
     }
 
     size_t device_staging = cfg_.copy_path == CopyPath::kStaged ? max_file_bytes : 0;
+    bool mapped = cfg_.copy_path == CopyPath::kZeroCopy;
     pool_ = std::make_unique<IoThreadPool>(
         cfg_.io_threads, cfg_.gpu_mode, cfg_.device, max_file_bytes, device_staging,
-        cfg_.read_preferring_ratio);
+        cfg_.read_preferring_ratio, mapped);
   }
 
   ~StorageOffloadEngine() {

@@ -39,10 +39,12 @@ class IoThreadPool {
   using Task = std::function<void(WorkerCtx&)>;
 
   IoThreadPool(int n_threads, bool gpu_mode, int device, size_t host_staging_bytes,
-               size_t device_staging_bytes, double read_preferring_ratio)
+               size_t device_staging_bytes, double read_preferring_ratio,
+               bool mapped_host_staging = false)
       : gpu_mode_(gpu_mode), device_(device),
         host_staging_bytes_(host_staging_bytes),
-        device_staging_bytes_(device_staging_bytes) {
+        device_staging_bytes_(device_staging_bytes),
+        mapped_host_staging_(mapped_host_staging) {
     n_threads = std::max(1, n_threads);
     int numa_node = -1;
     if (gpu_mode_) {
@@ -125,7 +127,8 @@ class IoThreadPool {
         pin_to_numa(i);
         KVO_HIP_CHECK(hipStreamCreateWithFlags(&ctx.stream, hipStreamNonBlocking));
       }
-      host_staging = std::make_unique<HostStaging>(host_staging_bytes_, gpu_mode_);
+      host_staging = std::make_unique<HostStaging>(host_staging_bytes_, gpu_mode_,
+                                                   mapped_host_staging_);
       device_staging =
           std::make_unique<DeviceStaging>(device_staging_bytes_, gpu_mode_);
       ctx.host_staging = host_staging.get();
@@ -168,6 +171,7 @@ class IoThreadPool {
   int numa_node_ = -1;
   size_t host_staging_bytes_;
   size_t device_staging_bytes_;
+  bool mapped_host_staging_ = false;
   std::vector<int> cpus_;
   std::mutex mu_;
   std::condition_variable cv_;

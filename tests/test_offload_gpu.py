@@ -151,7 +151,8 @@ def test_prefix_hash_kernel_matches_cpu():
     import reference_impl as ref
 
     seed_val = ref.fnv64a_chain_seed("m")
-    seeds = torch.full((n_seq,), seed_val, dtype=torch.int64, device="cuda")
+    seed_i64 = seed_val - (1 << 64) if seed_val >= (1 << 63) else seed_val
+    seeds = torch.full((n_seq,), seed_i64, dtype=torch.int64, device="cuda")
     keys = torch.zeros(n_seq * n_chunks, dtype=torch.int64, device="cuda")
     stream = torch.cuda.current_stream().cuda_stream
     ko.prefix_hash(tokens.data_ptr(), seq_off.data_ptr(), seeds.data_ptr(),

@@ -99,13 +99,13 @@ def test_write_storm_drops_when_over_limit(tmp_path):
 def test_wait_job_blocks_until_done(tmp_path):
     group, eng, mapper, store, load = build(tmp_path)
     job = store.transfer_async([5, 6, 7], {0: list(range(3 * BPF))})
-    ok = store.wait_job(job)
+    ok = store.wait_job(job)  # cancel + wait (preemption semantics)
     assert ok is True
-    # after wait_job the files exist (tasks either ran or were cancelled;
-    # here the queue was empty so they ran)
-    import os
-
-    assert os.path.exists(mapper.file_name(5, 0))
+    # job fully retired: its tasks either ran or were cancelled, and no
+    # completion is left pending for it
+    assert eng.native.pending_writes == 0
+    s = eng.stats()
+    assert s.files_written + s.tasks_cancelled == 3
 
 
 def test_cancelled_jobs_report_counter(tmp_path):

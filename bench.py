@@ -108,6 +108,55 @@ def bench_control_plane():
     }
 
 
+def bench_peer_phase(dist, torch, group, rank, world, local_rank):
+    """Measure cross-GPU block-pull bandwidth over RCCL/xGMI (or gloo on
+    CPU). Each rank pulls 16-block chunks from its ring neighbor."""
+    from llm_d_kv_cache_amd.peer import PeerMigrationService
+
+    gpu = torch.cuda.is_available()
+    ctrl_pg = dist.new_group(backend="gloo")
+    data_pg = dist.new_group(backend="nccl" if gpu else "gloo")
+    svc = PeerMigrationService([group], data_group=data_pg,
+                               control_group=ctrl_pg,
+                               device=local_rank if gpu else 0)
+    n_chunks = 32
+    bpf = BLOCKS_PER_FILE
+    for c in range(n_chunks):
+        svc.register_blocks(0xE000 + rank * 1000 + c, 0,
+                            list(range(c * bpf, (c + 1) * bpf)))
+    dist.barrier()
+    src = (rank + 1) % world
+    # warmup
+    svc.pull(0xE000 + src * 1000, 0, list(range(1024, 1024 + bpf)),
+             src_rank=src).result(timeout=60)
+    dist.barrier()
+    if gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    futs = [
+        svc.pull(0xE000 + src * 1000 + c, 0,
+                 list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
+                 src_rank=src, timeout=120)
+        for c in range(n_chunks)
+    ]
+    ok = all(f.result(timeout=120) for f in futs)
+    if gpu:
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    dist.barrier()
+    pulled = svc.stats().bytes_received
+    t = torch.tensor([dt], device="cuda" if gpu else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    svc.close()
+    return {
+        "ok": bool(ok),
+        "pull_GBps_per_gpu": round(pulled / dt / 1e9, 2),
+        "pull_GBps_aggregate": round(pulled * world / float(t.item()) / 1e9, 2),
+        "chunk_bytes": pulled // n_chunks,
+        "n_pulls": n_chunks,
+    }
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -239,7 +288,15 @@ def main():
     total_gbps = moved_bytes * world / elapsed / 1e9
     stats = eng.stats()
 
+    # xGMI peer-migration phase (aux, outside the headline timed region):
+    # each rank pulls its neighbor's cached chunks over RCCL send/recv.
+    peer_aux = None
+    if world > 1 and dist is not None:
+        peer_aux = bench_peer_phase(dist, torch, group, rank, world, local_rank)
+
     aux = bench_control_plane() if rank == 0 else None
+    if aux is not None and peer_aux is not None:
+        aux["peer_xgmi"] = peer_aux
 
     shutil.rmtree(rank_root, ignore_errors=True)
     if rank == 0:

@@ -7,6 +7,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "../offload/block_copier.h"
 #include "../offload/engine.h"
 
 namespace py = pybind11;
@@ -124,6 +125,44 @@ PYBIND11_MODULE(_kvoffload, m) {
       .def("stats", &StorageOffloadEngine::stats,
            py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("pending_writes", &StorageOffloadEngine::pending_writes);
+
+  py::class_<BlockCopier>(m, "BlockCopier")
+      .def(py::init([](std::vector<std::tuple<std::vector<uintptr_t>,
+                                              std::vector<uint64_t>, uint64_t>>
+                           groups,
+                       bool gpu_mode, int device) {
+             std::vector<GroupDesc> gs;
+             for (auto& [ptrs, strides, block_bytes] : groups) {
+               GroupDesc g;
+               for (auto p : ptrs) g.layer_ptrs.push_back(reinterpret_cast<void*>(p));
+               g.layer_strides = strides;
+               g.block_bytes = block_bytes;
+               gs.push_back(std::move(g));
+             }
+             py::gil_scoped_release rel;
+             return std::make_unique<BlockCopier>(std::move(gs), gpu_mode, device);
+           }),
+           py::arg("groups"), py::arg("gpu_mode") = false, py::arg("device") = 0)
+      .def("packed_bytes", &BlockCopier::packed_bytes, py::arg("group"),
+           py::arg("n_blocks"))
+      .def(
+          "gather",
+          [](BlockCopier& c, int group, std::vector<int32_t> ids, uintptr_t dst,
+             uintptr_t stream) {
+            py::gil_scoped_release rel;
+            c.gather(group, ids, reinterpret_cast<void*>(dst), stream);
+          },
+          py::arg("group"), py::arg("block_ids"), py::arg("dst"),
+          py::arg("stream") = 0)
+      .def(
+          "scatter",
+          [](BlockCopier& c, int group, std::vector<int32_t> ids, uintptr_t src,
+             uintptr_t stream) {
+            py::gil_scoped_release rel;
+            c.scatter(group, ids, reinterpret_cast<const void*>(src), stream);
+          },
+          py::arg("group"), py::arg("block_ids"), py::arg("src"),
+          py::arg("stream") = 0);
 
   m.def(
       "prefix_hash",

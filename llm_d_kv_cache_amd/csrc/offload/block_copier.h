@@ -1,0 +1,109 @@
+// BlockCopier: standalone gather/scatter between paged KV tensors and a
+// contiguous buffer — the packing primitive of the xGMI peer-migration
+// path (peer rank gathers blocks into a send slab, RCCL ships it, the
+// puller scatters into its own pages). Same kernels as the offload engine;
+// host mode memcpys for CPU-only tests.
+#pragma once
+
+#include "common.h"
+#include "engine.h"  // GroupDesc, kernel launcher decls
+
+namespace kvo {
+
+class BlockCopier {
+ public:
+  BlockCopier(std::vector<GroupDesc> groups, bool gpu_mode, int device)
+      : groups_(std::move(groups)), gpu_mode_(gpu_mode) {
+    if (gpu_mode_) {
+      KVO_HIP_CHECK(hipSetDevice(device));
+      for (auto& g : groups_) {
+        void* dp = nullptr;
+        KVO_HIP_CHECK(hipMalloc(&dp, g.layer_ptrs.size() * sizeof(void*)));
+        KVO_HIP_CHECK(hipMemcpy(dp, g.layer_ptrs.data(),
+                                g.layer_ptrs.size() * sizeof(void*),
+                                hipMemcpyHostToDevice));
+        dev_layer_ptrs_.push_back(static_cast<void**>(dp));
+        void* ds = nullptr;
+        KVO_HIP_CHECK(hipMalloc(&ds, g.layer_strides.size() * sizeof(uint64_t)));
+        KVO_HIP_CHECK(hipMemcpy(ds, g.layer_strides.data(),
+                                g.layer_strides.size() * sizeof(uint64_t),
+                                hipMemcpyHostToDevice));
+        dev_layer_strides_.push_back(static_cast<uint64_t*>(ds));
+      }
+    }
+  }
+
+  ~BlockCopier() {
+    for (auto p : dev_layer_ptrs_) (void)hipFree(p);
+    for (auto p : dev_layer_strides_) (void)hipFree(p);
+  }
+
+  size_t packed_bytes(int group, size_t n_blocks) const {
+    const GroupDesc& g = groups_.at(group);
+    return n_blocks * g.layer_ptrs.size() * g.block_bytes;
+  }
+
+  // Gather block_ids of `group` into contiguous dst (device ptr in GPU
+  // mode). Async on `stream`; caller synchronizes.
+  void gather(int group, const std::vector<int32_t>& ids, void* dst,
+              uintptr_t stream) {
+    const GroupDesc& g = groups_.at(group);
+    check_ids(ids);
+    if (!gpu_mode_) {
+      host_copy(g, ids, static_cast<uint8_t*>(dst), /*to_packed=*/true);
+      return;
+    }
+    hipError_t err = kvc_launch_gather(
+        const_cast<const void* const*>(dev_layer_ptrs_[group]),
+        dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        static_cast<uint8_t*>(dst), reinterpret_cast<hipStream_t>(stream));
+    if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+  }
+
+  void scatter(int group, const std::vector<int32_t>& ids, const void* src,
+               uintptr_t stream) {
+    const GroupDesc& g = groups_.at(group);
+    check_ids(ids);
+    if (!gpu_mode_) {
+      host_copy(g, ids, const_cast<uint8_t*>(static_cast<const uint8_t*>(src)),
+                /*to_packed=*/false);
+      return;
+    }
+    hipError_t err = kvc_launch_scatter(
+        const_cast<const void* const*>(dev_layer_ptrs_[group]),
+        dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        static_cast<const uint8_t*>(src), reinterpret_cast<hipStream_t>(stream));
+    if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+  }
+
+ private:
+  static void check_ids(const std::vector<int32_t>& ids) {
+    if (ids.empty() || ids.size() > kMaxBlocksPerFileHost)
+      throw std::invalid_argument("block count must be in [1, 64]");
+  }
+
+  void host_copy(const GroupDesc& g, const std::vector<int32_t>& ids,
+                 uint8_t* packed, bool to_packed) const {
+    const size_t nl = g.layer_ptrs.size();
+    for (size_t bi = 0; bi < ids.size(); ++bi) {
+      for (size_t l = 0; l < nl; ++l) {
+        uint8_t* page = static_cast<uint8_t*>(g.layer_ptrs[l]) +
+                        static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
+        uint8_t* slab = packed + (bi * nl + l) * g.block_bytes;
+        if (to_packed)
+          std::memcpy(slab, page, g.block_bytes);
+        else
+          std::memcpy(page, slab, g.block_bytes);
+      }
+    }
+  }
+
+  std::vector<GroupDesc> groups_;
+  bool gpu_mode_;
+  std::vector<void**> dev_layer_ptrs_;
+  std::vector<uint64_t*> dev_layer_strides_;
+};
+
+}  // namespace kvo

@@ -1,0 +1,163 @@
+"""Peer KV-block migration: multi-process (gloo, world_size=2) CPU tests.
+
+The same event-loop/protocol code drives RCCL over xGMI on a GPU node;
+here the data plane is gloo and packing is memcpy. GPU twin lives in
+test_peer_gpu.py.
+"""
+import multiprocessing as mp
+import os
+
+import pytest
+import torch
+
+
+def _run_rank(rank, world, init_file, fn_name, q):
+    try:
+        import torch.distributed as dist
+
+        dist.init_process_group(
+            "gloo", init_method=f"file://{init_file}", rank=rank,
+            world_size=world,
+        )
+        from llm_d_kv_cache_amd.peer import PeerMigrationService
+
+        # dedicated groups: the service thread must never share a group
+        # with main-thread collectives
+        ctrl_pg = dist.new_group(backend="gloo")
+        data_pg = dist.new_group(backend="gloo")
+        torch.manual_seed(100 + rank)
+        group = [torch.randint(0, 255, (32, 4096), dtype=torch.uint8)
+                 for _ in range(2)]
+        group_b = [torch.randint(0, 255, (32, 2048), dtype=torch.uint8)]
+        svc = PeerMigrationService([group, group_b], data_group=data_pg,
+                                   control_group=ctrl_pg)
+        fn = globals()[fn_name]
+        fn(rank, svc, group)
+        dist.barrier()
+        svc.close()
+        dist.barrier()
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+def spawn2(fn_name, tmp_path):
+    init_file = str(tmp_path / "pg_init")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_rank, args=(r, 2, init_file, fn_name, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, status = q.get(timeout=120)
+        results[rank] = status
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    assert all(v == "ok" for v in results.values()), results
+
+
+# ---- scenarios (run inside worker processes) --------------------------------
+
+def scenario_basic_pull(rank, svc, group):
+    import torch.distributed as dist
+
+    CHUNK = 0xC0FFEE
+    if rank == 0:
+        svc.register_blocks(CHUNK, 0, [3, 5, 7, 9])
+        golden = torch.cat([group[0][[3, 5, 7, 9]].reshape(4, 1, -1),
+                            group[1][[3, 5, 7, 9]].reshape(4, 1, -1)], dim=1)
+        # publish golden bytes for rank 1 to verify
+        dist.broadcast(golden.contiguous(), src=0)
+        dist.barrier()
+    else:
+        golden = torch.zeros(4, 2, 4096, dtype=torch.uint8)
+        dist.broadcast(golden, src=0)
+        ok = svc.pull(CHUNK, 0, [10, 11, 12, 13], src_rank=0).result(timeout=60)
+        assert ok is True
+        got = torch.stack([
+            torch.stack([group[0][10 + i], group[1][10 + i]]) for i in range(4)
+        ])
+        assert torch.equal(got, golden)
+        dist.barrier()
+
+
+def scenario_missing_chunk(rank, svc, group):
+    import torch.distributed as dist
+
+    if rank == 1:
+        ok = svc.pull(0xDEAD, 0, [1], src_rank=0).result(timeout=60)
+        assert ok is False
+        assert svc.stats().pulls_failed == 1
+    dist.barrier()
+
+
+def scenario_bidirectional(rank, svc, group):
+    """Both ranks pull from each other at the same time: the non-blocking
+    loop must not deadlock."""
+    import torch.distributed as dist
+
+    CHUNK = 0xAB00 + rank
+    svc.register_blocks(CHUNK, 0, [0, 1])
+    dist.barrier()
+    other = 1 - rank
+    futs = [svc.pull(0xAB00 + other, 0, [20 + 2 * i, 21 + 2 * i], src_rank=other)
+            for i in range(3)]
+    assert all(f.result(timeout=60) for f in futs)
+    assert svc.stats().pulls_served >= 1
+    dist.barrier()
+
+
+def scenario_multi_group(rank, svc, group):
+    import torch.distributed as dist
+
+    if rank == 0:
+        svc.register_blocks(0x11, 1, [2, 4])
+    dist.barrier()
+    if rank == 1:
+        ok = svc.pull(0x11, 1, [6, 8], src_rank=0).result(timeout=60)
+        assert ok is True
+    dist.barrier()
+
+
+# ---- tests ------------------------------------------------------------------
+
+@pytest.mark.parametrize("scenario", [
+    "scenario_basic_pull",
+    "scenario_missing_chunk",
+    "scenario_bidirectional",
+    "scenario_multi_group",
+])
+def test_peer_migration(scenario, tmp_path):
+    spawn2(scenario, tmp_path)
+
+
+def test_block_copier_roundtrip():
+    from llm_d_kv_cache_amd import _kvoffload as ko
+
+    g = [torch.randint(0, 255, (16, 1024), dtype=torch.uint8) for _ in range(3)]
+    copier = ko.BlockCopier(
+        [([t.data_ptr() for t in g], [t.stride(0) for t in g], 1024)],
+        gpu_mode=False,
+    )
+    packed = torch.zeros(copier.packed_bytes(0, 4), dtype=torch.uint8)
+    copier.gather(0, [1, 3, 5, 7], packed.data_ptr(), 0)
+    view = packed.view(4, 3, 1024)
+    for bi, b in enumerate([1, 3, 5, 7]):
+        for l in range(3):
+            assert torch.equal(view[bi, l], g[l][b])
+    g2 = [torch.zeros_like(t) for t in g]
+    copier2 = ko.BlockCopier(
+        [([t.data_ptr() for t in g2], [t.stride(0) for t in g2], 1024)],
+        gpu_mode=False,
+    )
+    copier2.scatter(0, [0, 2, 4, 6], packed.data_ptr(), 0)
+    for bi, b in enumerate([0, 2, 4, 6]):
+        for l in range(3):
+            assert torch.equal(g2[l][b], view[bi, l])

@@ -34,3 +34,29 @@ def ensure_native(build_if_missing: bool = True):
         from . import _kvcore  # type: ignore[attr-defined]
 
         return _kvcore
+
+
+def ensure_offload_native(build_if_missing: bool = True):
+    """Import (building if necessary) the native _kvoffload module.
+
+    torch MUST be imported first: _kvoffload links libamdhip64, and if it
+    loads the system copy before torch loads its bundled one, two HIP
+    runtimes coexist and device enumeration fails ("no ROCm-capable device
+    is detected") — observed on MI355X. Importing torch first makes the
+    loader reuse torch's runtime for our extension.
+    """
+    import torch  # noqa: F401  (load order matters; see docstring)
+
+    try:
+        from . import _kvoffload  # type: ignore[attr-defined]
+
+        return _kvoffload
+    except ImportError:
+        if not build_if_missing:
+            raise
+        from ._build import build_kvoffload
+
+        build_kvoffload()
+        from . import _kvoffload  # type: ignore[attr-defined]
+
+        return _kvoffload

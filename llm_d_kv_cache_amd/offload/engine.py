@@ -35,14 +35,9 @@ class TorchOffloadEngine:
     def __init__(self, groups: Sequence[Sequence], config: OffloadEngineConfig):
         import torch
 
-        from .. import _build
+        from .. import ensure_offload_native
 
-        try:
-            from .. import _kvoffload  # type: ignore[attr-defined]
-        except ImportError:
-            _build.build_kvoffload()
-            from .. import _kvoffload  # type: ignore[attr-defined]
-        self._ko = _kvoffload
+        self._ko = ensure_offload_native()
         self.config = config
 
         if not groups or not groups[0]:

@@ -83,13 +83,9 @@ class PeerMigrationService:
         import torch
         import torch.distributed as dist
 
-        from .. import _build
+        from .. import ensure_offload_native
 
-        try:
-            from .. import _kvoffload  # type: ignore[attr-defined]
-        except ImportError:
-            _build.build_kvoffload()
-            from .. import _kvoffload  # type: ignore[attr-defined]
+        _kvoffload = ensure_offload_native()
 
         self._torch = torch
         self._dist = dist

@@ -27,6 +27,7 @@
 
 #include "common.h"
 #include "file_io.h"
+#include "pcie_mover.h"
 #include "thread_pool.h"
 
 extern "C" hipError_t kvc_launch_gather(const void* const*, const uint64_t*, int,
@@ -144,6 +145,8 @@ class StorageOffloadEngine {
     pool_ = std::make_unique<IoThreadPool>(
         cfg_.io_threads, cfg_.gpu_mode, cfg_.device, max_file_bytes, device_staging,
         cfg_.read_preferring_ratio, mapped);
+    if (cfg_.gpu_mode && cfg_.copy_path == CopyPath::kStaged)
+      mover_ = std::make_unique<PcieMover>(true, cfg_.device);
   }
 
   ~StorageOffloadEngine() {
@@ -360,12 +363,22 @@ class StorageOffloadEngine {
           nb, kernel_dst, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       if (cfg_.copy_path == CopyPath::kStaged) {
-        // PCIe hop on the SDMA engines: zero CU occupancy.
-        KVO_HIP_CHECK(hipMemcpyAsync(ctx.host_staging->host(),
-                                     ctx.device_staging->ptr(), bytes,
-                                     hipMemcpyDeviceToHost, ctx.stream));
+        // PCIe hop via the serialized SDMA mover (zero CU occupancy; one
+        // stream per direction saturates the wire — see pcie_mover.h).
+        hipEvent_t gather_done;
+        KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
+        KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
+        try {
+          mover_->d2h(ctx.host_staging->host(), ctx.device_staging->ptr(), bytes,
+                      gather_done);
+        } catch (...) {
+          (void)hipEventDestroy(gather_done);
+          throw;
+        }
+        (void)hipEventDestroy(gather_done);
+      } else {
+        KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       }
-      KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
     }
     write_file_atomic(ft.path, ctx.host_staging->host(), bytes);
     stats_inc([&](EngineStats& s) {
@@ -393,9 +406,8 @@ class StorageOffloadEngine {
     } else {
       const uint8_t* kernel_src = ctx.host_staging->device();
       if (cfg_.copy_path == CopyPath::kStaged) {
-        KVO_HIP_CHECK(hipMemcpyAsync(ctx.device_staging->ptr(),
-                                     ctx.host_staging->host(), bytes,
-                                     hipMemcpyHostToDevice, ctx.stream));
+        // blocking SDMA hop; the scatter launched after return is ordered
+        mover_->h2d(ctx.device_staging->ptr(), ctx.host_staging->host(), bytes);
         kernel_src = ctx.device_staging->ptr();
       }
       hipError_t err = kvc_launch_scatter(
@@ -440,6 +452,7 @@ class StorageOffloadEngine {
   std::vector<void**> dev_layer_ptrs_;
   std::vector<uint64_t*> dev_layer_strides_;
   std::unique_ptr<IoThreadPool> pool_;
+  std::unique_ptr<PcieMover> mover_;
 
   std::mutex jobs_mu_;
   std::condition_variable done_cv_;

@@ -248,11 +248,15 @@ def main():
         done = 0
         while done < n_jobs:
             done += len(store.get_finished())
+            if done < n_jobs:
+                time.sleep(0.0002)
         for i in range(0, FILES_PER_STEP, 8):
             load.transfer_async(hashes[i:i + 8], {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
         done = 0
         while done < n_jobs:
             done += len(load.get_finished())
+            if done < n_jobs:
+                time.sleep(0.0002)
         # steady-state disk management: drop the previous generation
         prev = (step_id - 1) * FILES_PER_STEP + 1
         if step_id > 0:
@@ -327,6 +331,14 @@ def main():
                 "root": root,
                 "files_written": stats.files_written,
                 "engine_avg_write_ms": round(stats.avg_write_seconds * 1e3, 3),
+                "phase_ms": {
+                    "gather": round(stats.t_gather_ms, 1),
+                    "d2h": round(stats.t_d2h_ms, 1),
+                    "write": round(stats.t_write_ms, 1),
+                    "read": round(stats.t_read_ms, 1),
+                    "h2d": round(stats.t_h2d_ms, 1),
+                    "scatter": round(stats.t_scatter_ms, 1),
+                },
             },
         }))
     if dist is not None:

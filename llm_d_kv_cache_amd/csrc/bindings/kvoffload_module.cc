@@ -32,7 +32,13 @@ PYBIND11_MODULE(_kvoffload, m) {
       .def_readonly("errors", &EngineStats::errors)
       .def_readonly("avg_write_seconds", &EngineStats::avg_write_seconds)
       .def_readonly("bytes_stored", &EngineStats::bytes_stored)
-      .def_readonly("bytes_loaded", &EngineStats::bytes_loaded);
+      .def_readonly("bytes_loaded", &EngineStats::bytes_loaded)
+      .def_readonly("t_gather_ms", &EngineStats::t_gather_ms)
+      .def_readonly("t_d2h_ms", &EngineStats::t_d2h_ms)
+      .def_readonly("t_write_ms", &EngineStats::t_write_ms)
+      .def_readonly("t_read_ms", &EngineStats::t_read_ms)
+      .def_readonly("t_h2d_ms", &EngineStats::t_h2d_ms)
+      .def_readonly("t_scatter_ms", &EngineStats::t_scatter_ms);
 
   py::class_<StorageOffloadEngine>(m, "StorageOffloadEngine")
       .def(py::init([](std::vector<std::tuple<std::vector<uintptr_t>,

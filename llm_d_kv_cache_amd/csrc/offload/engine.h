@@ -87,6 +87,9 @@ struct EngineStats {
   double avg_write_seconds = 0;
   uint64_t bytes_stored = 0;
   uint64_t bytes_loaded = 0;
+  // accumulated per-phase wall time across all workers (ms)
+  double t_gather_ms = 0, t_d2h_ms = 0, t_write_ms = 0;
+  double t_read_ms = 0, t_h2d_ms = 0, t_scatter_ms = 0;
 };
 
 class StorageOffloadEngine {
@@ -353,6 +356,7 @@ class StorageOffloadEngine {
     } else {
       // KV-ready fence: the gather must observe the serving engine's
       // completed KV writes for these blocks.
+      double t0 = now_s();
       KVO_HIP_CHECK(hipStreamWaitEvent(ctx.stream, job.kv_ready, 0));
       uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
                                 ? ctx.device_staging->ptr()
@@ -362,6 +366,8 @@ class StorageOffloadEngine {
           dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
           nb, kernel_dst, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+      double t1 = now_s();
+      stats_inc([&](EngineStats& s) { s.t_gather_ms += (t1 - t0) * 1e3; });
       if (cfg_.copy_path == CopyPath::kStaged) {
         // PCIe hop via the serialized SDMA mover (zero CU occupancy; one
         // stream per direction saturates the wire — see pcie_mover.h).
@@ -379,11 +385,16 @@ class StorageOffloadEngine {
       } else {
         KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       }
+      double t2 = now_s();
+      stats_inc([&](EngineStats& s) { s.t_d2h_ms += (t2 - t1) * 1e3; });
     }
+    double tw = now_s();
     write_file_atomic(ft.path, ctx.host_staging->host(), bytes);
+    double tw2 = now_s();
     stats_inc([&](EngineStats& s) {
       s.files_written++;
       s.bytes_stored += bytes;
+      s.t_write_ms += (tw2 - tw) * 1e3;
     });
   }
 
@@ -398,8 +409,11 @@ class StorageOffloadEngine {
     if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
       throw FileIoError("file " + ft.path + " does not cover requested span");
 
+    double t0 = now_s();
     read_file_range(ft.path, offset, ctx.host_staging->host(), bytes);
     touch_atime(ft.path);
+    double t1 = now_s();
+    stats_inc([&](EngineStats& s) { s.t_read_ms += (t1 - t0) * 1e3; });
 
     if (cfg_.copy_path == CopyPath::kHostMemcpy) {
       scatter_host(g, ft.block_ids, ctx.host_staging->host());
@@ -410,12 +424,16 @@ class StorageOffloadEngine {
         mover_->h2d(ctx.device_staging->ptr(), ctx.host_staging->host(), bytes);
         kernel_src = ctx.device_staging->ptr();
       }
+      double t2 = now_s();
+      stats_inc([&](EngineStats& s) { s.t_h2d_ms += (t2 - t1) * 1e3; });
       hipError_t err = kvc_launch_scatter(
           const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
           dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
           nb, kernel_src, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
+      double t3 = now_s();
+      stats_inc([&](EngineStats& s) { s.t_scatter_ms += (t3 - t2) * 1e3; });
     }
     stats_inc([&](EngineStats& s) {
       s.files_read++;

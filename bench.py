@@ -236,11 +236,14 @@ def main():
     assert blocks_per_step <= args.device_blocks
     step_bytes = FILES_PER_STEP * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
 
+    phase_wall = {"store": 0.0, "load": 0.0, "unlink": 0.0}
+
     def run_step(step_id):
         base = step_id * FILES_PER_STEP + 1
         hashes = list(range(base, base + FILES_PER_STEP))
         ids = list(range(blocks_per_step))
         n_jobs = 0
+        p0 = time.perf_counter()
         # store in 8-file jobs to pipeline the I/O pool
         for i in range(0, FILES_PER_STEP, 8):
             store.transfer_async(hashes[i:i + 8], {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
@@ -250,6 +253,7 @@ def main():
             done += len(store.get_finished())
             if done < n_jobs:
                 time.sleep(0.0002)
+        p1 = time.perf_counter()
         for i in range(0, FILES_PER_STEP, 8):
             load.transfer_async(hashes[i:i + 8], {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
         done = 0
@@ -257,6 +261,7 @@ def main():
             done += len(load.get_finished())
             if done < n_jobs:
                 time.sleep(0.0002)
+        p2 = time.perf_counter()
         # steady-state disk management: drop the previous generation
         prev = (step_id - 1) * FILES_PER_STEP + 1
         if step_id > 0:
@@ -265,6 +270,10 @@ def main():
                     os.unlink(mapper.file_name(h, 0))
                 except OSError:
                     pass
+        p3 = time.perf_counter()
+        phase_wall["store"] += p1 - p0
+        phase_wall["load"] += p2 - p1
+        phase_wall["unlink"] += p3 - p2
 
     def barrier():
         if dist is not None:
@@ -331,6 +340,10 @@ def main():
                 "root": root,
                 "files_written": stats.files_written,
                 "engine_avg_write_ms": round(stats.avg_write_seconds * 1e3, 3),
+                "phase_wall_ms_per_step": {
+                    k: round(v / (args.steps + args.warmup) * 1e3, 1)
+                    for k, v in phase_wall.items()
+                },
                 "phase_ms": {
                     "gather": round(stats.t_gather_ms, 1),
                     "d2h": round(stats.t_d2h_ms, 1),

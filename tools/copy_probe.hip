@@ -82,6 +82,7 @@ void bench_threads(int nthreads) {
          nthreads, N * (double)ITERS * nthreads / dt / 1e9);
 }
 
+void run_file_bench();
 int main() {
   int count = 0;
   hipGetDeviceCount(&count);
@@ -97,5 +98,89 @@ int main() {
   bench("WithStream + default pinned", hipHostMallocDefault, true, true);
   bench("sync hipMemcpy + default pinned", hipHostMallocDefault, false, true, true);
   bench_threads(4);
+  run_file_bench();
   return 0;
+}
+
+// ---- file I/O section: tmpfs write/read from pinned vs malloc buffers ----
+#include <fcntl.h>
+#include <unistd.h>
+#include <cstring>
+#include <atomic>
+
+static void file_bench(const char* name, void* buf, size_t file_sz, int nthreads) {
+  std::vector<std::thread> ts;
+  std::atomic<int> ctr{0};
+  double t0 = now();
+  for (int t = 0; t < nthreads; ++t) {
+    ts.emplace_back([&, t] {
+      char path[128];
+      snprintf(path, sizeof(path), "/dev/shm/probe_%s_%d.bin", name, t);
+      for (int i = 0; i < 8; ++i) {
+        int fd = open(path, O_WRONLY | O_CREAT | O_TRUNC, 0644);
+        size_t off = 0;
+        while (off < file_sz) {
+          ssize_t w = pwrite(fd, (char*)buf + (t * file_sz) + off, file_sz - off, off);
+          if (w <= 0) break;
+          off += w;
+        }
+        close(fd);
+      }
+      ctr++;
+    });
+  }
+  for (auto& th : ts) th.join();
+  double wt = now() - t0;
+  printf("%-34s write %d thr: %.2f GB/s\n", name, nthreads,
+         file_sz * 8.0 * nthreads / wt / 1e9);
+  t0 = now();
+  ts.clear();
+  for (int t = 0; t < nthreads; ++t) {
+    ts.emplace_back([&, t] {
+      char path[128];
+      snprintf(path, sizeof(path), "/dev/shm/probe_%s_%d.bin", name, t);
+      for (int i = 0; i < 8; ++i) {
+        int fd = open(path, O_RDONLY);
+        size_t off = 0;
+        while (off < file_sz) {
+          ssize_t r = pread(fd, (char*)buf + (t * file_sz) + off, file_sz - off, off);
+          if (r <= 0) break;
+          off += r;
+        }
+        close(fd);
+      }
+    });
+  }
+  for (auto& th : ts) th.join();
+  double rt = now() - t0;
+  printf("%-34s read  %d thr: %.2f GB/s\n", name, nthreads,
+         file_sz * 8.0 * nthreads / rt / 1e9);
+  for (int t = 0; t < nthreads; ++t) {
+    char path[128];
+    snprintf(path, sizeof(path), "/dev/shm/probe_%s_%d.bin", name, t);
+    unlink(path);
+  }
+}
+
+void run_file_bench() {
+  const size_t FSZ = 32ull << 20;
+  const int NT = 16;
+  void* mbuf = malloc(FSZ * NT);
+  memset(mbuf, 1, FSZ * NT);
+  file_bench("malloc", mbuf, FSZ, 1);
+  file_bench("malloc", mbuf, FSZ, NT);
+  free(mbuf);
+  void* pbuf;
+  if (hipHostMalloc(&pbuf, FSZ * NT, hipHostMallocPortable) == hipSuccess) {
+    memset(pbuf, 1, FSZ * NT);
+    file_bench("pinned-portable", pbuf, FSZ, 1);
+    file_bench("pinned-portable", pbuf, FSZ, NT);
+    hipHostFree(pbuf);
+  }
+  if (hipHostMalloc(&pbuf, FSZ * NT, 0) == hipSuccess) {
+    memset(pbuf, 1, FSZ * NT);
+    file_bench("pinned-default", pbuf, FSZ, 1);
+    file_bench("pinned-default", pbuf, FSZ, NT);
+    hipHostFree(pbuf);
+  }
 }

@@ -40,6 +40,24 @@ class InMemoryIndexConfig:
 
 
 @dataclass
+class CostAwareMemoryIndexConfig:
+    """Byte-budget index (reference CostAwareMemoryIndex parity): evicts
+    LRU keys until the approximate resident size fits max_bytes."""
+    max_bytes: int = 2 * 1024**3  # reference default "2GiB"
+    pod_cache_size: int = 10
+    shards: int = 64
+
+
+@dataclass
+class RedisIndexConfig:
+    """Network-backed shared index (Redis or Valkey — wire compatible)."""
+    host: str = "127.0.0.1"
+    port: int = 6379
+    pool_size: int = 4
+    key_prefix: str = "kv"
+
+
+@dataclass
 class KVCacheBackendConfig:
     name: str
     weight: float
@@ -52,6 +70,10 @@ def default_backend_configs() -> List[KVCacheBackendConfig]:
 @dataclass
 class IndexerConfig:
     token_processor: TokenProcessorConfig = field(default_factory=TokenProcessorConfig)
+    # First-non-None backend wins (reference index.go:68-93 selection order:
+    # cost-aware > valkey/redis > in-memory).
+    cost_aware_index: Optional[CostAwareMemoryIndexConfig] = None
+    redis_index: Optional[RedisIndexConfig] = None
     index: InMemoryIndexConfig = field(default_factory=InMemoryIndexConfig)
     backends: List[KVCacheBackendConfig] = field(default_factory=default_backend_configs)
 
@@ -67,11 +89,24 @@ class KVCacheIndexer:
             self.config.token_processor.block_size_tokens,
             self.config.token_processor.hash_seed,
         )
-        self.index = k.InMemoryIndex(
-            size=self.config.index.size,
-            pods_per_key=self.config.index.pod_cache_size,
-            shards=self.config.index.shards,
-        )
+        if self.config.cost_aware_index is not None:
+            c = self.config.cost_aware_index
+            self.index = k.InMemoryIndex(
+                pods_per_key=c.pod_cache_size, shards=c.shards,
+                max_bytes=c.max_bytes,
+            )
+        elif self.config.redis_index is not None:
+            c = self.config.redis_index
+            self.index = k.RedisIndex(
+                host=c.host, port=c.port, pool_size=c.pool_size,
+                key_prefix=c.key_prefix,
+            )
+        else:
+            self.index = k.InMemoryIndex(
+                size=self.config.index.size,
+                pods_per_key=self.config.index.pod_cache_size,
+                shards=self.config.index.shards,
+            )
         weights = {b.name: b.weight for b in self.config.backends}
         self._indexer = k.Indexer(self.token_processor, self.index, weights)
 

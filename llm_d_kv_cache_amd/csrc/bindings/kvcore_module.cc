@@ -8,6 +8,7 @@
 #include <pybind11/stl.h>
 
 #include "../core/indexer.h"
+#include "../core/redis_index.h"
 #include "../core/token_processor.h"
 #include "../events/pool.h"
 #include "../events/zmtp.h"
@@ -25,7 +26,7 @@ struct PyPodEntry {
   std::optional<int32_t> group;
 };
 
-PodEntry to_native(InMemoryIndex& idx, const PyPodEntry& e) {
+PodEntry to_native(IndexBackend& idx, const PyPodEntry& e) {
   PodEntry n;
   n.pod = idx.strings().intern(e.pod);
   n.tier = idx.strings().intern(e.tier);
@@ -37,7 +38,7 @@ PodEntry to_native(InMemoryIndex& idx, const PyPodEntry& e) {
   return n;
 }
 
-PyPodEntry from_native(InMemoryIndex& idx, const PodEntry& e) {
+PyPodEntry from_native(IndexBackend& idx, const PodEntry& e) {
   PyPodEntry p;
   p.pod = idx.strings().get(e.pod);
   p.tier = idx.strings().get(e.tier);
@@ -144,19 +145,10 @@ PYBIND11_MODULE(_kvcore, m) {
       .def_readonly("hits", &IndexStats::hits)
       .def_readonly("keys", &IndexStats::keys);
 
-  py::class_<InMemoryIndex, std::shared_ptr<InMemoryIndex>>(m, "InMemoryIndex")
-      .def(py::init([](size_t size, size_t pods_per_key, size_t shards) {
-             InMemoryIndexConfig cfg;
-             cfg.size = size;
-             cfg.pods_per_key = pods_per_key;
-             cfg.shards = shards;
-             return std::make_shared<InMemoryIndex>(cfg);
-           }),
-           py::arg("size") = 100000000, py::arg("pods_per_key") = 10,
-           py::arg("shards") = 64)
+  py::class_<IndexBackend, std::shared_ptr<IndexBackend>>(m, "IndexBackend")
       .def(
           "lookup",
-          [](InMemoryIndex& idx, const std::vector<uint64_t>& keys,
+          [](IndexBackend& idx, const std::vector<uint64_t>& keys,
              const std::vector<std::string>& pods) {
             std::unordered_set<uint32_t> filter;
             bool any_unknown_only = !pods.empty();
@@ -183,7 +175,7 @@ PYBIND11_MODULE(_kvcore, m) {
           py::arg("keys"), py::arg("pods") = std::vector<std::string>{})
       .def(
           "add",
-          [](InMemoryIndex& idx, const std::vector<uint64_t>& engine_keys,
+          [](IndexBackend& idx, const std::vector<uint64_t>& engine_keys,
              const std::vector<uint64_t>& request_keys,
              const std::vector<PyPodEntry>& entries) {
             std::vector<PodEntry> native;
@@ -195,7 +187,7 @@ PYBIND11_MODULE(_kvcore, m) {
           py::arg("engine_keys"), py::arg("request_keys"), py::arg("entries"))
       .def(
           "evict",
-          [](InMemoryIndex& idx, uint64_t key, const std::string& key_type,
+          [](IndexBackend& idx, uint64_t key, const std::string& key_type,
              const std::vector<PyPodEntry>& entries) {
             KeyType t;
             if (key_type == "engine")
@@ -211,19 +203,54 @@ PYBIND11_MODULE(_kvcore, m) {
           },
           py::arg("key"), py::arg("key_type"), py::arg("entries"))
       .def("get_request_key",
-           [](InMemoryIndex& idx, uint64_t ek) -> py::object {
+           [](IndexBackend& idx, uint64_t ek) -> py::object {
              uint64_t rk;
-             if (idx.get_request_key(ek, &rk)) return py::cast(rk);
+             bool found;
+             {
+               py::gil_scoped_release rel;
+               found = idx.get_request_key(ek, &rk);
+             }
+             if (found) return py::cast(rk);
              return py::none();
            })
       .def("clear",
-           [](InMemoryIndex& idx, const std::string& pod) {
+           [](IndexBackend& idx, const std::string& pod) {
              uint32_t id = idx.strings().find(pod);
              if (id == StringTable::kInvalid) return;
              py::gil_scoped_release rel;
              idx.clear(id);
            })
-      .def("stats", &InMemoryIndex::stats);
+      .def("stats", &IndexBackend::stats,
+           py::call_guard<py::gil_scoped_release>());
+
+  py::class_<InMemoryIndex, IndexBackend, std::shared_ptr<InMemoryIndex>>(
+      m, "InMemoryIndex")
+      .def(py::init([](size_t size, size_t pods_per_key, size_t shards,
+                       size_t max_bytes) {
+             InMemoryIndexConfig cfg;
+             cfg.size = size;
+             cfg.pods_per_key = pods_per_key;
+             cfg.shards = shards;
+             cfg.max_bytes = max_bytes;
+             return std::make_shared<InMemoryIndex>(cfg);
+           }),
+           py::arg("size") = 100000000, py::arg("pods_per_key") = 10,
+           py::arg("shards") = 64, py::arg("max_bytes") = 0);
+
+  py::class_<RedisIndex, IndexBackend, std::shared_ptr<RedisIndex>>(
+      m, "RedisIndex")
+      .def(py::init([](const std::string& host, int port, size_t pool_size,
+                       const std::string& key_prefix) {
+             RedisIndexConfig cfg;
+             cfg.host = host;
+             cfg.port = port;
+             cfg.pool_size = pool_size;
+             cfg.key_prefix = key_prefix;
+             py::gil_scoped_release rel;
+             return std::make_shared<RedisIndex>(cfg);
+           }),
+           py::arg("host") = "127.0.0.1", py::arg("port") = 6379,
+           py::arg("pool_size") = 4, py::arg("key_prefix") = "kv");
 
   py::class_<ScoreResult>(m, "ScoreResult")
       .def_readonly("scores", &ScoreResult::scores)
@@ -231,7 +258,7 @@ PYBIND11_MODULE(_kvcore, m) {
       .def_readonly("hit_blocks", &ScoreResult::hit_blocks);
 
   py::class_<Indexer>(m, "Indexer")
-      .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<InMemoryIndex>,
+      .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<IndexBackend>,
                     std::unordered_map<std::string, double>>(),
            py::arg("token_processor"), py::arg("index"), py::arg("tier_weights"))
       .def(
@@ -262,7 +289,7 @@ PYBIND11_MODULE(_kvcore, m) {
       .def_readonly("dropped_parent_misses", &PoolStats::dropped_parent_misses);
 
   py::class_<EventPool, std::shared_ptr<EventPool>>(m, "EventPool")
-      .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<InMemoryIndex>,
+      .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<IndexBackend>,
                     size_t>(),
            py::arg("token_processor"), py::arg("index"), py::arg("concurrency") = 4)
       .def("start", &EventPool::start, py::call_guard<py::gil_scoped_release>())

@@ -11,6 +11,7 @@
 // whole check-and-act.
 #pragma once
 
+#include <algorithm>
 #include <atomic>
 #include <cstdint>
 #include <list>
@@ -51,9 +52,39 @@ struct InMemoryIndexConfig {
   size_t size = 100000000;   // max request keys (reference default 1e8)
   size_t pods_per_key = 10;  // max pod entries per key
   size_t shards = 64;        // power of two
+  // Byte budget (0 = unbounded): when set, the index evicts LRU keys until
+  // the approximate resident size fits (cost-aware mode, capability parity
+  // with the reference CostAwareMemoryIndex / ristretto backend —
+  // implemented as strict sharded LRU-by-bytes instead of probabilistic
+  // admission).
+  size_t max_bytes = 0;
 };
 
-class InMemoryIndex {
+// Pluggable index backend interface (capability parity with the reference
+// Index interface, pkg/kvcache/kvblock/index.go:120-155). Implementations:
+// InMemoryIndex (below, with optional byte budget) and RedisIndex
+// (redis_index.h, network-backed for Valkey/Redis).
+class IndexBackend {
+ public:
+  virtual ~IndexBackend() = default;
+  virtual std::vector<std::pair<uint64_t, std::vector<PodEntry>>> lookup(
+      const std::vector<uint64_t>& request_keys,
+      const std::unordered_set<uint32_t>& pod_filter) = 0;
+  virtual void add(const std::vector<uint64_t>& engine_keys,
+                   const std::vector<uint64_t>& request_keys,
+                   const std::vector<PodEntry>& entries) = 0;
+  virtual void evict(uint64_t key, KeyType type,
+                     const std::vector<PodEntry>& entries) = 0;
+  virtual bool get_request_key(uint64_t engine_key, uint64_t* out) = 0;
+  virtual void clear(uint32_t pod_id) = 0;
+  virtual IndexStats stats() const = 0;
+  StringTable& strings() { return strings_; }
+
+ protected:
+  StringTable strings_;
+};
+
+class InMemoryIndex : public IndexBackend {
  public:
   explicit InMemoryIndex(const InMemoryIndexConfig& cfg = {}) : cfg_(cfg) {
     if (cfg_.shards == 0 || (cfg_.shards & (cfg_.shards - 1)) != 0)
@@ -63,14 +94,12 @@ class InMemoryIndex {
     shard_cap_ = std::max<size_t>(1, cfg_.size / cfg_.shards);
   }
 
-  StringTable& strings() { return strings_; }
-
   // Lookup keys in order. Filter by pod ids when filter is non-empty.
   // A key that is present but has no (filtered) pods ends the scan early
   // (prefix chain is broken); an absent key is simply skipped.
   std::vector<std::pair<uint64_t, std::vector<PodEntry>>> lookup(
       const std::vector<uint64_t>& request_keys,
-      const std::unordered_set<uint32_t>& pod_filter) {
+      const std::unordered_set<uint32_t>& pod_filter) override {
     if (request_keys.empty())
       throw std::invalid_argument("no request keys provided for lookup");
     lookups_.fetch_add(1, std::memory_order_relaxed);
@@ -106,7 +135,7 @@ class InMemoryIndex {
   // the length ratio (both derive from one token count, so they divide).
   void add(const std::vector<uint64_t>& engine_keys,
            const std::vector<uint64_t>& request_keys,
-           const std::vector<PodEntry>& entries) {
+           const std::vector<PodEntry>& entries) override {
     if (request_keys.empty() || entries.empty())
       throw std::invalid_argument("no keys or entries provided for add");
 
@@ -138,11 +167,13 @@ class InMemoryIndex {
         it = sh.map.emplace(key, KeyEntry{}).first;
         sh.lru.push_front(key);
         it->second.lru_it = sh.lru.begin();
+        sh.bytes += kKeyOverheadBytes;
         if (sh.map.size() > shard_cap_) evict_lru_locked(sh);
       } else {
         sh.touch(it);
       }
       auto& pods = it->second.pods;
+      size_t before = pods.size();
       for (const auto& e : entries) {
         // Per-key pod LRU: move-to-front on re-add, bounded capacity.
         for (size_t i = 0; i < pods.size(); ++i) {
@@ -154,7 +185,14 @@ class InMemoryIndex {
         pods.insert(pods.begin(), e);
         if (pods.size() > cfg_.pods_per_key) pods.pop_back();
       }
+      sh.bytes += (pods.size() - before) * sizeof(PodEntry);
       admissions_.fetch_add(entries.size(), std::memory_order_relaxed);
+      if (cfg_.max_bytes > 0) {
+        // byte-budget mode: evict LRU keys until this shard fits its slice
+        size_t shard_budget = cfg_.max_bytes / cfg_.shards;
+        while (sh.bytes > shard_budget && sh.map.size() > 1)
+          evict_lru_locked(sh);
+      }
     }
   }
 
@@ -162,7 +200,8 @@ class InMemoryIndex {
   // the bridge (possibly to several request keys); request keys apply
   // directly. When every resolved request key is gone/empty the bridge
   // mapping itself is dropped.
-  void evict(uint64_t key, KeyType type, const std::vector<PodEntry>& entries) {
+  void evict(uint64_t key, KeyType type,
+             const std::vector<PodEntry>& entries) override {
     if (entries.empty())
       throw std::invalid_argument("no entries provided for evict");
     if (type == KeyType::kRequest) {
@@ -202,7 +241,7 @@ class InMemoryIndex {
   // Last request key of the engine key's chain segment (the one whose chunk
   // ends where the engine block ends) — what parent-hash resolution needs.
   // Returns false when the mapping is missing (e.g. already evicted).
-  bool get_request_key(uint64_t engine_key, uint64_t* out) {
+  bool get_request_key(uint64_t engine_key, uint64_t* out) override {
     EngShard& es = eng_shard(engine_key);
     std::lock_guard<std::mutex> g(es.mu);
     auto it = es.map.find(engine_key);
@@ -216,7 +255,7 @@ class InMemoryIndex {
   // path (backs AllBlocksCleared). The engine bridge is intentionally left
   // alone: stale mappings resolve to emptied keys that break the prefix
   // chain correctly and the LRU self-heals.
-  void clear(uint32_t pod_id) {
+  void clear(uint32_t pod_id) override {
     for (auto& sh : shards_) {
       std::lock_guard<std::mutex> g(sh.mu);
       for (auto it = sh.map.begin(); it != sh.map.end();) {
@@ -226,8 +265,10 @@ class InMemoryIndex {
                                   [pod_id](const PodEntry& e) { return e.pod == pod_id; }),
                    pods.end());
         evictions_.fetch_add(before - pods.size(), std::memory_order_relaxed);
+        sh.bytes -= (before - pods.size()) * sizeof(PodEntry);
         if (pods.empty()) {
           sh.lru.erase(it->second.lru_it);
+          sh.bytes -= kKeyOverheadBytes;
           it = sh.map.erase(it);
         } else {
           ++it;
@@ -236,7 +277,7 @@ class InMemoryIndex {
     }
   }
 
-  IndexStats stats() const {
+  IndexStats stats() const override {
     IndexStats s;
     s.admissions = admissions_.load(std::memory_order_relaxed);
     s.evictions = evictions_.load(std::memory_order_relaxed);
@@ -258,6 +299,7 @@ class InMemoryIndex {
     std::mutex mu;
     std::unordered_map<uint64_t, KeyEntry> map;
     std::list<uint64_t> lru;  // front = most recent
+    size_t bytes = 0;  // approximate resident cost (byte-budget mode)
 
     Shard() = default;
     Shard(const Shard&) {}
@@ -322,6 +364,7 @@ class InMemoryIndex {
       for (size_t i = 0; i < pods.size(); ++i) {
         if (pods[i] == e) {
           pods.erase(pods.begin() + i);
+          sh.bytes -= sizeof(PodEntry);
           evictions_.fetch_add(1, std::memory_order_relaxed);
           break;
         }
@@ -329,9 +372,12 @@ class InMemoryIndex {
     }
     if (pods.empty()) {
       sh.lru.erase(it->second.lru_it);
+      sh.bytes -= kKeyOverheadBytes;
       sh.map.erase(it);
     }
   }
+
+  static constexpr size_t kKeyOverheadBytes = 96;  // map node + LRU node
 
   void evict_lru_locked(Shard& sh) {
     if (sh.lru.empty()) return;
@@ -340,6 +386,7 @@ class InMemoryIndex {
     auto it = sh.map.find(victim);
     if (it != sh.map.end()) {
       evictions_.fetch_add(it->second.pods.size(), std::memory_order_relaxed);
+      sh.bytes -= kKeyOverheadBytes + it->second.pods.size() * sizeof(PodEntry);
       sh.map.erase(it);
     }
   }
@@ -348,7 +395,6 @@ class InMemoryIndex {
   size_t shard_cap_;
   std::vector<Shard> shards_;
   std::vector<EngShard> eng_shards_;
-  StringTable strings_;
   std::atomic<uint64_t> admissions_{0}, evictions_{0}, lookups_{0}, hits_{0};
 };
 

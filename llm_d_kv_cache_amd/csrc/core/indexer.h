@@ -25,7 +25,7 @@ struct ScoreResult {
 
 class Indexer {
  public:
-  Indexer(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<InMemoryIndex> index,
+  Indexer(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<IndexBackend> index,
           std::unordered_map<std::string, double> tier_weights)
       : tp_(std::move(tp)), index_(std::move(index)) {
     std::unordered_map<uint32_t, double> w;
@@ -34,7 +34,7 @@ class Indexer {
     scorer_ = std::make_unique<LongestPrefixScorer>(std::move(w));
   }
 
-  InMemoryIndex& index() { return *index_; }
+  IndexBackend& index() { return *index_; }
   TokenProcessor& token_processor() { return *tp_; }
 
   std::vector<uint64_t> compute_block_keys(const uint32_t* tokens, size_t n,
@@ -70,7 +70,7 @@ class Indexer {
 
  private:
   std::shared_ptr<TokenProcessor> tp_;
-  std::shared_ptr<InMemoryIndex> index_;
+  std::shared_ptr<IndexBackend> index_;
   std::unique_ptr<LongestPrefixScorer> scorer_;
 };
 

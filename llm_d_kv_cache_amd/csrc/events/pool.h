@@ -68,7 +68,7 @@ struct PoolStats {
 
 class EventPool {
  public:
-  EventPool(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<InMemoryIndex> index,
+  EventPool(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<IndexBackend> index,
             size_t concurrency = 4)
       : tp_(std::move(tp)), index_(std::move(index)),
         queues_(std::max<size_t>(1, concurrency)) {}
@@ -287,7 +287,7 @@ class EventPool {
   }
 
   std::shared_ptr<TokenProcessor> tp_;
-  std::shared_ptr<InMemoryIndex> index_;
+  std::shared_ptr<IndexBackend> index_;
   GroupCatalog catalog_;
   std::vector<Queue> queues_;
   std::vector<std::thread> workers_;

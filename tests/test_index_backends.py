@@ -1,0 +1,130 @@
+"""Alternative index backends: byte-budget (cost-aware) mode and the
+Redis/Valkey network backend against an embedded fake server."""
+import pytest
+
+from fake_redis import FakeRedis
+from llm_d_kv_cache_amd import ensure_native
+
+k = ensure_native()
+
+
+def entry(pod, tier="gpu", group=None):
+    return k.PodEntry(pod, tier, False, group)
+
+
+# ---- cost-aware (byte budget) ----------------------------------------------
+
+def test_cost_aware_evicts_to_budget():
+    # budget for roughly a handful of keys per shard
+    idx = k.InMemoryIndex(shards=1, max_bytes=10 * (96 + 16))
+    for i in range(100):
+        idx.add([], [i], [entry("a")])
+    found = idx.lookup(list(range(100)))
+    assert 0 < len(found) <= 10
+    assert 99 in found  # newest survives
+    assert idx.stats().evictions > 0
+
+
+def test_cost_aware_unbounded_by_default():
+    idx = k.InMemoryIndex(shards=1)
+    for i in range(100):
+        idx.add([], [i], [entry("a")])
+    assert len(idx.lookup(list(range(100)))) == 100
+
+
+# ---- redis / valkey ---------------------------------------------------------
+
+@pytest.fixture
+def redis_pair():
+    srv = FakeRedis()
+    idx = k.RedisIndex(host="127.0.0.1", port=srv.port)
+    yield srv, idx
+    srv.close()
+
+
+def test_redis_add_lookup(redis_pair):
+    srv, idx = redis_pair
+    idx.add([], [1, 2], [entry("pod-a"), entry("pod-b", "cpu")])
+    got = idx.lookup([1, 2, 3])
+    assert set(got.keys()) == {1, 2}
+    pods = {(e.pod, e.tier) for e in got[1]}
+    assert pods == {("pod-a", "gpu"), ("pod-b", "cpu")}
+
+
+def test_redis_pod_filter(redis_pair):
+    srv, idx = redis_pair
+    idx.add([], [5], [entry("pod-a"), entry("pod-b")])
+    got = idx.lookup([5], ["pod-b"])
+    assert [e.pod for e in got[5]] == ["pod-b"]
+
+
+def test_redis_engine_bridge(redis_pair):
+    srv, idx = redis_pair
+    idx.add([10, 11], [1, 2, 3, 4], [entry("a")])  # 1:many
+    assert idx.get_request_key(10) == 2
+    assert idx.get_request_key(11) == 4
+    assert idx.get_request_key(99) is None
+
+
+def test_redis_evict_engine_key(redis_pair):
+    srv, idx = redis_pair
+    idx.add([10], [1, 2], [entry("a")])
+    idx.evict(10, "engine", [entry("a")])
+    assert idx.lookup([1, 2]) == {}
+    assert idx.get_request_key(10) is None
+
+
+def test_redis_evict_partial(redis_pair):
+    srv, idx = redis_pair
+    idx.add([10], [1], [entry("a"), entry("b")])
+    idx.evict(10, "engine", [entry("a")])
+    got = idx.lookup([1])
+    assert [e.pod for e in got[1]] == ["b"]
+    assert idx.get_request_key(10) == 1  # mapping survives: key non-empty
+
+
+def test_redis_clear_pod(redis_pair):
+    srv, idx = redis_pair
+    idx.add([], [1, 2], [entry("a"), entry("b")])
+    idx.clear("a")
+    got = idx.lookup([1, 2])
+    assert all(e.pod == "b" for kk in got for e in got[kk])
+
+
+def test_redis_group_flags_roundtrip(redis_pair):
+    srv, idx = redis_pair
+    idx.add([], [7], [entry("a", group=3)])
+    got = idx.lookup([7])
+    assert got[7][0].group == 3
+
+
+def test_redis_shared_state_between_instances(redis_pair):
+    """Two index instances (two replicas) see each other's writes."""
+    srv, idx = redis_pair
+    idx2 = k.RedisIndex(host="127.0.0.1", port=srv.port)
+    idx.add([], [42], [entry("pod-x")])
+    got = idx2.lookup([42])
+    assert [e.pod for e in got[42]] == ["pod-x"]
+
+
+def test_redis_connection_refused():
+    with pytest.raises(Exception):
+        k.RedisIndex(host="127.0.0.1", port=1)  # nothing listens there
+
+
+def test_redis_backend_drives_indexer_and_pool(redis_pair):
+    """Full wiring: Indexer + EventPool on the Redis backend."""
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_stored_payload,
+        encode_batch,
+    )
+
+    srv, idx = redis_pair
+    tp = k.TokenProcessor(16, "")
+    ix = k.Indexer(tp, idx, {"gpu": 1.0, "cpu": 0.8})
+    pool = k.EventPool(tp, idx, 2)
+    tokens = list(range(32))
+    pool.process("kv@pod-r@m", 0,
+                 encode_batch([block_stored_payload([1, 2], None, tokens, 16)]))
+    res = ix.score_tokens(tokens, "m", [])
+    assert res.scores == {"pod-r": 2.0}

@@ -1,0 +1,115 @@
+"""Prometheus metrics for the index + events plane.
+
+Capability parity with the reference pkg/kvcache/metrics/collector.go:
+kvcache_index_{admissions,evictions,lookup_requests,lookup_hits}_total and
+the periodic "metrics beat" log line. Implemented as a custom collector
+that reads the native stats counters on scrape — the C++ hot paths carry
+plain atomics and never touch Python.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Optional
+
+log = logging.getLogger(__name__)
+
+try:
+    from prometheus_client import REGISTRY
+    from prometheus_client.core import CounterMetricFamily, GaugeMetricFamily
+
+    HAVE_PROMETHEUS = True
+except ImportError:  # pragma: no cover
+    HAVE_PROMETHEUS = False
+
+
+class KVCacheMetricsCollector:
+    """Scrape-time collector over native index/pool/engine stats."""
+
+    def __init__(self, indexer=None, events_pool=None, offload_engine=None):
+        self.indexer = indexer
+        self.events_pool = events_pool
+        self.offload_engine = offload_engine
+
+    def collect(self):
+        if self.indexer is not None:
+            s = self.indexer.index.stats()
+            yield CounterMetricFamily(
+                "kvcache_index_admissions_total",
+                "pod entries admitted to the index", value=s.admissions)
+            yield CounterMetricFamily(
+                "kvcache_index_evictions_total",
+                "pod entries evicted from the index", value=s.evictions)
+            yield CounterMetricFamily(
+                "kvcache_index_lookup_requests_total",
+                "index lookups", value=s.lookups)
+            yield CounterMetricFamily(
+                "kvcache_index_lookup_hits_total",
+                "index lookups with at least one hit", value=s.hits)
+            yield GaugeMetricFamily(
+                "kvcache_index_keys", "request keys resident", value=s.keys)
+        if self.events_pool is not None:
+            p = self.events_pool.stats()
+            yield CounterMetricFamily(
+                "kvcache_events_enqueued_total", "raw messages enqueued",
+                value=p.enqueued)
+            yield CounterMetricFamily(
+                "kvcache_events_processed_total", "messages processed",
+                value=p.processed)
+            yield CounterMetricFamily(
+                "kvcache_events_parse_failures_total", "undecodable payloads",
+                value=p.parse_failures)
+            yield CounterMetricFamily(
+                "kvcache_events_parent_misses_total",
+                "BlockStored dropped on unknown parent chain",
+                value=p.dropped_parent_misses)
+        if self.offload_engine is not None:
+            e = self.offload_engine.stats()
+            yield CounterMetricFamily(
+                "kv_offload_files_written_total", "KV files written",
+                value=e.files_written)
+            yield CounterMetricFamily(
+                "kv_offload_files_read_total", "KV files read",
+                value=e.files_read)
+            yield CounterMetricFamily(
+                "kv_offload_files_deduped_total", "stores skipped (exists)",
+                value=e.files_deduped)
+            yield CounterMetricFamily(
+                "kv_offload_writes_dropped_total",
+                "stores dropped by the queue limit", value=e.writes_dropped)
+            yield CounterMetricFamily(
+                "kv_offload_bytes_stored_total", "bytes offloaded",
+                value=e.bytes_stored)
+            yield CounterMetricFamily(
+                "kv_offload_bytes_loaded_total", "bytes loaded back",
+                value=e.bytes_loaded)
+
+
+def register(indexer=None, events_pool=None, offload_engine=None,
+             registry=None):
+    if not HAVE_PROMETHEUS:  # pragma: no cover
+        raise RuntimeError("prometheus_client is not installed")
+    collector = KVCacheMetricsCollector(indexer, events_pool, offload_engine)
+    (registry or REGISTRY).register(collector)
+    return collector
+
+
+def start_metrics_logging(indexer, interval_s: float = 60.0,
+                          stop_event: Optional[threading.Event] = None):
+    """Periodic 'metrics beat' log line (reference collector.go:97-165)."""
+    stop_event = stop_event or threading.Event()
+
+    def beat():
+        while not stop_event.wait(interval_s):
+            s = indexer.index.stats()
+            hit_pct = 100.0 * s.hits / s.lookups if s.lookups else 0.0
+            log.info(
+                "kvcache metrics beat: keys=%d admissions=%d evictions=%d "
+                "lookups=%d hit%%=%.1f", s.keys, s.admissions, s.evictions,
+                s.lookups, hit_pct,
+            )
+
+    t = threading.Thread(target=beat, daemon=True, name="kvcache-metrics-beat")
+    t.start()
+    return stop_event

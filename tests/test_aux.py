@@ -1,0 +1,201 @@
+"""Auxiliary subsystems: Prometheus metrics, tracing, PVC evictor."""
+import os
+import time
+
+import pytest
+
+from llm_d_kv_cache_amd import ensure_native
+from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+
+k = ensure_native()
+
+
+# ---- metrics ----------------------------------------------------------------
+
+def test_prometheus_collector():
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from llm_d_kv_cache_amd.utils.metrics import KVCacheMetricsCollector
+
+    ix = KVCacheIndexer(IndexerConfig())
+    tokens = list(range(32))
+    keys = ix.compute_block_keys(tokens, "m")
+    ix.index.add([], keys, [k.PodEntry("pod-a", "gpu")])
+    ix.score_tokens(tokens, "m")
+
+    reg = CollectorRegistry()
+    reg.register(KVCacheMetricsCollector(indexer=ix))
+    text = generate_latest(reg).decode()
+    assert "kvcache_index_admissions_total 2.0" in text
+    assert "kvcache_index_lookup_requests_total 1.0" in text
+    assert "kvcache_index_lookup_hits_total 1.0" in text
+    assert "kvcache_index_keys 2.0" in text
+
+
+def test_metrics_beat_logs(caplog):
+    import logging
+
+    from llm_d_kv_cache_amd.utils.metrics import start_metrics_logging
+
+    ix = KVCacheIndexer(IndexerConfig())
+    with caplog.at_level(logging.INFO, logger="llm_d_kv_cache_amd.utils.metrics"):
+        stop = start_metrics_logging(ix, interval_s=0.05)
+        time.sleep(0.2)
+        stop.set()
+    assert any("metrics beat" in r.message for r in caplog.records)
+
+
+# ---- tracing ----------------------------------------------------------------
+
+def test_tracing_jsonl_export(tmp_path):
+    import json
+
+    from llm_d_kv_cache_amd.utils import tracing
+
+    path = str(tmp_path / "trace.jsonl")
+    t = tracing.Tracer("test", 1.0, tracing._JsonlExporter(path))
+    with t.span("outer", foo=1) as outer:
+        outer.set_attribute("bar", 2)
+        with t.span("inner"):
+            pass
+    recs = [json.loads(line) for line in open(path)]
+    assert len(recs) == 2
+    inner, outer_rec = recs
+    assert inner["name"] == "test.inner"
+    assert inner["trace_id"] == outer_rec["trace_id"]
+    assert inner["parent_id"] == outer_rec["span_id"]
+    assert outer_rec["attributes"] == {"foo": 1, "bar": 2}
+
+
+def test_tracing_sampling_zero_ratio():
+    from llm_d_kv_cache_amd.utils import tracing
+
+    calls = []
+
+    class Exp:
+        def export(self, span):
+            calls.append(span)
+
+    t = tracing.Tracer("test", 0.0, Exp())
+    with t.span("never") as s:
+        s.set_attribute("x", 1)  # noop span accepts attributes
+    assert calls == []
+
+
+def test_traced_indexer_score_path():
+    from llm_d_kv_cache_amd.utils import tracing
+
+    spans = []
+
+    class Exp:
+        def export(self, span):
+            spans.append(span)
+
+    ix = KVCacheIndexer(IndexerConfig())
+    tokens = list(range(64))
+    keys = ix.compute_block_keys(tokens, "m")
+    ix.index.add([], keys[:2], [k.PodEntry("p", "gpu")])
+    traced = tracing.TracedIndexer(ix, tracing.Tracer("kvc", 1.0, Exp()))
+    scores = traced.score_tokens(tokens, "m")
+    assert scores == {"p": 2.0}
+    assert len(spans) == 1
+    assert spans[0].attributes["total_blocks"] == 4
+    assert spans[0].attributes["hit_blocks"] == 2
+    assert abs(spans[0].attributes["block_hit_ratio"] - 0.5) < 1e-9
+
+
+# ---- evictor ----------------------------------------------------------------
+
+def util_always_high(root):
+    return 0.99
+
+
+def util_always_low(root):
+    return 0.01
+
+
+def _layout_with_files(tmp_path, n=8, age_s=7200, offset=0):
+    """Build a mapper-style layout with aged files."""
+    from llm_d_kv_cache_amd.offload import FileMapper, KVCacheLayoutConfig
+
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="evict-test"))
+    paths = []
+    old = time.time() - age_s
+    for h in range(offset + 1, offset + n + 1):
+        p = mapper.file_name(h * 7919, 0)
+        os.makedirs(os.path.dirname(p), exist_ok=True)
+        with open(p, "wb") as f:
+            f.write(b"x" * 128)
+        os.utime(p, (old, old))
+        paths.append(p)
+    return paths
+
+
+def test_evictor_deletes_cold_files_under_pressure(tmp_path):
+    from llm_d_kv_cache_amd.evictor import EvictorConfig, PvcEvictor
+
+    paths = _layout_with_files(tmp_path, n=8)
+    cfg = EvictorConfig(root=str(tmp_path), crawlers=2, atime_threshold_s=60,
+                        crawl_interval_s=0.1, check_interval_s=0.05)
+    ev = PvcEvictor(cfg, utilization=util_always_high)
+    ev.start()
+    try:
+        deadline = time.time() + 30
+        while any(os.path.exists(p) for p in paths) and time.time() < deadline:
+            time.sleep(0.1)
+        assert not any(os.path.exists(p) for p in paths)
+        assert ev.deleted.value >= 8
+    finally:
+        ev.shutdown()
+
+
+def test_evictor_idle_below_threshold(tmp_path):
+    from llm_d_kv_cache_amd.evictor import EvictorConfig, PvcEvictor
+
+    paths = _layout_with_files(tmp_path, n=4)
+    cfg = EvictorConfig(root=str(tmp_path), crawlers=1, atime_threshold_s=60,
+                        crawl_interval_s=0.1, check_interval_s=0.05)
+    ev = PvcEvictor(cfg, utilization=util_always_low)
+    ev.start()
+    try:
+        time.sleep(1.0)
+        assert all(os.path.exists(p) for p in paths)
+    finally:
+        ev.shutdown()
+
+
+def test_evictor_spares_warm_files(tmp_path):
+    from llm_d_kv_cache_amd.evictor import EvictorConfig, PvcEvictor
+
+    cold = _layout_with_files(tmp_path, n=4, age_s=7200)
+    warm = _layout_with_files(tmp_path, n=2, age_s=0, offset=100)
+    # same run dir; warm files have fresh atime
+    cfg = EvictorConfig(root=str(tmp_path), crawlers=1, atime_threshold_s=3600,
+                        crawl_interval_s=0.1, check_interval_s=0.05)
+    ev = PvcEvictor(cfg, utilization=util_always_high)
+    ev.start()
+    try:
+        deadline = time.time() + 30
+        while any(os.path.exists(p) for p in cold) and time.time() < deadline:
+            time.sleep(0.1)
+        assert not any(os.path.exists(p) for p in cold)
+        assert all(os.path.exists(p) for p in warm)
+    finally:
+        ev.shutdown()
+
+
+def test_evictor_restarts_dead_children(tmp_path):
+    from llm_d_kv_cache_amd.evictor import EvictorConfig, PvcEvictor
+
+    cfg = EvictorConfig(root=str(tmp_path), crawlers=1, crawl_interval_s=0.1,
+                        check_interval_s=0.05)
+    ev = PvcEvictor(cfg, utilization=util_always_low)
+    ev.start()
+    try:
+        victim = ev._procs["activator"]
+        victim.terminate()
+        victim.join(timeout=10)
+        ev.supervise_once()
+        assert ev._procs["activator"].is_alive()
+    finally:
+        ev.shutdown()

@@ -165,6 +165,8 @@ def main():
     ap.add_argument("--root", type=str, default=None)
     ap.add_argument("--copy-path", type=str, default="staged",
                     choices=["staged", "zero_copy"])
+    ap.add_argument("--serialize", type=str, default="raw",
+                    choices=["raw", "fp8_e4m3"])
     ap.add_argument("--io-threads", type=int, default=16)
     ap.add_argument("--device-blocks", type=int, default=2048)
     args = ap.parse_args()
@@ -224,7 +226,8 @@ def main():
         [group],
         OffloadEngineConfig(io_threads=args.io_threads,
                             gpu_blocks_per_file=BLOCKS_PER_FILE,
-                            copy_path=args.copy_path, device=local_rank),
+                            copy_path=args.copy_path, serialize=args.serialize,
+                            device=local_rank),
     )
     mapper = FileMapper(rank_root, KVCacheLayoutConfig(
         model=MODEL, tp_size=1, kv_cache_groups=(("full_attention", 16, BLOCK_BYTES),),
@@ -360,6 +363,7 @@ def main():
                                f"{BLOCK_TOKENS}-token blocks",
                 "bytes_per_step_per_gpu": step_bytes * 2,
                 "copy_path": args.copy_path,
+                "serialize": args.serialize,
                 "io_threads": args.io_threads,
                 "root": root,
                 "files_written": stats.files_written,

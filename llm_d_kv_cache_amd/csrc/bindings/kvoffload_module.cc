@@ -46,7 +46,8 @@ PYBIND11_MODULE(_kvoffload, m) {
                            groups,
                        int io_threads, int gpu_blocks_per_file,
                        double read_preferring_ratio, double max_write_queued_seconds,
-                       bool gpu_mode, int device, const std::string& copy_path) {
+                       bool gpu_mode, int device, const std::string& copy_path,
+                       const std::string& serialize) {
              EngineConfig cfg;
              cfg.io_threads = io_threads;
              cfg.gpu_blocks_per_file = gpu_blocks_per_file;
@@ -63,6 +64,12 @@ PYBIND11_MODULE(_kvoffload, m) {
              else
                throw std::invalid_argument(
                    "copy_path must be staged|zero_copy|host");
+             if (serialize == "raw")
+               cfg.serialize = Serialize::kRaw;
+             else if (serialize == "fp8_e4m3")
+               cfg.serialize = Serialize::kFp8E4M3;
+             else
+               throw std::invalid_argument("serialize must be raw|fp8_e4m3");
              std::vector<GroupDesc> gs;
              for (auto& [ptrs, strides, block_bytes] : groups) {
                GroupDesc g;
@@ -78,7 +85,8 @@ PYBIND11_MODULE(_kvoffload, m) {
            py::arg("gpu_blocks_per_file") = 16,
            py::arg("read_preferring_ratio") = 0.75,
            py::arg("max_write_queued_seconds") = 30.0, py::arg("gpu_mode") = false,
-           py::arg("device") = 0, py::arg("copy_path") = "staged")
+           py::arg("device") = 0, py::arg("copy_path") = "staged",
+           py::arg("serialize") = "raw")
       .def(
           "async_store",
           [](StorageOffloadEngine& e,

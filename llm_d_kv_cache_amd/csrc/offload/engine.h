@@ -19,6 +19,7 @@
 #pragma once
 
 #include <algorithm>
+#include <cmath>
 #include <chrono>
 #include <thread>
 #include <list>
@@ -36,12 +37,23 @@ extern "C" hipError_t kvc_launch_gather(const void* const*, const uint64_t*, int
 extern "C" hipError_t kvc_launch_scatter(const void* const*, const uint64_t*, int,
                                          uint64_t, const int32_t*, int,
                                          const uint8_t*, hipStream_t);
+extern "C" hipError_t kvc_launch_gather_fp8(const void* const*, const uint64_t*,
+                                            int, uint64_t, const int32_t*, int,
+                                            uint8_t*, hipStream_t);
+extern "C" hipError_t kvc_launch_scatter_fp8(const void* const*, const uint64_t*,
+                                             int, uint64_t, const int32_t*, int,
+                                             const uint8_t*, hipStream_t);
 
 namespace kvo {
 
 constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerFile
 
 enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };
+
+// Payload serialization: raw bytes, or fp8 e4m3fn quantization of bf16
+// pages (halves wire + storage bytes; one f32 scale per (block, layer)
+// tile appended to each tile record).
+enum class Serialize { kRaw, kFp8E4M3 };
 
 struct GroupDesc {
   std::vector<void*> layer_ptrs;        // per-layer block-0 base address
@@ -57,6 +69,7 @@ struct EngineConfig {
   bool gpu_mode = false;
   int device = 0;
   CopyPath copy_path = CopyPath::kStaged;
+  Serialize serialize = Serialize::kRaw;
 };
 
 struct FileTransfer {
@@ -341,8 +354,13 @@ class StorageOffloadEngine {
     f(stats_);
   }
 
+  size_t tile_record_bytes(const GroupDesc& g) const {
+    return cfg_.serialize == Serialize::kFp8E4M3 ? g.block_bytes / 2 + 4
+                                                 : g.block_bytes;
+  }
+
   size_t file_bytes(const GroupDesc& g, size_t n_blocks) const {
-    return n_blocks * g.layer_ptrs.size() * g.block_bytes;
+    return n_blocks * g.layer_ptrs.size() * tile_record_bytes(g);
   }
 
   void store_one(WorkerCtx& ctx, Job& job, const FileTransfer& ft) {
@@ -361,10 +379,16 @@ class StorageOffloadEngine {
       uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
                                 ? ctx.device_staging->ptr()
                                 : ctx.host_staging->device();
-      hipError_t err = kvc_launch_gather(
-          const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
-          dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
-          nb, kernel_dst, ctx.stream);
+      hipError_t err =
+          cfg_.serialize == Serialize::kFp8E4M3
+              ? kvc_launch_gather_fp8(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, kernel_dst, ctx.stream)
+              : kvc_launch_gather(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, kernel_dst, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       double t1 = now_s();
       stats_inc([&](EngineStats& s) { s.t_gather_ms += (t1 - t0) * 1e3; });
@@ -403,7 +427,8 @@ class StorageOffloadEngine {
     const int nb = static_cast<int>(ft.block_ids.size());
     const int nl = static_cast<int>(g.layer_ptrs.size());
     const size_t bytes = file_bytes(g, nb);
-    const uint64_t offset = static_cast<uint64_t>(ft.slot_offset) * nl * g.block_bytes;
+    const uint64_t offset =
+        static_cast<uint64_t>(ft.slot_offset) * nl * tile_record_bytes(g);
 
     int64_t fsz = file_size(ft.path);
     if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
@@ -426,10 +451,16 @@ class StorageOffloadEngine {
       }
       double t2 = now_s();
       stats_inc([&](EngineStats& s) { s.t_h2d_ms += (t2 - t1) * 1e3; });
-      hipError_t err = kvc_launch_scatter(
-          const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
-          dev_layer_strides_[ft.group], nl, g.block_bytes, ft.block_ids.data(),
-          nb, kernel_src, ctx.stream);
+      hipError_t err =
+          cfg_.serialize == Serialize::kFp8E4M3
+              ? kvc_launch_scatter_fp8(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, kernel_src, ctx.stream)
+              : kvc_launch_scatter(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, kernel_src, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       double t3 = now_s();
@@ -444,11 +475,16 @@ class StorageOffloadEngine {
   void gather_host(const GroupDesc& g, const std::vector<int32_t>& ids,
                    uint8_t* dst) const {
     const size_t nl = g.layer_ptrs.size();
+    const size_t rec = tile_record_bytes(g);
     for (size_t bi = 0; bi < ids.size(); ++bi) {
       for (size_t l = 0; l < nl; ++l) {
         const uint8_t* src = static_cast<const uint8_t*>(g.layer_ptrs[l]) +
                              static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
-        std::memcpy(dst + (bi * nl + l) * g.block_bytes, src, g.block_bytes);
+        uint8_t* out = dst + (bi * nl + l) * rec;
+        if (cfg_.serialize == Serialize::kFp8E4M3)
+          fp8_quantize_tile(src, g.block_bytes, out);
+        else
+          std::memcpy(out, src, g.block_bytes);
       }
     }
   }
@@ -456,13 +492,101 @@ class StorageOffloadEngine {
   void scatter_host(const GroupDesc& g, const std::vector<int32_t>& ids,
                     const uint8_t* src) const {
     const size_t nl = g.layer_ptrs.size();
+    const size_t rec = tile_record_bytes(g);
     for (size_t bi = 0; bi < ids.size(); ++bi) {
       for (size_t l = 0; l < nl; ++l) {
         uint8_t* dst = static_cast<uint8_t*>(g.layer_ptrs[l]) +
                        static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
-        std::memcpy(dst, src + (bi * nl + l) * g.block_bytes, g.block_bytes);
+        const uint8_t* in = src + (bi * nl + l) * rec;
+        if (cfg_.serialize == Serialize::kFp8E4M3)
+          fp8_dequantize_tile(in, g.block_bytes, dst);
+        else
+          std::memcpy(dst, in, g.block_bytes);
       }
     }
+  }
+
+  // ---- software fp8 e4m3fn (host-mode twin of the CDNA4 kernels) -----------
+
+  static float bf16_to_f32(uint16_t u) {
+    uint32_t w = static_cast<uint32_t>(u) << 16;
+    float f;
+    std::memcpy(&f, &w, 4);
+    return f;
+  }
+  static uint16_t f32_to_bf16(float f) {
+    uint32_t w;
+    std::memcpy(&w, &f, 4);
+    uint32_t rounding = 0x7fff + ((w >> 16) & 1);
+    return static_cast<uint16_t>((w + rounding) >> 16);
+  }
+  static uint8_t f32_to_fp8(float x) {
+    // OCP e4m3fn: 1s 4e 3m, bias 7, max 448, no inf, single NaN.
+    if (x != x) return 0x7f;
+    uint8_t sign = x < 0 ? 0x80 : 0;
+    float a = std::abs(x);
+    if (a >= 448.0f) return sign | 0x7e;  // saturate to max normal
+    if (a < 0.0009765625f) {              // subnormal range (< 2^-10 = min subnormal/2... )
+      // subnormals: value = m * 2^-9, m in [0,7]
+      int m = static_cast<int>(a * 512.0f + 0.5f);
+      if (m > 7) m = 7;
+      return sign | static_cast<uint8_t>(m);
+    }
+    int e;
+    float frac = std::frexp(a, &e);  // a = frac * 2^e, frac in [0.5, 1)
+    // normalized: a = 1.mmm * 2^(e-1); exponent field = (e-1)+7
+    int exp_field = e - 1 + 7;
+    float mant = frac * 2.0f - 1.0f;  // [0,1)
+    int m = static_cast<int>(mant * 8.0f + 0.5f);
+    if (m == 8) {
+      m = 0;
+      exp_field += 1;
+    }
+    if (exp_field >= 16) return sign | 0x7e;
+    if (exp_field <= 0) {
+      // underflow into subnormal
+      int ms = static_cast<int>(a * 512.0f + 0.5f);
+      if (ms > 7) ms = 7;
+      return sign | static_cast<uint8_t>(ms);
+    }
+    return sign | static_cast<uint8_t>((exp_field << 3) | m);
+  }
+  static float fp8_to_f32(uint8_t b) {
+    uint8_t sign = b & 0x80;
+    int exp_field = (b >> 3) & 0xf;
+    int m = b & 0x7;
+    if (exp_field == 0xf && m == 0x7) return NAN;
+    float v;
+    if (exp_field == 0)
+      v = m * 0.001953125f;  // m * 2^-9
+    else
+      v = std::ldexp(1.0f + m / 8.0f, exp_field - 7);
+    return sign ? -v : v;
+  }
+
+  void fp8_quantize_tile(const uint8_t* src, size_t block_bytes,
+                         uint8_t* out) const {
+    const size_t n = block_bytes / 2;
+    const uint16_t* in = reinterpret_cast<const uint16_t*>(src);
+    float amax = 0.0f;
+    for (size_t i = 0; i < n; ++i)
+      amax = std::max(amax, std::abs(bf16_to_f32(in[i])));
+    if (amax <= 0.0f) amax = 1.0f;
+    float scale = amax / 448.0f;
+    float inv = 448.0f / amax;
+    for (size_t i = 0; i < n; ++i)
+      out[i] = f32_to_fp8(bf16_to_f32(in[i]) * inv);
+    std::memcpy(out + n, &scale, 4);
+  }
+
+  void fp8_dequantize_tile(const uint8_t* in, size_t block_bytes,
+                           uint8_t* dst) const {
+    const size_t n = block_bytes / 2;
+    float scale;
+    std::memcpy(&scale, in + n, 4);
+    uint16_t* out = reinterpret_cast<uint16_t*>(dst);
+    for (size_t i = 0; i < n; ++i)
+      out[i] = f32_to_bf16(fp8_to_f32(in[i]) * scale);
   }
 
   EngineConfig cfg_;

@@ -177,3 +177,133 @@ extern "C" hipError_t kvc_launch_prefix_hash(
 }
 
 }  // namespace kvo
+
+namespace kvo {
+// ---- fp8 (OCP e4m3fn) serialization ----------------------------------------
+// Fused gather+quantize on store and dequantize+scatter on load: bf16 KV
+// pages are serialized as fp8 with one f32 scale per (block, layer) tile,
+// halving PCIe and storage bytes. Packed slab layout is self-contained per
+// tile record (so partial-span tail-seek loads work):
+//   [ tile fp8 payload (block_bytes/2) | f32 scale ] x tiles
+// One workgroup per tile (amax needs a tile-wide reduction): pass 1
+// computes amax over the tile's bf16 elements (wave + LDS reduce), pass 2
+// converts with the derived scale. gfx950 is OCP e4m3fn (not fnuz).
+
+namespace {
+
+__device__ __forceinline__ float bf16_to_f32(uint16_t u) {
+  uint32_t w = static_cast<uint32_t>(u) << 16;
+  return __uint_as_float(w);
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  uint32_t w = __float_as_uint(f);
+  // round-to-nearest-even
+  uint32_t rounding = 0x7fff + ((w >> 16) & 1);
+  return static_cast<uint16_t>((w + rounding) >> 16);
+}
+
+constexpr float kFp8Max = 448.0f;  // e4m3fn max normal
+
+__device__ __forceinline__ uint8_t f32_to_fp8_e4m3(float x, float inv_scale) {
+  float v = x * inv_scale;
+  // clamp to representable range; builtin handles rounding + saturation
+  uint32_t packed = __builtin_amdgcn_cvt_pk_fp8_f32(v, 0.0f, 0, false);
+  return static_cast<uint8_t>(packed & 0xff);
+}
+
+__device__ __forceinline__ float fp8_e4m3_to_f32(uint8_t b) {
+  return __builtin_amdgcn_cvt_f32_fp8(static_cast<uint32_t>(b), 0);
+}
+
+}  // namespace
+
+__global__ __launch_bounds__(256) void kvc_gather_fp8(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    uint64_t block_bytes, BlockList blocks, int num_tiles,
+    uint8_t* __restrict__ dst) {
+  (void)num_tiles;
+  const uint32_t tile = blockIdx.x;
+  const int l = tile % num_layers;
+  const int bi = tile / num_layers;
+  const uint64_t n_elems = block_bytes / 2;  // bf16 elements per tile
+  const uint64_t record = n_elems + 4;       // payload + f32 scale
+  const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
+      static_cast<const uint8_t*>(layer_ptrs[l]) +
+      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+  uint8_t* __restrict__ payload = dst + static_cast<uint64_t>(tile) * record;
+  float* __restrict__ scale_out =
+      reinterpret_cast<float*>(payload + n_elems);
+
+  // pass 1: tile amax
+  __shared__ float lds_max[4];
+  float amax = 0.0f;
+  for (uint64_t i = threadIdx.x; i < n_elems; i += blockDim.x)
+    amax = fmaxf(amax, fabsf(bf16_to_f32(src[i])));
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_down(amax, off, 64));
+  if ((threadIdx.x & 63) == 0) lds_max[threadIdx.x >> 6] = amax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = fmaxf(fmaxf(lds_max[0], lds_max[1]), fmaxf(lds_max[2], lds_max[3]));
+    lds_max[0] = m > 0.0f ? m : 1.0f;
+    *scale_out = lds_max[0] / kFp8Max;
+  }
+  __syncthreads();
+  const float inv_scale = kFp8Max / lds_max[0];
+
+  // pass 2: quantize
+  for (uint64_t i = threadIdx.x; i < n_elems; i += blockDim.x)
+    payload[i] = f32_to_fp8_e4m3(bf16_to_f32(src[i]), inv_scale);
+}
+
+__global__ __launch_bounds__(256) void kvc_scatter_fp8(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    uint64_t block_bytes, BlockList blocks, int num_tiles,
+    const uint8_t* __restrict__ src) {
+  (void)num_tiles;
+  const uint32_t tile = blockIdx.x;
+  const int l = tile % num_layers;
+  const int bi = tile / num_layers;
+  const uint64_t n_elems = block_bytes / 2;
+  const uint64_t record = n_elems + 4;
+  uint16_t* __restrict__ out = reinterpret_cast<uint16_t*>(
+      static_cast<uint8_t*>(const_cast<void*>(layer_ptrs[l])) +
+      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+  const uint8_t* __restrict__ payload =
+      src + static_cast<uint64_t>(tile) * record;
+  const float scale =
+      *reinterpret_cast<const float*>(payload + n_elems);
+  for (uint64_t i = threadIdx.x; i < n_elems; i += blockDim.x)
+    out[i] = f32_to_bf16(fp8_e4m3_to_f32(payload[i]) * scale);
+}
+
+extern "C" hipError_t kvc_launch_gather_fp8(
+    const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
+    int num_layers, uint64_t block_bytes, const int32_t* block_ids,
+    int num_blocks, uint8_t* dst, hipStream_t stream) {
+  BlockList bl;
+  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  int tiles = num_blocks * num_layers;
+  hipLaunchKernelGGL(kvc_gather_fp8, dim3(tiles), dim3(256), 0, stream,
+                     layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
+                     bl, tiles, dst);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t kvc_launch_scatter_fp8(
+    const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
+    int num_layers, uint64_t block_bytes, const int32_t* block_ids,
+    int num_blocks, const uint8_t* src, hipStream_t stream) {
+  BlockList bl;
+  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  int tiles = num_blocks * num_layers;
+  hipLaunchKernelGGL(kvc_scatter_fp8, dim3(tiles), dim3(256), 0, stream,
+                     layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
+                     bl, tiles, src);
+  return hipGetLastError();
+}
+
+}  // namespace kvo

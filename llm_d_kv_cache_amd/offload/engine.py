@@ -21,6 +21,7 @@ class OffloadEngineConfig:
     read_preferring_ratio: float = 0.75
     max_write_queued_seconds: float = 30.0
     copy_path: str = "staged"  # staged | zero_copy | host
+    serialize: str = "raw"     # raw | fp8_e4m3 (bf16 pages -> fp8 + scale)
     device: int = 0
     staging_budget_bytes: int = DEFAULT_STAGING_BUDGET_BYTES
 
@@ -71,10 +72,12 @@ class TorchOffloadEngine:
             # the actual payload per block may be smaller than the stride;
             # canonical layouts are dense so stride == payload
             native_groups.append((ptrs, strides, bb))
+            record_bytes = bb // 2 + 4 if config.serialize == "fp8_e4m3" else bb
             self.group_geometry.append(
                 {
                     "num_layers": len(g),
                     "block_bytes": bb,
+                    "record_bytes": record_bytes,
                     "num_device_blocks": g[0].shape[0],
                 }
             )
@@ -103,6 +106,7 @@ class TorchOffloadEngine:
             gpu_mode=gpu_mode,
             device=config.device,
             copy_path=config.copy_path,
+            serialize=config.serialize,
         )
         del stream
         # keep tensor refs: the native engine holds raw pointers

@@ -44,7 +44,8 @@ class _BaseHandler:
 
     def _file_bytes(self, group: int, n_blocks: int) -> int:
         geo = self.engine.group_geometry[group]
-        return n_blocks * geo["num_layers"] * geo["block_bytes"]
+        return n_blocks * geo["num_layers"] * geo.get("record_bytes",
+                                                      geo["block_bytes"])
 
     def get_finished(self) -> List[TransferResult]:
         out = []

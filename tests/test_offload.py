@@ -233,3 +233,72 @@ def test_cpu_tensors_require_host_mode():
     group = make_group()
     with pytest.raises(ValueError, match="copy_path='host'"):
         TorchOffloadEngine([group], OffloadEngineConfig(copy_path="staged"))
+
+
+def test_fp8_serialize_roundtrip(tmp_path):
+    """fp8 e4m3fn serialization: files are ~half size, values return within
+    quantization tolerance (per-tile scaling => rel err <= ~6%)."""
+    torch.manual_seed(3)
+    nl = 2
+    group = [
+        (torch.randn(NUM_BLOCKS, BLOCK_BYTES // 2) * 4).to(torch.bfloat16)
+        for _ in range(nl)
+    ]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host", serialize="fp8_e4m3"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="fp8"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    ids = list(range(8))
+    store.transfer_async([0xF8], {0: ids})
+    assert wait_finished(store)[0].success
+    raw_bytes = BLOCKS_PER_FILE * nl * BLOCK_BYTES
+    fp8_bytes = os.path.getsize(mapper.file_name(0xF8, 0))
+    assert fp8_bytes == BLOCKS_PER_FILE * nl * (BLOCK_BYTES // 2 + 4)
+    assert fp8_bytes < raw_bytes * 0.51
+
+    orig = [t.clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xF8], {0: ids})
+    assert wait_finished(load)[0].success
+    for t, o in zip(group, orig):
+        got = t[:8].float()
+        want = o[:8].float()
+        amax = want.abs().amax()
+        assert (got - want).abs().max() <= 0.07 * amax
+        # untouched blocks stay zero
+        assert (t[8:] == 0).all()
+
+
+def test_fp8_software_codec_exactness():
+    """Values exactly representable in e4m3 round-trip bit-exact through
+    the host-mode software codec (scale 1.0 when amax == 448)."""
+    import struct
+
+    nl = 1
+    vals = [448.0, 1.0, -2.0, 0.5, 0.0, 240.0, -448.0, 0.001953125]
+    data = torch.tensor(vals * (BLOCK_BYTES // 2 // len(vals)),
+                        dtype=torch.bfloat16).unsqueeze(0).repeat(NUM_BLOCKS, 1)
+    group = [data.clone()]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=1, gpu_blocks_per_file=4,
+                            copy_path="host", serialize="fp8_e4m3"),
+    )
+    import tempfile
+
+    d = tempfile.mkdtemp()
+    mapper = FileMapper(d, KVCacheLayoutConfig(model="fp8exact"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    load = StorageToGPUHandler(eng, mapper, [4])
+    store.transfer_async([1], {0: [0, 1, 2, 3]})
+    assert wait_finished(store)[0].success
+    orig = group[0][:4].clone()
+    group[0].zero_()
+    load.transfer_async([1], {0: [0, 1, 2, 3]})
+    assert wait_finished(load)[0].success
+    assert torch.equal(group[0][:4], orig)

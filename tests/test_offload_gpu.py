@@ -163,3 +163,41 @@ def test_prefix_hash_kernel_matches_cpu():
     for s in range(0, n_seq, 7):
         want = tp.tokens_to_block_keys(cpu_tokens[s * toks_per_seq:(s + 1) * toks_per_seq], "m")
         assert list(got[s * n_chunks:(s + 1) * n_chunks]) == want
+
+
+def test_gpu_fp8_serialize_roundtrip(tmp_path):
+    """CDNA4 fused gather+quantize / dequantize+scatter kernels: round-trip
+    within e4m3 tolerance; compare against plain PyTorch fp32 reference of
+    the same quantization."""
+    torch.manual_seed(5)
+    nl = 4
+    group = [
+        (torch.randn(NUM_BLOCKS, BLOCK_BYTES // 2, device="cuda") * 3)
+        .to(torch.bfloat16)
+        for _ in range(nl)
+    ]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BPF,
+                            serialize="fp8_e4m3"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gpu-fp8"))
+    store = GPUToStorageHandler(eng, mapper, [BPF])
+    load = StorageToGPUHandler(eng, mapper, [BPF])
+    ids = list(range(BPF))
+    store.transfer_async([0xF9], {0: ids})
+    assert wait_finished(store)[0].success
+    assert os.path.getsize(mapper.file_name(0xF9, 0)) == \
+        BPF * nl * (BLOCK_BYTES // 2 + 4)
+    orig = [t[:BPF].float().cpu() for t in group]
+    for t in group:
+        t.zero_()
+    torch.cuda.synchronize()
+    load.transfer_async([0xF9], {0: ids})
+    assert wait_finished(load)[0].success
+    torch.cuda.synchronize()
+    for t, want in zip(group, orig):
+        got = t[:BPF].float().cpu()
+        amax = want.abs().amax()
+        assert (got - want).abs().max() <= 0.07 * amax, \
+            f"fp8 error too large: {(got - want).abs().max()} vs amax {amax}"

@@ -48,6 +48,11 @@ namespace kvo {
 
 constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerFile
 
+// Intra-transfer pipeline granularity: the PCIe hop and the file I/O of one
+// transfer overlap at this chunk size (copy chunk i+1 rides the SDMA stream
+// while chunk i is written/was read).
+constexpr size_t kPipelineChunkBytes = 8ull << 20;
+
 enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };
 
 // Payload serialization: raw bytes, or fp8 e4m3fn quantization of bf16
@@ -393,22 +398,48 @@ class StorageOffloadEngine {
       double t1 = now_s();
       stats_inc([&](EngineStats& s) { s.t_gather_ms += (t1 - t0) * 1e3; });
       if (cfg_.copy_path == CopyPath::kStaged) {
-        // PCIe hop via the serialized SDMA mover (zero CU occupancy; one
-        // stream per direction saturates the wire — see pcie_mover.h).
+        // Chunked pipeline: D2H of chunk i+1 rides the SDMA mover while
+        // chunk i is written to the file (zero CU occupancy on the wire;
+        // one stream per direction saturates it — see pcie_mover.h).
         hipEvent_t gather_done;
         KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
         KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
+        double t_copy = 0, t_io = 0;
         try {
-          mover_->d2h(ctx.host_staging->host(), ctx.device_staging->ptr(), bytes,
-                      gather_done);
+          AtomicFileWriter writer(ft.path);
+          std::vector<std::future<void>> futs;
+          for (size_t off = 0; off < bytes; off += kPipelineChunkBytes) {
+            size_t n = std::min(kPipelineChunkBytes, bytes - off);
+            futs.push_back(mover_->d2h_async(
+                ctx.host_staging->host() + off, ctx.device_staging->ptr() + off,
+                n, off == 0 ? gather_done : nullptr));
+          }
+          size_t i = 0;
+          for (size_t off = 0; off < bytes; off += kPipelineChunkBytes, ++i) {
+            size_t n = std::min(kPipelineChunkBytes, bytes - off);
+            double w0 = now_s();
+            futs[i].get();
+            double w1 = now_s();
+            writer.write_at(off, ctx.host_staging->host() + off, n);
+            double w2 = now_s();
+            t_copy += w1 - w0;
+            t_io += w2 - w1;
+          }
+          writer.commit();
         } catch (...) {
           (void)hipEventDestroy(gather_done);
           throw;
         }
         (void)hipEventDestroy(gather_done);
-      } else {
-        KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
+        stats_inc([&](EngineStats& s) {
+          s.t_d2h_ms += t_copy * 1e3;
+          s.t_write_ms += t_io * 1e3;
+          s.files_written++;
+          s.bytes_stored += bytes;
+        });
+        return;
       }
+      KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       double t2 = now_s();
       stats_inc([&](EngineStats& s) { s.t_d2h_ms += (t2 - t1) * 1e3; });
     }
@@ -434,6 +465,50 @@ class StorageOffloadEngine {
     if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
       throw FileIoError("file " + ft.path + " does not cover requested span");
 
+    if (cfg_.copy_path == CopyPath::kStaged) {
+      // Chunked pipeline: pread of chunk i+1 overlaps the H2D of chunk i.
+      double t_read = 0, t_h2d = 0;
+      std::vector<std::future<void>> futs;
+      double r0 = now_s();
+      for (size_t off = 0; off < bytes; off += kPipelineChunkBytes) {
+        size_t n = std::min(kPipelineChunkBytes, bytes - off);
+        read_file_range(ft.path, offset + off, ctx.host_staging->host() + off, n);
+        double r1 = now_s();
+        t_read += r1 - r0;
+        futs.push_back(mover_->h2d_async(ctx.device_staging->ptr() + off,
+                                         ctx.host_staging->host() + off, n));
+        r0 = now_s();
+      }
+      double h0 = now_s();
+      for (auto& f : futs) f.get();
+      t_h2d = now_s() - h0;
+      touch_atime(ft.path);
+      hipError_t err =
+          cfg_.serialize == Serialize::kFp8E4M3
+              ? kvc_launch_scatter_fp8(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, ctx.device_staging->ptr(),
+                    ctx.stream)
+              : kvc_launch_scatter(
+                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+                    dev_layer_strides_[ft.group], nl, g.block_bytes,
+                    ft.block_ids.data(), nb, ctx.device_staging->ptr(),
+                    ctx.stream);
+      if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+      double s0 = now_s();
+      KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
+      double s1 = now_s();
+      stats_inc([&](EngineStats& s) {
+        s.t_read_ms += t_read * 1e3;
+        s.t_h2d_ms += t_h2d * 1e3;
+        s.t_scatter_ms += (s1 - s0) * 1e3;
+        s.files_read++;
+        s.bytes_loaded += bytes;
+      });
+      return;
+    }
+
     double t0 = now_s();
     read_file_range(ft.path, offset, ctx.host_staging->host(), bytes);
     touch_atime(ft.path);
@@ -444,11 +519,6 @@ class StorageOffloadEngine {
       scatter_host(g, ft.block_ids, ctx.host_staging->host());
     } else {
       const uint8_t* kernel_src = ctx.host_staging->device();
-      if (cfg_.copy_path == CopyPath::kStaged) {
-        // blocking SDMA hop; the scatter launched after return is ordered
-        mover_->h2d(ctx.device_staging->ptr(), ctx.host_staging->host(), bytes);
-        kernel_src = ctx.device_staging->ptr();
-      }
       double t2 = now_s();
       stats_inc([&](EngineStats& s) { s.t_h2d_ms += (t2 - t1) * 1e3; });
       hipError_t err =

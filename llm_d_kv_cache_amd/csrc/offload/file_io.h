@@ -87,6 +87,60 @@ inline void write_file_atomic(const std::string& path, const uint8_t* data,
   }
 }
 
+// Incremental atomic writer: open <path>.<rand>.tmp, pwrite chunks, then
+// commit() renames into place (abort unlinks). Enables overlapping file
+// writes with the PCIe pipeline inside one transfer.
+class AtomicFileWriter {
+ public:
+  explicit AtomicFileWriter(const std::string& path) : path_(path) {
+    static thread_local std::mt19937_64 rng{std::random_device{}()};
+    tmp_ = path + "." + std::to_string(rng()) + ".tmp";
+    fd_ = ::open(tmp_.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0644);
+    if (fd_ < 0 && errno == ENOENT) {
+      make_parent_dirs(tmp_);
+      fd_ = ::open(tmp_.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0644);
+    }
+    if (fd_ < 0)
+      throw FileIoError("open " + tmp_ + ": " + std::strerror(errno));
+  }
+
+  ~AtomicFileWriter() {
+    if (fd_ >= 0) {
+      ::close(fd_);
+      ::unlink(tmp_.c_str());
+    }
+  }
+  AtomicFileWriter(const AtomicFileWriter&) = delete;
+
+  void write_at(uint64_t offset, const uint8_t* data, size_t len) {
+    size_t off = 0;
+    while (off < len) {
+      ssize_t w = ::pwrite(fd_, data + off, len - off,
+                           static_cast<off_t>(offset + off));
+      if (w < 0) {
+        if (errno == EINTR) continue;
+        throw FileIoError("pwrite " + tmp_ + ": " + std::strerror(errno));
+      }
+      off += static_cast<size_t>(w);
+    }
+  }
+
+  void commit() {
+    ::close(fd_);
+    fd_ = -1;
+    if (::rename(tmp_.c_str(), path_.c_str()) != 0) {
+      int err = errno;
+      ::unlink(tmp_.c_str());
+      throw FileIoError("rename " + path_ + ": " + std::strerror(err));
+    }
+  }
+
+ private:
+  std::string path_;
+  std::string tmp_;
+  int fd_ = -1;
+};
+
 // Read [offset, offset+len) into buf; the file may be a head-partial
 // (shorter than the nominal full span) — callers validate coverage first.
 inline void read_file_range(const std::string& path, uint64_t offset,

@@ -48,16 +48,24 @@ class PcieMover {
     }
   }
 
-  // Blocking: device -> pinned host. pre_event (optional) is waited on the
-  // mover stream before the copy (producer-kernel ordering without a host
-  // sync on the worker).
-  void d2h(void* dst, const void* src, size_t n, hipEvent_t pre_event) {
-    submit(lanes_[0], dst, src, n, pre_event);
+  // Async submit: the returned future resolves (or rethrows) when the copy
+  // has completed on the wire. Same-direction requests execute FIFO on one
+  // stream, so chunked callers can overlap file I/O of chunk i with the
+  // copy of chunk i+1.
+  std::future<void> d2h_async(void* dst, const void* src, size_t n,
+                              hipEvent_t pre_event = nullptr) {
+    return submit(lanes_[0], dst, src, n, pre_event);
+  }
+  std::future<void> h2d_async(void* dst, const void* src, size_t n) {
+    return submit(lanes_[1], dst, src, n, nullptr);
   }
 
-  // Blocking: pinned host -> device.
+  // Blocking conveniences.
+  void d2h(void* dst, const void* src, size_t n, hipEvent_t pre_event) {
+    d2h_async(dst, src, n, pre_event).get();
+  }
   void h2d(void* dst, const void* src, size_t n) {
-    submit(lanes_[1], dst, src, n, nullptr);
+    h2d_async(dst, src, n).get();
   }
 
  private:
@@ -67,50 +75,49 @@ class PcieMover {
     size_t n;
     hipEvent_t pre;
     std::promise<void> done;
-    std::exception_ptr error;
   };
   struct Lane {
     std::mutex mu;
     std::condition_variable cv;
-    std::deque<Req*> q;
+    std::deque<std::unique_ptr<Req>> q;
     bool stopping = false;
     std::thread thread;
   };
 
-  void submit(Lane& lane, void* dst, const void* src, size_t n, hipEvent_t pre) {
-    Req req;
-    req.dst = dst;
-    req.src = src;
-    req.n = n;
-    req.pre = pre;
-    auto fut = req.done.get_future();
+  std::future<void> submit(Lane& lane, void* dst, const void* src, size_t n,
+                           hipEvent_t pre) {
+    auto req = std::make_unique<Req>();
+    req->dst = dst;
+    req->src = src;
+    req->n = n;
+    req->pre = pre;
+    auto fut = req->done.get_future();
     {
       std::lock_guard<std::mutex> g(lane.mu);
-      lane.q.push_back(&req);
+      lane.q.push_back(std::move(req));
     }
     lane.cv.notify_one();
-    fut.get();
-    if (req.error) std::rethrow_exception(req.error);
+    return fut;
   }
 
   void lane_loop(Lane& lane, hipStream_t stream, hipMemcpyKind kind) {
     for (;;) {
-      Req* req = nullptr;
+      std::unique_ptr<Req> req;
       {
         std::unique_lock<std::mutex> g(lane.mu);
         lane.cv.wait(g, [&] { return lane.stopping || !lane.q.empty(); });
         if (lane.q.empty()) return;
-        req = lane.q.front();
+        req = std::move(lane.q.front());
         lane.q.pop_front();
       }
       try {
         if (req->pre) KVO_HIP_CHECK(hipStreamWaitEvent(stream, req->pre, 0));
         KVO_HIP_CHECK(hipMemcpyAsync(req->dst, req->src, req->n, kind, stream));
         KVO_HIP_CHECK(hipStreamSynchronize(stream));
+        req->done.set_value();
       } catch (...) {
-        req->error = std::current_exception();
+        req->done.set_exception(std::current_exception());
       }
-      req->done.set_value();
     }
   }
 

@@ -50,8 +50,15 @@ constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerF
 
 // Intra-transfer pipeline granularity: the PCIe hop and the file I/O of one
 // transfer overlap at this chunk size (copy chunk i+1 rides the SDMA stream
-// while chunk i is written/was read).
+// while chunk i is written/was read). Transfers below 3 chunks run
+// single-shot — the overlap cannot pay for its future/sync overhead there
+// (measured: fp8 16 MiB files regressed 68.6 -> 61 GB/s with forced
+// 2-chunk splits).
 constexpr size_t kPipelineChunkBytes = 8ull << 20;
+
+inline size_t pipeline_chunk(size_t bytes) {
+  return bytes < 3 * kPipelineChunkBytes ? bytes : kPipelineChunkBytes;
+}
 
 enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };
 
@@ -406,17 +413,18 @@ class StorageOffloadEngine {
         KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
         double t_copy = 0, t_io = 0;
         try {
+          const size_t chunk = pipeline_chunk(bytes);
           AtomicFileWriter writer(ft.path);
           std::vector<std::future<void>> futs;
-          for (size_t off = 0; off < bytes; off += kPipelineChunkBytes) {
-            size_t n = std::min(kPipelineChunkBytes, bytes - off);
+          for (size_t off = 0; off < bytes; off += chunk) {
+            size_t n = std::min(chunk, bytes - off);
             futs.push_back(mover_->d2h_async(
                 ctx.host_staging->host() + off, ctx.device_staging->ptr() + off,
                 n, off == 0 ? gather_done : nullptr));
           }
           size_t i = 0;
-          for (size_t off = 0; off < bytes; off += kPipelineChunkBytes, ++i) {
-            size_t n = std::min(kPipelineChunkBytes, bytes - off);
+          for (size_t off = 0; off < bytes; off += chunk, ++i) {
+            size_t n = std::min(chunk, bytes - off);
             double w0 = now_s();
             futs[i].get();
             double w1 = now_s();
@@ -467,12 +475,14 @@ class StorageOffloadEngine {
 
     if (cfg_.copy_path == CopyPath::kStaged) {
       // Chunked pipeline: pread of chunk i+1 overlaps the H2D of chunk i.
+      const size_t chunk = pipeline_chunk(bytes);
       double t_read = 0, t_h2d = 0;
       std::vector<std::future<void>> futs;
+      FileReader reader(ft.path);
       double r0 = now_s();
-      for (size_t off = 0; off < bytes; off += kPipelineChunkBytes) {
-        size_t n = std::min(kPipelineChunkBytes, bytes - off);
-        read_file_range(ft.path, offset + off, ctx.host_staging->host() + off, n);
+      for (size_t off = 0; off < bytes; off += chunk) {
+        size_t n = std::min(chunk, bytes - off);
+        reader.read_at(offset + off, ctx.host_staging->host() + off, n);
         double r1 = now_s();
         t_read += r1 - r0;
         futs.push_back(mover_->h2d_async(ctx.device_staging->ptr() + off,

@@ -141,6 +141,38 @@ class AtomicFileWriter {
   int fd_ = -1;
 };
 
+// RAII reader for chunked loads: one open per transfer.
+class FileReader {
+ public:
+  explicit FileReader(const std::string& path) : path_(path) {
+    fd_ = ::open(path.c_str(), O_RDONLY);
+    if (fd_ < 0)
+      throw FileIoError("open " + path + ": " + std::strerror(errno));
+  }
+  ~FileReader() {
+    if (fd_ >= 0) ::close(fd_);
+  }
+  FileReader(const FileReader&) = delete;
+
+  void read_at(uint64_t offset, uint8_t* buf, size_t len) {
+    size_t got = 0;
+    while (got < len) {
+      ssize_t r = ::pread(fd_, buf + got, len - got,
+                          static_cast<off_t>(offset + got));
+      if (r < 0) {
+        if (errno == EINTR) continue;
+        throw FileIoError("pread " + path_ + ": " + std::strerror(errno));
+      }
+      if (r == 0) throw FileIoError("short read from " + path_);
+      got += static_cast<size_t>(r);
+    }
+  }
+
+ private:
+  std::string path_;
+  int fd_ = -1;
+};
+
 // Read [offset, offset+len) into buf; the file may be a head-partial
 // (shorter than the nominal full span) — callers validate coverage first.
 inline void read_file_range(const std::string& path, uint64_t offset,

@@ -94,3 +94,54 @@ def test_unknown_tier_defaults_to_weight_1():
     tokens = list(range(16))
     add_prefix(ix, "pod-x", tokens, tier="exotic-tier")
     assert ix.score_tokens(tokens, MODEL)["pod-x"] == 1.0
+
+
+# ---- multimodal read side ---------------------------------------------------
+
+def test_compute_block_extra_features():
+    from llm_d_kv_cache_amd.core.extra_keys import (
+        PlaceholderRange,
+        compute_block_extra_features,
+    )
+
+    # 64 tokens, 16-token blocks; image A covers tokens [10, 30),
+    # image B covers [30, 34)
+    out = compute_block_extra_features(
+        64, 16, ["img-A", "img-B"],
+        [PlaceholderRange(10, 20), PlaceholderRange(30, 4)],
+    )
+    assert out == [["img-A"], ["img-A", "img-B"], ["img-B"], None]
+    # pure text
+    assert compute_block_extra_features(64, 16, [], []) is None
+    # partial tail dropped: only full blocks get entries
+    out = compute_block_extra_features(40, 16, ["x"], [PlaceholderRange(38, 2)])
+    assert out == [None, None]
+
+
+def test_mm_features_read_write_agree():
+    """The read-side features reproduce the write-side tainted hashes: a
+    scorer query with mm features matches blocks stored with engine
+    extra_keys."""
+    from llm_d_kv_cache_amd.core.extra_keys import (
+        PlaceholderRange,
+        compute_block_extra_features,
+    )
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_stored_payload,
+        encode_batch,
+    )
+    from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+
+    ix = make_indexer()
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    tokens = list(range(48))
+    extra = compute_block_extra_features(
+        48, 16, ["mm-7"], [PlaceholderRange(20, 10)])
+    assert extra == [None, ["mm-7"], None]
+    pool.process("kv@pod-mm@m", 0, encode_batch([
+        block_stored_payload([1, 2, 3], None, tokens, 16, extra_keys=extra)
+    ]))
+    # untainted query matches only the untainted first block; the chain
+    # diverges at the tainted block
+    assert ix.score_tokens(tokens, MODEL) == {"pod-mm": 1.0}
+    assert ix.score_tokens(tokens, MODEL, extra_features=extra) == {"pod-mm": 3.0}

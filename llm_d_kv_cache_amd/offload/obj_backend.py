@@ -1,0 +1,250 @@
+"""Object-store offload backend (S3-compatible HTTP).
+
+Capability parity with the reference llmd_nixl OBJ backend
+(llmd_nixl/nixl_offload.py, obj_backend.py, nixl_lookup.py): the same
+handler protocol as the filesystem engine, but chunks live in an object
+store. The data path is GPU pages -> CDNA4 gather into a packed slab ->
+pinned staging -> HTTP PUT (and GET -> staging -> scatter on load); lookup
+is a HEAD per chunk. Works against any S3-compatible endpoint (MinIO,
+Ceph RGW, S3 with anonymous/bucket-policy auth; request signing is left
+to a fronting proxy, the common in-cluster deployment).
+"""
+from __future__ import annotations
+
+import concurrent.futures
+import threading
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Sequence, Tuple
+from urllib import request as urlrequest
+from urllib.error import HTTPError, URLError
+
+
+@dataclass
+class ObjStorageConfig:
+    endpoint: str = "http://127.0.0.1:9000"
+    bucket: str = "kvcache"
+    io_threads: int = 8
+    timeout_s: float = 30.0
+    serialize: str = "raw"  # raw | fp8_e4m3 (via the native codec kernels)
+
+
+class ObjClient:
+    """Minimal S3-compatible object client (PUT/GET/HEAD/DELETE)."""
+
+    def __init__(self, cfg: ObjStorageConfig):
+        self.cfg = cfg
+
+    def _url(self, key: str) -> str:
+        return f"{self.cfg.endpoint}/{self.cfg.bucket}/{key}"
+
+    def put(self, key: str, data: bytes) -> None:
+        req = urlrequest.Request(self._url(key), data=data, method="PUT")
+        req.add_header("Content-Type", "application/octet-stream")
+        with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
+            if resp.status not in (200, 201, 204):
+                raise IOError(f"PUT {key}: HTTP {resp.status}")
+
+    def get(self, key: str, offset: int = 0, length: Optional[int] = None) -> bytes:
+        req = urlrequest.Request(self._url(key), method="GET")
+        if offset or length is not None:
+            end = "" if length is None else str(offset + length - 1)
+            req.add_header("Range", f"bytes={offset}-{end}")
+        with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
+            return resp.read()
+
+    def head(self, key: str) -> bool:
+        req = urlrequest.Request(self._url(key), method="HEAD")
+        try:
+            with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
+                return resp.status == 200
+        except HTTPError:
+            return False
+        except URLError:
+            return False
+
+    def delete(self, key: str) -> None:
+        req = urlrequest.Request(self._url(key), method="DELETE")
+        try:
+            urlrequest.urlopen(req, timeout=self.cfg.timeout_s)
+        except HTTPError:
+            pass
+
+
+class ObjStorageEngine:
+    """Same surface as TorchOffloadEngine for the handler layer, but
+    transfers target object keys instead of file paths."""
+
+    def __init__(self, groups: Sequence[Sequence], config: ObjStorageConfig):
+        import torch
+
+        from .. import ensure_offload_native
+
+        ko = ensure_offload_native()
+        self._torch = torch
+        self.config = config
+        self.gpu_mode = groups[0][0].is_cuda
+        native_groups = []
+        self.group_geometry: List[dict] = []
+        for g in groups:
+            ptrs = [t.data_ptr() for t in g]
+            strides = [t.stride(0) * t.element_size() for t in g]
+            bb = strides[0]
+            native_groups.append((ptrs, strides, bb))
+            record = bb // 2 + 4 if config.serialize == "fp8_e4m3" else bb
+            self.group_geometry.append(
+                {"num_layers": len(g), "block_bytes": bb, "record_bytes": record,
+                 "num_device_blocks": g[0].shape[0]}
+            )
+        if config.serialize == "fp8_e4m3":
+            raise NotImplementedError(
+                "fp8 object offload rides the filesystem engine for now")
+        device = groups[0][0].device.index or 0 if self.gpu_mode else 0
+        self._copier = ko.BlockCopier(native_groups, self.gpu_mode, device)
+        self._tensors = [list(g) for g in groups]
+        self.client = ObjClient(config)
+        self._pool = concurrent.futures.ThreadPoolExecutor(
+            max_workers=config.io_threads)
+        self._jobs: Dict[int, List[concurrent.futures.Future]] = {}
+        self._next_job = 1
+        self._mu = threading.Lock()
+        self._fin_lock = threading.Lock()
+        self._fin_buffer: Dict[int, Tuple[bool, bool]] = {}
+
+    # ---- engine surface -----------------------------------------------------
+
+    def _slab(self, group: int, n_blocks: int):
+        nb = self._copier.packed_bytes(group, n_blocks)
+        t = self._torch.empty(nb, dtype=self._torch.uint8,
+                              pin_memory=self.gpu_mode)
+        return t
+
+    def _store_one(self, group: int, key: str, ids: List[int]) -> None:
+        if self.client.head(key):
+            return  # dedupe
+        if self.gpu_mode:
+            dev = self._torch.empty(self._copier.packed_bytes(group, len(ids)),
+                                    dtype=self._torch.uint8, device="cuda")
+            stream = self._torch.cuda.Stream()
+            with self._torch.cuda.stream(stream):
+                self._copier.gather(group, ids, dev.data_ptr(),
+                                    stream.cuda_stream)
+                host = self._slab(group, len(ids))
+                host.copy_(dev, non_blocking=True)
+            stream.synchronize()
+        else:
+            host = self._slab(group, len(ids))
+            self._copier.gather(group, ids, host.data_ptr(), 0)
+        self.client.put(key, host.numpy().tobytes())
+
+    def _load_one(self, group: int, key: str, ids: List[int],
+                  slot_offset: int) -> None:
+        geo = self.group_geometry[group]
+        rec = geo["record_bytes"] * geo["num_layers"]
+        data = self.client.get(key, offset=slot_offset * rec,
+                               length=len(ids) * rec)
+        if len(data) != len(ids) * rec:
+            raise IOError(f"short object read for {key}")
+        host = self._torch.frombuffer(bytearray(data), dtype=self._torch.uint8)
+        if self.gpu_mode:
+            dev = host.cuda()
+            self._torch.cuda.synchronize()
+            self._copier.scatter(group, ids, dev.data_ptr(),
+                                 self._torch.cuda.current_stream().cuda_stream)
+            self._torch.cuda.synchronize()
+        else:
+            self._copier.scatter(group, ids, host.data_ptr(), 0)
+
+    def async_store(self, files, stream: int = None) -> int:
+        if self.gpu_mode:
+            self._torch.cuda.synchronize()  # KV-ready fence (coarse)
+        with self._mu:
+            job = self._next_job
+            self._next_job += 1
+        futs = [self._pool.submit(self._store_one, g, key, ids)
+                for (g, key, ids, _off) in files]
+        with self._mu:
+            self._jobs[job] = futs
+        return job
+
+    def async_load(self, files) -> int:
+        with self._mu:
+            job = self._next_job
+            self._next_job += 1
+        futs = [self._pool.submit(self._load_one, g, key, ids, off)
+                for (g, key, ids, off) in files]
+        with self._mu:
+            self._jobs[job] = futs
+        return job
+
+    def get_finished(self):
+        out = []
+        with self._mu:
+            for job, futs in list(self._jobs.items()):
+                if all(f.done() for f in futs):
+                    success = all(f.exception() is None for f in futs)
+                    out.append((job, success, False))
+                    del self._jobs[job]
+        return out
+
+    def poll_finished(self, job_ids):
+        with self._fin_lock:
+            for jid, success, dropped in self.get_finished():
+                self._fin_buffer[jid] = (success, dropped)
+            out = []
+            for jid in list(job_ids):
+                if jid in self._fin_buffer:
+                    success, dropped = self._fin_buffer.pop(jid)
+                    out.append((jid, success, dropped))
+            return out
+
+    def wait_job(self, job_id: int) -> bool:
+        with self._mu:
+            futs = self._jobs.pop(job_id, [])
+        ok = True
+        for f in futs:
+            if not f.cancel():
+                try:
+                    f.result(timeout=60)
+                except Exception:
+                    ok = False
+        return ok
+
+    def current_stream_handle(self) -> int:
+        return 0
+
+    def close(self):
+        self._pool.shutdown(wait=True)
+
+
+class ObjKeyMapper:
+    """FileMapper-compatible naming for object keys (no leading root)."""
+
+    def __init__(self, mapper):
+        self._mapper = mapper
+        self.run_dir = mapper.run_dir
+
+    def file_name(self, chunk_hash: int, group: int = 0) -> str:
+        import os
+
+        return os.path.relpath(self._mapper.file_name(chunk_hash, group),
+                               self._mapper.root)
+
+
+class ObjOffloadManager:
+    """Scheduler-side lookup over HEAD (reference NixlLookup parity)."""
+
+    def __init__(self, key_mapper: ObjKeyMapper, client: ObjClient,
+                 num_groups: int = 1):
+        self.mapper = key_mapper
+        self.client = client
+        self.num_groups = num_groups
+
+    def lookup(self, chunk_hashes: Sequence[int]) -> int:
+        hits = 0
+        for h in chunk_hashes:
+            if all(self.client.head(self.mapper.file_name(h, g))
+                   for g in range(self.num_groups)):
+                hits += 1
+            else:
+                break
+        return hits

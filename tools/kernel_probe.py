@@ -1,0 +1,65 @@
+#!/usr/bin/env python3
+"""Kernel microbench: gather/scatter HBM bandwidth on Llama-3-8B KV geometry.
+
+Times the CDNA4 gather (paged KV pages -> contiguous slab) and scatter
+kernels in isolation (HBM->HBM, both directions r+w) — the compute-side
+ceiling of the offload path, independent of PCIe/file I/O. Run under
+rocprofv3 (--stats or --pmc) for counter evidence.
+"""
+import sys
+import time
+
+import torch
+
+from llm_d_kv_cache_amd import ensure_offload_native
+
+NUM_LAYERS = 32
+BLOCK_BYTES = 64 * 1024
+NUM_BLOCKS = 2048
+BPF = 16
+ITERS = int(sys.argv[1]) if len(sys.argv) > 1 else 50
+
+
+def main():
+    ko = ensure_offload_native()
+    assert torch.cuda.is_available()
+    group = [
+        torch.randint(0, 255, (NUM_BLOCKS, BLOCK_BYTES), dtype=torch.uint8,
+                      device="cuda")
+        for _ in range(NUM_LAYERS)
+    ]
+    copier = ko.BlockCopier(
+        [([t.data_ptr() for t in group], [t.stride(0) for t in group],
+          BLOCK_BYTES)],
+        gpu_mode=True,
+    )
+    slab_bytes = copier.packed_bytes(0, BPF)
+    slab = torch.empty(slab_bytes, dtype=torch.uint8, device="cuda")
+    stream = torch.cuda.current_stream().cuda_stream
+    ids = list(range(0, BPF * 8, 8))  # strided pages: realistic gather
+
+    # warmup
+    copier.gather(0, ids, slab.data_ptr(), stream)
+    copier.scatter(0, ids, slab.data_ptr(), stream)
+    torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(ITERS):
+        copier.gather(0, ids, slab.data_ptr(), stream)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    moved = slab_bytes * ITERS
+    print(f"gather:  {moved / dt / 1e9:.1f} GB/s payload "
+          f"({2 * moved / dt / 1e9:.1f} GB/s r+w), {dt / ITERS * 1e6:.0f} us/launch")
+
+    t0 = time.perf_counter()
+    for _ in range(ITERS):
+        copier.scatter(0, ids, slab.data_ptr(), stream)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"scatter: {moved / dt / 1e9:.1f} GB/s payload "
+          f"({2 * moved / dt / 1e9:.1f} GB/s r+w), {dt / ITERS * 1e6:.0f} us/launch")
+
+
+if __name__ == "__main__":
+    main()

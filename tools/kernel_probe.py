@@ -6,8 +6,11 @@ kernels in isolation (HBM->HBM, both directions r+w) — the compute-side
 ceiling of the offload path, independent of PCIe/file I/O. Run under
 rocprofv3 (--stats or --pmc) for counter evidence.
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -44,6 +44,7 @@ class FakeRedis:
             self._threads.append(t)
 
     def _serve(self, conn):
+        conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         buf = b""
 
         def read_line():

@@ -184,7 +184,9 @@ def main():
         backend = "nccl" if torch.cuda.is_available() else "gloo"
         dist.init_process_group(backend=backend)
 
-    if not torch.cuda.is_available():
+    gpu = torch.cuda.is_available()
+    cpu_full = bool(os.environ.get("KVC_BENCH_CPU_FULL"))
+    if not gpu and not cpu_full:
         log("no GPU: control-plane phases only; offload GB/s unmeasured")
         aux = bench_control_plane()
         if rank == 0:
@@ -198,7 +200,8 @@ def main():
             }))
         return
 
-    torch.cuda.set_device(local_rank)
+    if gpu:
+        torch.cuda.set_device(local_rank)
     from llm_d_kv_cache_amd.offload import (
         FileMapper,
         GPUToStorageHandler,
@@ -219,14 +222,15 @@ def main():
         f"{args.device_blocks * NUM_LAYERS * BLOCK_BYTES / 1e9:.1f} GB")
     group = [
         torch.randint(0, 255, (args.device_blocks, BLOCK_BYTES), dtype=torch.uint8,
-                      device="cuda")
+                      device="cuda" if gpu else "cpu")
         for _ in range(NUM_LAYERS)
     ]
     eng = TorchOffloadEngine(
         [group],
         OffloadEngineConfig(io_threads=args.io_threads,
                             gpu_blocks_per_file=BLOCKS_PER_FILE,
-                            copy_path=args.copy_path, serialize=args.serialize,
+                            copy_path=args.copy_path if gpu else "host",
+                            serialize=args.serialize,
                             device=local_rank),
     )
     mapper = FileMapper(rank_root, KVCacheLayoutConfig(
@@ -299,10 +303,16 @@ def main():
     def barrier():
         if dist is not None:
             dist.barrier()
-        torch.cuda.synchronize()
+        if gpu:
+            torch.cuda.synchronize()
 
     for w in range(args.warmup):
-        run_step(-1 - w)  # negative ids: cleaned below
+        run_step(-1 - w)
+    # drop warmup generations so steady state holds exactly two generations
+    for w in range(args.warmup):
+        base = (-1 - w) * FILES_PER_STEP + 1
+        for h in range(base, base + FILES_PER_STEP):
+            del_q.put(mapper.file_name(h, 0))
     log("warmup done")
 
     barrier()
@@ -320,7 +330,8 @@ def main():
 
     # max over ranks
     if dist is not None:
-        t = torch.tensor([elapsed], device="cuda" if dist.get_backend() == "nccl" else "cpu")
+        t = torch.tensor([elapsed],
+                         device="cuda" if dist.get_backend() == "nccl" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
@@ -332,7 +343,12 @@ def main():
     # each rank pulls its neighbor's cached chunks over RCCL send/recv.
     peer_aux = None
     if world > 1 and dist is not None:
-        peer_aux = bench_peer_phase(dist, torch, group, rank, world, local_rank)
+        try:
+            peer_aux = bench_peer_phase(dist, torch, group, rank, world,
+                                        local_rank)
+        except Exception as e:  # aux phase must never sink the headline run
+            log(f"peer phase failed: {e}")
+            peer_aux = {"ok": False, "error": str(e)}
 
     aux = bench_control_plane() if rank == 0 else None
     if aux is not None and peer_aux is not None:
@@ -344,7 +360,7 @@ def main():
             "metric": "kv_block_offload_GBps",
             "value": round(total_gbps, 2),
             "unit": "GB/s",
-            "n_gpus": world,
+            "n_gpus": world if gpu else 0,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1e3, 2),

@@ -308,11 +308,14 @@ def main():
 
     for w in range(args.warmup):
         run_step(-1 - w)
-    # drop warmup generations so steady state holds exactly two generations
+    # drop warmup generations BEFORE the timed region starts (pre-existing
+    # state, not steady-state work; one generation per step remains inside)
     for w in range(args.warmup):
         base = (-1 - w) * FILES_PER_STEP + 1
         for h in range(base, base + FILES_PER_STEP):
             del_q.put(mapper.file_name(h, 0))
+    while not del_q.empty():
+        time.sleep(0.005)
     log("warmup done")
 
     barrier()

@@ -50,14 +50,14 @@ constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerF
 
 // Intra-transfer pipeline granularity: the PCIe hop and the file I/O of one
 // transfer overlap at this chunk size (copy chunk i+1 rides the SDMA stream
-// while chunk i is written/was read). Transfers below 3 chunks run
-// single-shot — the overlap cannot pay for its future/sync overhead there
-// (measured: fp8 16 MiB files regressed 68.6 -> 61 GB/s with forced
-// 2-chunk splits).
-constexpr size_t kPipelineChunkBytes = 8ull << 20;
+// while chunk i is written/was read). Transfers at or below 64 MiB run
+// single-shot: with 16 workers the pipeline across tasks already overlaps
+// file I/O with the wire, and measured intra-task chunking at 32 MiB files
+// only added memory-traffic contention (read-side accumulated time 3x).
+constexpr size_t kPipelineChunkBytes = 16ull << 20;
 
 inline size_t pipeline_chunk(size_t bytes) {
-  return bytes < 3 * kPipelineChunkBytes ? bytes : kPipelineChunkBytes;
+  return bytes <= 4 * kPipelineChunkBytes ? bytes : kPipelineChunkBytes;
 }
 
 enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };

@@ -169,6 +169,8 @@ def main():
                     choices=["raw", "fp8_e4m3"])
     ap.add_argument("--io-threads", type=int, default=16)
     ap.add_argument("--device-blocks", type=int, default=2048)
+    ap.add_argument("--host-cache-gb", type=float, default=6.0,
+                    help="pinned-DRAM cache tier size (0 disables)")
     args = ap.parse_args()
 
     import torch
@@ -231,6 +233,7 @@ def main():
                             gpu_blocks_per_file=BLOCKS_PER_FILE,
                             copy_path=args.copy_path if gpu else "host",
                             serialize=args.serialize,
+                            host_cache_bytes=int(args.host_cache_gb * 1024**3),
                             device=local_rank),
     )
     mapper = FileMapper(rank_root, KVCacheLayoutConfig(
@@ -384,6 +387,9 @@ def main():
                 "copy_path": args.copy_path,
                 "serialize": args.serialize,
                 "io_threads": args.io_threads,
+                "host_cache_gb": args.host_cache_gb,
+                "host_cache_hits": stats.host_cache_hits,
+                "host_cache_stores": stats.host_cache_stores,
                 "root": root,
                 "files_written": stats.files_written,
                 "engine_avg_write_ms": round(stats.avg_write_seconds * 1e3, 3),

@@ -28,6 +28,7 @@
 
 #include "common.h"
 #include "file_io.h"
+#include "host_cache.h"
 #include "pcie_mover.h"
 #include "thread_pool.h"
 
@@ -82,6 +83,9 @@ struct EngineConfig {
   int device = 0;
   CopyPath copy_path = CopyPath::kStaged;
   Serialize serialize = Serialize::kRaw;
+  // Pinned host-DRAM cache tier (0 = disabled): write-through on store,
+  // cache-hit loads skip the filesystem (host_cache.h).
+  size_t host_cache_bytes = 0;
 };
 
 struct FileTransfer {
@@ -108,6 +112,8 @@ struct EngineStats {
   uint64_t files_read = 0;
   uint64_t writes_dropped = 0;
   uint64_t tasks_cancelled = 0;
+  uint64_t host_cache_hits = 0;
+  uint64_t host_cache_stores = 0;
   uint64_t errors = 0;
   double avg_write_seconds = 0;
   uint64_t bytes_stored = 0;
@@ -175,6 +181,10 @@ class StorageOffloadEngine {
         cfg_.read_preferring_ratio, mapped);
     if (cfg_.gpu_mode && cfg_.copy_path == CopyPath::kStaged)
       mover_ = std::make_unique<PcieMover>(true, cfg_.device);
+    if (cfg_.host_cache_bytes > 0 && cfg_.copy_path != CopyPath::kZeroCopy &&
+        max_file_bytes > 0)
+      cache_ = std::make_unique<HostPinnedCache>(cfg_.host_cache_bytes,
+                                                 max_file_bytes, cfg_.gpu_mode);
   }
 
   ~StorageOffloadEngine() {
@@ -381,8 +391,26 @@ class StorageOffloadEngine {
     const int nl = static_cast<int>(g.layer_ptrs.size());
     const size_t bytes = file_bytes(g, nb);
 
+    HostPinnedCache::Slot* slot =
+        cache_ ? cache_->acquire(ft.path, bytes) : nullptr;
+    uint8_t* host_buf = slot ? slot->buf->host() : ctx.host_staging->host();
+    struct SlotGuard {
+      HostPinnedCache* c;
+      HostPinnedCache::Slot* s;
+      bool ok = false;
+      ~SlotGuard() {
+        if (!c || !s) return;
+        if (ok) {
+          c->publish(s);
+          c->release(s);
+        } else {
+          c->abandon(s);
+        }
+      }
+    } guard{cache_.get(), slot};
+
     if (cfg_.copy_path == CopyPath::kHostMemcpy) {
-      gather_host(g, ft.block_ids, ctx.host_staging->host());
+      gather_host(g, ft.block_ids, host_buf);
     } else {
       // KV-ready fence: the gather must observe the serving engine's
       // completed KV writes for these blocks.
@@ -419,7 +447,7 @@ class StorageOffloadEngine {
           for (size_t off = 0; off < bytes; off += chunk) {
             size_t n = std::min(chunk, bytes - off);
             futs.push_back(mover_->d2h_async(
-                ctx.host_staging->host() + off, ctx.device_staging->ptr() + off,
+                host_buf + off, ctx.device_staging->ptr() + off,
                 n, off == 0 ? gather_done : nullptr));
           }
           size_t i = 0;
@@ -428,7 +456,7 @@ class StorageOffloadEngine {
             double w0 = now_s();
             futs[i].get();
             double w1 = now_s();
-            writer.write_at(off, ctx.host_staging->host() + off, n);
+            writer.write_at(off, host_buf + off, n);
             double w2 = now_s();
             t_copy += w1 - w0;
             t_io += w2 - w1;
@@ -439,11 +467,13 @@ class StorageOffloadEngine {
           throw;
         }
         (void)hipEventDestroy(gather_done);
+        guard.ok = true;
         stats_inc([&](EngineStats& s) {
           s.t_d2h_ms += t_copy * 1e3;
           s.t_write_ms += t_io * 1e3;
           s.files_written++;
           s.bytes_stored += bytes;
+          if (slot) s.host_cache_stores++;
         });
         return;
       }
@@ -452,12 +482,14 @@ class StorageOffloadEngine {
       stats_inc([&](EngineStats& s) { s.t_d2h_ms += (t2 - t1) * 1e3; });
     }
     double tw = now_s();
-    write_file_atomic(ft.path, ctx.host_staging->host(), bytes);
+    write_file_atomic(ft.path, host_buf, bytes);
     double tw2 = now_s();
+    guard.ok = true;
     stats_inc([&](EngineStats& s) {
       s.files_written++;
       s.bytes_stored += bytes;
       s.t_write_ms += (tw2 - tw) * 1e3;
+      if (slot) s.host_cache_stores++;
     });
   }
 
@@ -469,25 +501,69 @@ class StorageOffloadEngine {
     const uint64_t offset =
         static_cast<uint64_t>(ft.slot_offset) * nl * tile_record_bytes(g);
 
-    int64_t fsz = file_size(ft.path);
-    if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
-      throw FileIoError("file " + ft.path + " does not cover requested span");
+    // (file coverage is validated on the miss paths — a pinned-DRAM cache
+    // hit must serve even when the file was already evicted from disk)
+    auto check_file_span = [&]() {
+      int64_t fsz = file_size(ft.path);
+      if (fsz < 0 || static_cast<uint64_t>(fsz) < offset + bytes)
+        throw FileIoError("file " + ft.path + " does not cover requested span");
+    };
 
     if (cfg_.copy_path == CopyPath::kStaged) {
-      // Chunked pipeline: pread of chunk i+1 overlaps the H2D of chunk i.
+      // Pinned-DRAM tier: a cache hit skips the filesystem read entirely —
+      // the H2D streams straight from the resident slab.
+      HostPinnedCache::Slot* hit = cache_ ? cache_->lookup(ft.path) : nullptr;
+      if (hit && hit->bytes_used < offset + bytes) {
+        cache_->release(hit);
+        hit = nullptr;
+      }
+      HostPinnedCache::Slot* fill = nullptr;
+      uint8_t* host_buf = ctx.host_staging->host();
+      if (hit) {
+        host_buf = hit->buf->host() + offset;
+      } else if (cache_ && offset == 0) {
+        // populate the tier on a miss (full-span loads only)
+        fill = cache_->acquire(ft.path, bytes);
+        if (fill) host_buf = fill->buf->host();
+      }
+      struct LoadGuard {
+        HostPinnedCache* c;
+        HostPinnedCache::Slot* hit;
+        HostPinnedCache::Slot* fill;
+        bool ok = false;
+        ~LoadGuard() {
+          if (!c) return;
+          if (hit) c->release(hit);
+          if (fill) {
+            if (ok) {
+              c->publish(fill);
+              c->release(fill);
+            } else {
+              c->abandon(fill);
+            }
+          }
+        }
+      } lguard{cache_.get(), hit, fill};
+
       const size_t chunk = pipeline_chunk(bytes);
       double t_read = 0, t_h2d = 0;
       std::vector<std::future<void>> futs;
-      FileReader reader(ft.path);
-      double r0 = now_s();
-      for (size_t off = 0; off < bytes; off += chunk) {
-        size_t n = std::min(chunk, bytes - off);
-        reader.read_at(offset + off, ctx.host_staging->host() + off, n);
-        double r1 = now_s();
-        t_read += r1 - r0;
-        futs.push_back(mover_->h2d_async(ctx.device_staging->ptr() + off,
-                                         ctx.host_staging->host() + off, n));
-        r0 = now_s();
+      if (hit) {
+        futs.push_back(mover_->h2d_async(ctx.device_staging->ptr(), host_buf,
+                                         bytes));
+      } else {
+        check_file_span();
+        FileReader reader(ft.path);
+        double r0 = now_s();
+        for (size_t off = 0; off < bytes; off += chunk) {
+          size_t n = std::min(chunk, bytes - off);
+          reader.read_at(offset + off, host_buf + off, n);
+          double r1 = now_s();
+          t_read += r1 - r0;
+          futs.push_back(mover_->h2d_async(ctx.device_staging->ptr() + off,
+                                           host_buf + off, n));
+          r0 = now_s();
+        }
       }
       double h0 = now_s();
       for (auto& f : futs) f.get();
@@ -509,16 +585,34 @@ class StorageOffloadEngine {
       double s0 = now_s();
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       double s1 = now_s();
+      lguard.ok = true;
       stats_inc([&](EngineStats& s) {
         s.t_read_ms += t_read * 1e3;
         s.t_h2d_ms += t_h2d * 1e3;
         s.t_scatter_ms += (s1 - s0) * 1e3;
         s.files_read++;
         s.bytes_loaded += bytes;
+        if (hit) s.host_cache_hits++;
       });
       return;
     }
 
+    if (cfg_.copy_path == CopyPath::kHostMemcpy && cache_) {
+      HostPinnedCache::Slot* hit = cache_->lookup(ft.path);
+      if (hit && hit->bytes_used >= offset + bytes) {
+        scatter_host(g, ft.block_ids, hit->buf->host() + offset);
+        touch_atime(ft.path);
+        cache_->release(hit);
+        stats_inc([&](EngineStats& s) {
+          s.files_read++;
+          s.bytes_loaded += bytes;
+          s.host_cache_hits++;
+        });
+        return;
+      }
+      if (hit) cache_->release(hit);
+    }
+    check_file_span();
     double t0 = now_s();
     read_file_range(ft.path, offset, ctx.host_staging->host(), bytes);
     touch_atime(ft.path);
@@ -675,6 +769,7 @@ class StorageOffloadEngine {
   std::vector<uint64_t*> dev_layer_strides_;
   std::unique_ptr<IoThreadPool> pool_;
   std::unique_ptr<PcieMover> mover_;
+  std::unique_ptr<HostPinnedCache> cache_;
 
   std::mutex jobs_mu_;
   std::condition_variable done_cv_;

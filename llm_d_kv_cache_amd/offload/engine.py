@@ -22,6 +22,7 @@ class OffloadEngineConfig:
     max_write_queued_seconds: float = 30.0
     copy_path: str = "staged"  # staged | zero_copy | host
     serialize: str = "raw"     # raw | fp8_e4m3 (bf16 pages -> fp8 + scale)
+    host_cache_bytes: int = 0  # pinned-DRAM cache tier (0 = off)
     device: int = 0
     staging_budget_bytes: int = DEFAULT_STAGING_BUDGET_BYTES
 
@@ -107,6 +108,7 @@ class TorchOffloadEngine:
             device=config.device,
             copy_path=config.copy_path,
             serialize=config.serialize,
+            host_cache_bytes=config.host_cache_bytes,
         )
         del stream
         # keep tensor refs: the native engine holds raw pointers

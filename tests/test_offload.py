@@ -302,3 +302,87 @@ def test_fp8_software_codec_exactness():
     load.transfer_async([1], {0: [0, 1, 2, 3]})
     assert wait_finished(load)[0].success
     assert torch.equal(group[0][:4], orig)
+
+
+def test_host_cache_tier(tmp_path):
+    """Pinned-DRAM tier: write-through stores; loads hit the cache even
+    after the file is gone (proof the filesystem read was skipped)."""
+    group = make_group(seed=9)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host",
+                            host_cache_bytes=8 * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="hostcache"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    ids = list(range(8))
+    store.transfer_async([0xCC], {0: ids})
+    assert wait_finished(store)[0].success
+    assert eng.stats().host_cache_stores == 1
+    path = mapper.file_name(0xCC, 0)
+    assert os.path.exists(path)  # write-through
+    os.unlink(path)
+
+    orig = [t[:8].clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xCC], {0: ids})
+    assert wait_finished(load)[0].success
+    assert eng.stats().host_cache_hits == 1
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:8], o)
+
+
+def test_host_cache_partial_offset_hit(tmp_path):
+    group = make_group(seed=10)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host",
+                            host_cache_bytes=4 * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="hc2"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    store.transfer_async([0xCD], {0: list(range(8))})
+    assert wait_finished(store)[0].success
+    os.unlink(mapper.file_name(0xCD, 0))
+    orig = [t[:8].clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xCD], {0: [5, 6, 7]}, skip_leading_blocks=5)
+    assert wait_finished(load)[0].success
+    assert eng.stats().host_cache_hits == 1
+    for t, o in zip(group, orig):
+        assert torch.equal(t[5:8], o[5:8])
+        assert (t[:5] == 0).all()
+
+
+def test_host_cache_eviction_falls_back_to_file(tmp_path):
+    # cache holds exactly 2 slots; store 4 files; oldest 2 must come from disk
+    slot = BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
+    group = make_group(seed=11)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=1, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host", host_cache_bytes=2 * slot),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="hc3"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    for i in range(4):
+        store.transfer_async([0xD0 + i], {0: list(range(i * 8, i * 8 + 8))})
+    assert len(wait_finished(store, n=4)) == 4
+    orig = [t.clone() for t in group]
+    for t in group:
+        t.zero_()
+    for i in range(4):
+        load.transfer_async([0xD0 + i], {0: list(range(i * 8, i * 8 + 8))})
+    assert len(wait_finished(load, n=4)) == 4
+    s = eng.stats()
+    assert s.host_cache_hits >= 2  # the resident generation
+    assert s.files_read + s.host_cache_hits >= 4
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:32], o[:32])

@@ -140,3 +140,20 @@ def test_connector_rejects_bad_geometry(tmp_path):
                 group_block_tokens=(16,),
             ),
         )
+
+
+def test_epp_scorer_demo_runs():
+    out = subprocess.run(
+        [sys.executable, str(REPO / "examples" / "epp_scorer.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "ok" in out.stdout
+
+
+def test_valkey_example_runs():
+    out = subprocess.run(
+        [sys.executable, str(REPO / "examples" / "valkey_example.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr

@@ -201,3 +201,66 @@ def test_gpu_fp8_serialize_roundtrip(tmp_path):
         amax = want.abs().amax()
         assert (got - want).abs().max() <= 0.07 * amax, \
             f"fp8 error too large: {(got - want).abs().max()} vs amax {amax}"
+
+
+def test_gpu_host_cache_tier(tmp_path):
+    """DRAM tier on the GPU path: loads hit the pinned cache (file
+    deleted to prove the filesystem was skipped), bit-exact."""
+    group = make_group(num_layers=4)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BPF,
+                            host_cache_bytes=4 * BPF * 4 * BLOCK_BYTES),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gpu-hc"))
+    store = GPUToStorageHandler(eng, mapper, [BPF])
+    load = StorageToGPUHandler(eng, mapper, [BPF])
+    ids = list(range(BPF))
+    store.transfer_async([0xCA], {0: ids})
+    assert wait_finished(store)[0].success
+    assert eng.stats().host_cache_stores == 1
+    os.unlink(mapper.file_name(0xCA, 0))
+    golden = [t[:BPF].cpu().clone() for t in group]
+    for t in group:
+        t.zero_()
+    torch.cuda.synchronize()
+    load.transfer_async([0xCA], {0: ids})
+    assert wait_finished(load)[0].success
+    torch.cuda.synchronize()
+    assert eng.stats().host_cache_hits == 1
+    for t, g in zip(group, golden):
+        assert torch.equal(t[:BPF].cpu(), g)
+
+
+def test_gpu_reads_overtake_write_storm(tmp_path):
+    """QoS on the real GPU path: HIGH-priority loads jump a deep store
+    backlog."""
+    import time
+
+    group = make_group(num_layers=2)
+    eng = TorchOffloadEngine(
+        [group], OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=4),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gpu-qos"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    load = StorageToGPUHandler(eng, mapper, [4])
+    for i in range(4):
+        store.transfer_async([i], {0: list(range(4))})
+    assert len(wait_finished(store, n=4)) == 4
+    for i in range(150):
+        store.transfer_async([1000 + i], {0: list(range(4))})
+    t0 = time.time()
+    for i in range(4):
+        load.transfer_async([i], {0: list(range(4))})
+    got = 0
+    while got < 4 and time.time() < t0 + 20:
+        got += len(load.get_finished())
+        time.sleep(0.002)
+    latency = time.time() - t0
+    assert got == 4
+    assert eng.native.pending_writes > 0, "storm drained before reads measured"
+    assert latency < 5.0
+    # drain
+    deadline = time.time() + 60
+    while eng.native.pending_writes > 0 and time.time() < deadline:
+        time.sleep(0.05)

@@ -14,7 +14,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-from llm_d_kv_cache_amd import ensure_offload_native
+from llm_d_kv_cache_amd import ensure_offload_native  # noqa: E402
 
 NUM_LAYERS = 32
 BLOCK_BYTES = 64 * 1024
@@ -66,3 +66,31 @@ def main():
 
 if __name__ == "__main__":
     main()
+    prefix_hash_probe()
+
+
+def prefix_hash_probe(n_seq=4096, toks=2048, iters=20):
+    """Batched prefix-hash kernel throughput (sequences hashed /s)."""
+    ko = ensure_offload_native()
+    tokens = torch.randint(0, 128000, (n_seq * toks,), dtype=torch.int32,
+                           device="cuda")
+    seq_off = torch.arange(0, (n_seq + 1) * toks, toks, dtype=torch.int64,
+                           device="cuda")
+    seeds = torch.full((n_seq,), 12345, dtype=torch.int64, device="cuda")
+    nchunks = toks // 16
+    key_off = torch.arange(0, (n_seq + 1) * nchunks, nchunks,
+                           dtype=torch.int64, device="cuda")
+    keys = torch.zeros(n_seq * nchunks, dtype=torch.int64, device="cuda")
+    s = torch.cuda.current_stream().cuda_stream
+    ko.prefix_hash(tokens.data_ptr(), seq_off.data_ptr(), seeds.data_ptr(),
+                   keys.data_ptr(), key_off.data_ptr(), 16, n_seq, s)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ko.prefix_hash(tokens.data_ptr(), seq_off.data_ptr(), seeds.data_ptr(),
+                       keys.data_ptr(), key_off.data_ptr(), 16, n_seq, s)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"prefix_hash: {n_seq * iters / dt:,.0f} seq/s "
+          f"({n_seq * toks * iters / dt / 1e9:.2f} Gtok/s, "
+          f"{n_seq}x{toks} tokens per launch)")

@@ -64,9 +64,6 @@ def main():
           f"({2 * moved / dt / 1e9:.1f} GB/s r+w), {dt / ITERS * 1e6:.0f} us/launch")
 
 
-if __name__ == "__main__":
-    main()
-    prefix_hash_probe()
 
 
 def prefix_hash_probe(n_seq=4096, toks=2048, iters=20):
@@ -94,3 +91,8 @@ def prefix_hash_probe(n_seq=4096, toks=2048, iters=20):
     print(f"prefix_hash: {n_seq * iters / dt:,.0f} seq/s "
           f"({n_seq * toks * iters / dt / 1e9:.2f} Gtok/s, "
           f"{n_seq}x{toks} tokens per launch)")
+
+
+if __name__ == "__main__":
+    main()
+    prefix_hash_probe()

@@ -31,6 +31,7 @@ PYBIND11_MODULE(_kvoffload, m) {
       .def_readonly("tasks_cancelled", &EngineStats::tasks_cancelled)
       .def_readonly("host_cache_hits", &EngineStats::host_cache_hits)
       .def_readonly("host_cache_stores", &EngineStats::host_cache_stores)
+      .def_readonly("writeback_flushes", &EngineStats::writeback_flushes)
       .def_readonly("errors", &EngineStats::errors)
       .def_readonly("avg_write_seconds", &EngineStats::avg_write_seconds)
       .def_readonly("bytes_stored", &EngineStats::bytes_stored)
@@ -49,7 +50,8 @@ PYBIND11_MODULE(_kvoffload, m) {
                        int io_threads, int gpu_blocks_per_file,
                        double read_preferring_ratio, double max_write_queued_seconds,
                        bool gpu_mode, int device, const std::string& copy_path,
-                       const std::string& serialize, size_t host_cache_bytes) {
+                       const std::string& serialize, size_t host_cache_bytes,
+                       const std::string& write_policy) {
              EngineConfig cfg;
              cfg.io_threads = io_threads;
              cfg.gpu_blocks_per_file = gpu_blocks_per_file;
@@ -73,6 +75,12 @@ PYBIND11_MODULE(_kvoffload, m) {
              else
                throw std::invalid_argument("serialize must be raw|fp8_e4m3");
              cfg.host_cache_bytes = host_cache_bytes;
+             if (write_policy == "through")
+               cfg.write_policy = WritePolicy::kThrough;
+             else if (write_policy == "back")
+               cfg.write_policy = WritePolicy::kBack;
+             else
+               throw std::invalid_argument("write_policy must be through|back");
              std::vector<GroupDesc> gs;
              for (auto& [ptrs, strides, block_bytes] : groups) {
                GroupDesc g;
@@ -89,7 +97,8 @@ PYBIND11_MODULE(_kvoffload, m) {
            py::arg("read_preferring_ratio") = 0.75,
            py::arg("max_write_queued_seconds") = 30.0, py::arg("gpu_mode") = false,
            py::arg("device") = 0, py::arg("copy_path") = "staged",
-           py::arg("serialize") = "raw", py::arg("host_cache_bytes") = 0)
+           py::arg("serialize") = "raw", py::arg("host_cache_bytes") = 0,
+           py::arg("write_policy") = "through")
       .def(
           "async_store",
           [](StorageOffloadEngine& e,

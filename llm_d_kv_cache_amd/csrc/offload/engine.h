@@ -68,6 +68,14 @@ enum class CopyPath { kStaged, kZeroCopy, kHostMemcpy };
 // tile appended to each tile record).
 enum class Serialize { kRaw, kFp8E4M3 };
 
+// Store completion semantics: kThrough = the file rename happened (crash
+// durable); kBack = the slab is resident in the pinned-DRAM tier and the
+// file flush runs asynchronously (store latency ~= gather + D2H; loses
+// only not-yet-flushed cache entries on crash — acceptable for a cache).
+// kBack needs host_cache_bytes > 0; transfers that cannot get a slot fall
+// back to write-through.
+enum class WritePolicy { kThrough, kBack };
+
 struct GroupDesc {
   std::vector<void*> layer_ptrs;        // per-layer block-0 base address
   std::vector<uint64_t> layer_strides;  // bytes between consecutive blocks
@@ -86,6 +94,7 @@ struct EngineConfig {
   // Pinned host-DRAM cache tier (0 = disabled): write-through on store,
   // cache-hit loads skip the filesystem (host_cache.h).
   size_t host_cache_bytes = 0;
+  WritePolicy write_policy = WritePolicy::kThrough;
 };
 
 struct FileTransfer {
@@ -114,6 +123,7 @@ struct EngineStats {
   uint64_t tasks_cancelled = 0;
   uint64_t host_cache_hits = 0;
   uint64_t host_cache_stores = 0;
+  uint64_t writeback_flushes = 0;
   uint64_t errors = 0;
   double avg_write_seconds = 0;
   uint64_t bytes_stored = 0;
@@ -411,6 +421,32 @@ class StorageOffloadEngine {
 
     if (cfg_.copy_path == CopyPath::kHostMemcpy) {
       gather_host(g, ft.block_ids, host_buf);
+      if (cfg_.write_policy == WritePolicy::kBack && slot != nullptr) {
+        guard.ok = true;
+        cache_->addref(slot);
+        auto* cache = cache_.get();
+        std::string path = ft.path;
+        pool_->enqueue(Priority::kNormal,
+                       [this, cache, slot, path, bytes](WorkerCtx&) {
+                         try {
+                           write_file_atomic(path, slot->buf->host(), bytes);
+                           stats_inc([](EngineStats& s) {
+                             s.files_written++;
+                             s.writeback_flushes++;
+                           });
+                         } catch (const std::exception& e) {
+                           KVO_LOG_ERROR("writeback flush %s failed: %s",
+                                         path.c_str(), e.what());
+                           stats_inc([](EngineStats& s) { s.errors++; });
+                         }
+                         cache->release(slot);
+                       });
+        stats_inc([&](EngineStats& s) {
+          s.bytes_stored += bytes;
+          s.host_cache_stores++;
+        });
+        return;
+      }
     } else {
       // KV-ready fence: the gather must observe the serving engine's
       // completed KV writes for these blocks.
@@ -432,6 +468,50 @@ class StorageOffloadEngine {
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       double t1 = now_s();
       stats_inc([&](EngineStats& s) { s.t_gather_ms += (t1 - t0) * 1e3; });
+      if (cfg_.copy_path == CopyPath::kStaged &&
+          cfg_.write_policy == WritePolicy::kBack && slot != nullptr) {
+        // Write-back: the store completes once the slab is resident in
+        // the DRAM tier; the file flush runs as a background task.
+        hipEvent_t gather_done;
+        KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
+        KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
+        double c0 = now_s();
+        try {
+          mover_->d2h(host_buf, ctx.device_staging->ptr(), bytes, gather_done);
+        } catch (...) {
+          (void)hipEventDestroy(gather_done);
+          throw;
+        }
+        (void)hipEventDestroy(gather_done);
+        double c1 = now_s();
+        guard.ok = true;  // publish + release the store's reference
+        cache_->addref(slot);
+        auto* cache = cache_.get();
+        std::string path = ft.path;
+        pool_->enqueue(Priority::kNormal,
+                       [this, cache, slot, path, bytes](WorkerCtx&) {
+                         double w0 = now_s();
+                         try {
+                           write_file_atomic(path, slot->buf->host(), bytes);
+                           stats_inc([&](EngineStats& s) {
+                             s.files_written++;
+                             s.writeback_flushes++;
+                             s.t_write_ms += (now_s() - w0) * 1e3;
+                           });
+                         } catch (const std::exception& e) {
+                           KVO_LOG_ERROR("writeback flush %s failed: %s",
+                                         path.c_str(), e.what());
+                           stats_inc([](EngineStats& s) { s.errors++; });
+                         }
+                         cache->release(slot);
+                       });
+        stats_inc([&](EngineStats& s) {
+          s.t_d2h_ms += (c1 - c0) * 1e3;
+          s.bytes_stored += bytes;
+          s.host_cache_stores++;
+        });
+        return;
+      }
       if (cfg_.copy_path == CopyPath::kStaged) {
         // Chunked pipeline: D2H of chunk i+1 rides the SDMA mover while
         // chunk i is written to the file (zero CU occupancy on the wire;

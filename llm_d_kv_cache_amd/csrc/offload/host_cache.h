@@ -108,6 +108,12 @@ class HostPinnedCache {
     s->refs--;
   }
 
+  // Extra reference on a held slot (e.g. handing it to an async flush).
+  void addref(Slot* s) {
+    std::lock_guard<std::mutex> g(mu_);
+    s->refs++;
+  }
+
   uint64_t hit_count() const { return hits_; }
 
  private:

@@ -23,6 +23,7 @@ class OffloadEngineConfig:
     copy_path: str = "staged"  # staged | zero_copy | host
     serialize: str = "raw"     # raw | fp8_e4m3 (bf16 pages -> fp8 + scale)
     host_cache_bytes: int = 0  # pinned-DRAM cache tier (0 = off)
+    write_policy: str = "through"  # through | back (flush async via DRAM tier)
     device: int = 0
     staging_budget_bytes: int = DEFAULT_STAGING_BUDGET_BYTES
 
@@ -109,6 +110,7 @@ class TorchOffloadEngine:
             copy_path=config.copy_path,
             serialize=config.serialize,
             host_cache_bytes=config.host_cache_bytes,
+            write_policy=config.write_policy,
         )
         del stream
         # keep tensor refs: the native engine holds raw pointers

@@ -386,3 +386,57 @@ def test_host_cache_eviction_falls_back_to_file(tmp_path):
     assert s.files_read + s.host_cache_hits >= 4
     for t, o in zip(group, orig):
         assert torch.equal(t[:32], o[:32])
+
+
+def test_writeback_policy(tmp_path):
+    """write_policy=back: the store completes once the slab is in the
+    DRAM tier; the file appears asynchronously; loads hit the tier."""
+    group = make_group(seed=12)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host", write_policy="back",
+                            host_cache_bytes=8 * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="wb"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    ids = list(range(8))
+    store.transfer_async([0xE1], {0: ids})
+    assert wait_finished(store)[0].success
+    # flush lands eventually
+    path = mapper.file_name(0xE1, 0)
+    deadline = time.time() + 10
+    while not os.path.exists(path) and time.time() < deadline:
+        time.sleep(0.01)
+    assert os.path.exists(path)
+    deadline = time.time() + 10
+    while eng.stats().writeback_flushes < 1 and time.time() < deadline:
+        time.sleep(0.01)
+    assert eng.stats().writeback_flushes == 1
+
+    orig = [t[:8].clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xE1], {0: ids})
+    assert wait_finished(load)[0].success
+    assert eng.stats().host_cache_hits == 1
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:8], o)
+
+
+def test_writeback_without_cache_falls_through(tmp_path):
+    """No DRAM tier -> write-back degrades to write-through gracefully."""
+    group = make_group(seed=13)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=1, gpu_blocks_per_file=4,
+                            copy_path="host", write_policy="back",
+                            host_cache_bytes=0),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="wb2"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    store.transfer_async([0xE2], {0: [0, 1, 2, 3]})
+    assert wait_finished(store)[0].success
+    assert os.path.exists(mapper.file_name(0xE2, 0))
+    assert eng.stats().writeback_flushes == 0

@@ -171,6 +171,8 @@ def main():
     ap.add_argument("--device-blocks", type=int, default=2048)
     ap.add_argument("--host-cache-gb", type=float, default=6.0,
                     help="pinned-DRAM cache tier size (0 disables)")
+    ap.add_argument("--write-policy", type=str, default="through",
+                    choices=["through", "back"])
     args = ap.parse_args()
 
     import torch
@@ -234,6 +236,7 @@ def main():
                             copy_path=args.copy_path if gpu else "host",
                             serialize=args.serialize,
                             host_cache_bytes=int(args.host_cache_gb * 1024**3),
+                            write_policy=args.write_policy,
                             device=local_rank),
     )
     mapper = FileMapper(rank_root, KVCacheLayoutConfig(
@@ -388,6 +391,8 @@ def main():
                 "serialize": args.serialize,
                 "io_threads": args.io_threads,
                 "host_cache_gb": args.host_cache_gb,
+                "write_policy": args.write_policy,
+                "writeback_flushes": stats.writeback_flushes,
                 "host_cache_hits": stats.host_cache_hits,
                 "host_cache_stores": stats.host_cache_stores,
                 "root": root,

@@ -173,6 +173,10 @@ def main():
                     help="pinned-DRAM cache tier size (0 disables)")
     ap.add_argument("--write-policy", type=str, default="through",
                     choices=["through", "back"])
+    ap.add_argument("--overlap", action="store_true",
+                    help="full-duplex steady state: stores of generation N "
+                         "run concurrently with loads of generation N-1 "
+                         "(both PCIe directions busy; disjoint GPU pages)")
     args = ap.parse_args()
 
     import torch
@@ -272,7 +276,40 @@ def main():
     for t in deleters:
         t.start()
 
+    def run_step_overlap(step_id):
+        # store gen step_id from pages [0, B); load gen step_id-1 into
+        # pages [B, 2B) — disjoint, so both directions run concurrently
+        base = step_id * FILES_PER_STEP + 1
+        hashes = list(range(base, base + FILES_PER_STEP))
+        prev_base = (step_id - 1) * FILES_PER_STEP + 1
+        prev_hashes = list(range(prev_base, prev_base + FILES_PER_STEP))
+        do_load = step_id - 1 >= -args.warmup
+        n_store = n_load = 0
+        p0 = time.perf_counter()
+        for i in range(0, FILES_PER_STEP, 8):
+            ids = list(range(i * BLOCKS_PER_FILE, (i + 8) * BLOCKS_PER_FILE))
+            store.transfer_async(hashes[i:i + 8], {0: ids})
+            n_store += 1
+            if do_load:
+                lids = [b + blocks_per_step for b in ids]
+                load.transfer_async(prev_hashes[i:i + 8], {0: lids})
+                n_load += 1
+        ds = dl = 0
+        while ds < n_store or dl < n_load:
+            ds += len(store.get_finished())
+            dl += len(load.get_finished())
+            if ds < n_store or dl < n_load:
+                time.sleep(0.0002)
+        p1 = time.perf_counter()
+        old = (step_id - 2) * FILES_PER_STEP + 1
+        if step_id - 2 >= -args.warmup:
+            for h in range(old, old + FILES_PER_STEP):
+                del_q.put(mapper.file_name(h, 0))
+        phase_wall["store"] += p1 - p0
+
     def run_step(step_id):
+        if args.overlap:
+            return run_step_overlap(step_id)
         base = step_id * FILES_PER_STEP + 1
         hashes = list(range(base, base + FILES_PER_STEP))
         ids = list(range(blocks_per_step))
@@ -312,12 +349,16 @@ def main():
         if gpu:
             torch.cuda.synchronize()
 
-    for w in range(args.warmup):
-        run_step(-1 - w)
+    for sid in range(-args.warmup, 0):  # ascending: overlap loads gen-1
+        run_step(sid)
     # drop warmup generations BEFORE the timed region starts (pre-existing
-    # state, not steady-state work; one generation per step remains inside)
-    for w in range(args.warmup):
-        base = (-1 - w) * FILES_PER_STEP + 1
+    # state, not steady-state work; one generation per step remains inside).
+    # Overlap mode keeps the newest warmup generation: timed step 0 loads it.
+    keep = {-1} if args.overlap else set()
+    for gen in range(-args.warmup, 0):
+        if gen in keep:
+            continue
+        base = gen * FILES_PER_STEP + 1
         for h in range(base, base + FILES_PER_STEP):
             del_q.put(mapper.file_name(h, 0))
     while not del_q.empty():
@@ -345,6 +386,8 @@ def main():
         elapsed = float(t.item())
 
     moved_bytes = 2 * step_bytes * args.steps  # store + load, per rank
+    # (overlap mode: every timed step stores gen N and loads gen N-1 — the
+    # same total volume)
     total_gbps = moved_bytes * world / elapsed / 1e9
     stats = eng.stats()
 
@@ -392,6 +435,7 @@ def main():
                 "io_threads": args.io_threads,
                 "host_cache_gb": args.host_cache_gb,
                 "write_policy": args.write_policy,
+                "overlap": args.overlap,
                 "writeback_flushes": stats.writeback_flushes,
                 "host_cache_hits": stats.host_cache_hits,
                 "host_cache_stores": stats.host_cache_stores,

@@ -173,10 +173,14 @@ def main():
                     help="pinned-DRAM cache tier size (0 disables)")
     ap.add_argument("--write-policy", type=str, default="through",
                     choices=["through", "back"])
-    ap.add_argument("--overlap", action="store_true",
-                    help="full-duplex steady state: stores of generation N "
-                         "run concurrently with loads of generation N-1 "
-                         "(both PCIe directions busy; disjoint GPU pages)")
+    ap.add_argument("--overlap", dest="overlap", action="store_true",
+                    default=True,
+                    help="full-duplex steady state (default): stores of "
+                         "generation N run concurrently with loads of "
+                         "generation N-1 (both PCIe directions busy; "
+                         "disjoint GPU pages)")
+    ap.add_argument("--no-overlap", dest="overlap", action="store_false",
+                    help="sequential store-then-load phases")
     args = ap.parse_args()
 
     import torch
@@ -223,6 +227,12 @@ def main():
     rank_root = os.path.join(root, f"rank{rank}")
     shutil.rmtree(rank_root, ignore_errors=True)
     os.makedirs(rank_root, exist_ok=True)
+
+    need = FILES_PER_STEP * BLOCKS_PER_FILE * (2 if args.overlap else 1)
+    if args.device_blocks < need:
+        log(f"raising --device-blocks {args.device_blocks} -> {need} "
+            f"(overlap mode uses disjoint store/load pages)")
+        args.device_blocks = need
 
     # Llama-3-8B canonical KV: one group, 32 layers, (num_blocks, 128 KiB)
     log(f"allocating KV cache: {args.device_blocks} blocks x {NUM_LAYERS} layers "

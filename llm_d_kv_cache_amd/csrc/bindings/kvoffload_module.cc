@@ -45,7 +45,8 @@ PYBIND11_MODULE(_kvoffload, m) {
 
   py::class_<StorageOffloadEngine>(m, "StorageOffloadEngine")
       .def(py::init([](std::vector<std::tuple<std::vector<uintptr_t>,
-                                              std::vector<uint64_t>, uint64_t>>
+                                              std::vector<uint64_t>, uint64_t,
+                                              int64_t>>
                            groups,
                        int io_threads, int gpu_blocks_per_file,
                        double read_preferring_ratio, double max_write_queued_seconds,
@@ -82,11 +83,12 @@ PYBIND11_MODULE(_kvoffload, m) {
              else
                throw std::invalid_argument("write_policy must be through|back");
              std::vector<GroupDesc> gs;
-             for (auto& [ptrs, strides, block_bytes] : groups) {
+             for (auto& [ptrs, strides, block_bytes, num_blocks] : groups) {
                GroupDesc g;
                for (auto p : ptrs) g.layer_ptrs.push_back(reinterpret_cast<void*>(p));
                g.layer_strides = strides;
                g.block_bytes = block_bytes;
+               g.num_blocks = num_blocks;
                gs.push_back(std::move(g));
              }
              py::gil_scoped_release rel;

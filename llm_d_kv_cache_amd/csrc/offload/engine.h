@@ -80,6 +80,7 @@ struct GroupDesc {
   std::vector<void*> layer_ptrs;        // per-layer block-0 base address
   std::vector<uint64_t> layer_strides;  // bytes between consecutive blocks
   uint64_t block_bytes = 0;             // bytes per (block, layer)
+  int64_t num_blocks = 0;               // device pages per layer (0 = unchecked)
 };
 
 struct EngineConfig {
@@ -355,6 +356,14 @@ class StorageOffloadEngine {
     if (ft.block_ids.empty() ||
         ft.block_ids.size() > static_cast<size_t>(cfg_.gpu_blocks_per_file))
       throw std::invalid_argument("bad block count for file transfer");
+    int64_t nb = groups_[ft.group].num_blocks;
+    if (nb > 0) {
+      for (int32_t id : ft.block_ids)
+        if (id < 0 || id >= nb)
+          throw std::invalid_argument(
+              "block id " + std::to_string(id) + " out of range [0, " +
+              std::to_string(nb) + ")");
+    }
   }
 
   std::shared_ptr<Job> new_job(size_t n_tasks, bool) {

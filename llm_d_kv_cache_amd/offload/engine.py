@@ -73,7 +73,7 @@ class TorchOffloadEngine:
             bb = block_bytes.pop()
             # the actual payload per block may be smaller than the stride;
             # canonical layouts are dense so stride == payload
-            native_groups.append((ptrs, strides, bb))
+            native_groups.append((ptrs, strides, bb, int(g[0].shape[0])))
             record_bytes = bb // 2 + 4 if config.serialize == "fp8_e4m3" else bb
             self.group_geometry.append(
                 {

@@ -155,16 +155,16 @@ PYBIND11_MODULE(_kvoffload, m) {
       .def_property_readonly("pending_writes", &StorageOffloadEngine::pending_writes);
 
   py::class_<BlockCopier>(m, "BlockCopier")
-      .def(py::init([](std::vector<std::tuple<std::vector<uintptr_t>,
-                                              std::vector<uint64_t>, uint64_t>>
-                           groups,
-                       bool gpu_mode, int device) {
+      .def(py::init([](std::vector<py::tuple> groups, bool gpu_mode,
+                       int device) {
              std::vector<GroupDesc> gs;
-             for (auto& [ptrs, strides, block_bytes] : groups) {
+             for (auto& t : groups) {
                GroupDesc g;
-               for (auto p : ptrs) g.layer_ptrs.push_back(reinterpret_cast<void*>(p));
-               g.layer_strides = strides;
-               g.block_bytes = block_bytes;
+               for (auto p : t[0].cast<std::vector<uintptr_t>>())
+                 g.layer_ptrs.push_back(reinterpret_cast<void*>(p));
+               g.layer_strides = t[1].cast<std::vector<uint64_t>>();
+               g.block_bytes = t[2].cast<uint64_t>();
+               if (t.size() > 3) g.num_blocks = t[3].cast<int64_t>();
                gs.push_back(std::move(g));
              }
              py::gil_scoped_release rel;

@@ -48,7 +48,7 @@ class BlockCopier {
   void gather(int group, const std::vector<int32_t>& ids, void* dst,
               uintptr_t stream) {
     const GroupDesc& g = groups_.at(group);
-    check_ids(ids);
+    check_ids(ids, g);
     if (!gpu_mode_) {
       host_copy(g, ids, static_cast<uint8_t*>(dst), /*to_packed=*/true);
       return;
@@ -64,7 +64,7 @@ class BlockCopier {
   void scatter(int group, const std::vector<int32_t>& ids, const void* src,
                uintptr_t stream) {
     const GroupDesc& g = groups_.at(group);
-    check_ids(ids);
+    check_ids(ids, g);
     if (!gpu_mode_) {
       host_copy(g, ids, const_cast<uint8_t*>(static_cast<const uint8_t*>(src)),
                 /*to_packed=*/false);
@@ -79,9 +79,14 @@ class BlockCopier {
   }
 
  private:
-  static void check_ids(const std::vector<int32_t>& ids) {
+  static void check_ids(const std::vector<int32_t>& ids, const GroupDesc& g) {
     if (ids.empty() || ids.size() > kMaxBlocksPerFileHost)
       throw std::invalid_argument("block count must be in [1, 64]");
+    if (g.num_blocks > 0) {
+      for (int32_t id : ids)
+        if (id < 0 || id >= g.num_blocks)
+          throw std::invalid_argument("block id out of range");
+    }
   }
 
   void host_copy(const GroupDesc& g, const std::vector<int32_t>& ids,

@@ -89,7 +89,7 @@ class ObjStorageEngine:
             ptrs = [t.data_ptr() for t in g]
             strides = [t.stride(0) * t.element_size() for t in g]
             bb = strides[0]
-            native_groups.append((ptrs, strides, bb))
+            native_groups.append((ptrs, strides, bb, int(g[0].shape[0])))
             record = bb // 2 + 4 if config.serialize == "fp8_e4m3" else bb
             self.group_geometry.append(
                 {"num_layers": len(g), "block_bytes": bb, "record_bytes": record,

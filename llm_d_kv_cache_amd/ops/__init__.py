@@ -22,7 +22,8 @@ def block_copier(groups: Sequence[Sequence], device: Optional[int] = None):
     native = [
         ([t.data_ptr() for t in g],
          [t.stride(0) * t.element_size() for t in g],
-         g[0].stride(0) * g[0].element_size())
+         g[0].stride(0) * g[0].element_size(),
+         int(g[0].shape[0]))
         for g in groups
     ]
     dev = device if device is not None else (

@@ -104,7 +104,7 @@ class PeerMigrationService:
         for g in groups:
             ptrs = [t.data_ptr() for t in g]
             strides = [t.stride(0) * t.element_size() for t in g]
-            native_groups.append((ptrs, strides, strides[0]))
+            native_groups.append((ptrs, strides, strides[0], int(g[0].shape[0])))
             self._geo.append({"num_layers": len(g), "block_bytes": strides[0]})
         self._copier = _kvoffload.BlockCopier(native_groups, self.gpu_mode,
                                               self.device)

@@ -15,7 +15,7 @@ import logging
 import multiprocessing as mp
 import os
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Optional
 
 log = logging.getLogger(__name__)

@@ -15,8 +15,8 @@ import hashlib
 import json
 import os
 import re
-from dataclasses import asdict, dataclass, field
-from typing import Optional, Tuple
+from dataclasses import asdict, dataclass
+from typing import Tuple
 
 
 @dataclass

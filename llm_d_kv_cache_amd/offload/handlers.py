@@ -11,7 +11,7 @@ already-on-GPU head both at file granularity and inside the first file
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Tuple
 
 from .engine import TorchOffloadEngine

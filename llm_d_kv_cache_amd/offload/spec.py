@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import math
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence
+from typing import Optional, Sequence
 
 from .engine import OffloadEngineConfig, TorchOffloadEngine
 from .events import StorageEventPublisher

@@ -118,14 +118,14 @@ inline bool parse_event(MsgCursor& c, ParsedEvent& out) {
     out.type = EventType::kBlockStored;
     auto& e = out.stored;
     if (n < 5) throw MsgpackError("BlockStored: need at least 5 fields");
-    uint32_t nh = c.array_len();
+    uint32_t nh = c.bounded_len(c.array_len());
     e.block_hashes.reserve(nh);
     for (uint32_t i = 0; i < nh; ++i) e.block_hashes.push_back(decode_hash(c));
     if (c.is_nil())
       c.nil();
     else
       e.parent_hash = decode_hash(c);
-    uint32_t nt = c.array_len();
+    uint32_t nt = c.bounded_len(c.array_len());
     e.tokens.reserve(nt);
     for (uint32_t i = 0; i < nt; ++i)
       e.tokens.push_back(static_cast<uint32_t>(c.uint64()));
@@ -148,7 +148,7 @@ inline bool parse_event(MsgCursor& c, ParsedEvent& out) {
         c.nil();
       } else {
         e.has_extra_keys = true;
-        uint32_t nb = c.array_len();
+        uint32_t nb = c.bounded_len(c.array_len());
         e.extra_keys.resize(nb);
         for (uint32_t b = 0; b < nb; ++b) {
           if (c.is_nil()) {
@@ -198,7 +198,7 @@ inline bool parse_event(MsgCursor& c, ParsedEvent& out) {
     out.type = EventType::kBlockRemoved;
     auto& e = out.removed;
     if (n < 2) throw MsgpackError("BlockRemoved: need at least 2 fields");
-    uint32_t nh = c.array_len();
+    uint32_t nh = c.bounded_len(c.array_len());
     e.block_hashes.reserve(nh);
     for (uint32_t i = 0; i < nh; ++i) e.block_hashes.push_back(decode_hash(c));
     consumed = 2;
@@ -231,7 +231,7 @@ inline EventBatch parse_batch(const uint8_t* data, size_t n) {
   uint32_t nf = c.array_len();
   if (nf < 2) throw MsgpackError("batch: need [ts, events]");
   batch.timestamp = c.f64();
-  uint32_t ne = c.array_len();
+  uint32_t ne = c.bounded_len(c.array_len());
   batch.events.reserve(ne);
   for (uint32_t i = 0; i < ne; ++i) {
     ParsedEvent ev;

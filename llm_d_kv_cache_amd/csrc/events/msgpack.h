@@ -24,6 +24,15 @@ class MsgCursor {
 
   bool done() const { return p_ >= end_; }
   const uint8_t* pos() const { return p_; }
+  size_t remaining() const { return static_cast<size_t>(end_ - p_); }
+
+  // An N-element container needs at least N bytes of payload; a declared
+  // count beyond that is malformed (and would otherwise drive unbounded
+  // reserve() on attacker-controlled input).
+  uint32_t bounded_len(uint32_t n) const {
+    if (n > remaining()) throw MsgpackError("declared count exceeds payload");
+    return n;
+  }
 
   uint8_t peek() const {
     need(1);

@@ -174,10 +174,10 @@ class MsgCursor {
       return;
     }
     switch (b) {
-      case 0xcc: case 0xd0: p_ += 2; return;
-      case 0xcd: case 0xd1: p_ += 3; return;
-      case 0xce: case 0xd2: case 0xca: p_ += 5; return;
-      case 0xcf: case 0xd3: case 0xcb: p_ += 9; return;
+      case 0xcc: case 0xd0: need(2); p_ += 2; return;
+      case 0xcd: case 0xd1: need(3); p_ += 3; return;
+      case 0xce: case 0xd2: case 0xca: need(5); p_ += 5; return;
+      case 0xcf: case 0xd3: case 0xcb: need(9); p_ += 9; return;
       default: throw MsgpackError("skip: unsupported type byte");
     }
   }

@@ -135,7 +135,7 @@ inline bool send_frame(int fd, const void* body, size_t n, bool more, bool comma
   return write_all(fd, hdr, hlen) && write_all(fd, body, n);
 }
 
-inline bool recv_frame(int fd, Frame& f, size_t max_size = 1ull << 30) {
+inline bool recv_frame(int fd, Frame& f, size_t max_size = 256ull << 20) {
   uint8_t flags;
   if (!read_all(fd, &flags, 1)) return false;
   f.more = flags & 1;

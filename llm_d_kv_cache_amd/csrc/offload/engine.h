@@ -210,6 +210,10 @@ class StorageOffloadEngine {
 
   int64_t async_store(std::vector<FileTransfer> files, uintptr_t caller_stream) {
     auto job = new_job(files.size(), /*is_store=*/true);
+    if (files.empty()) {
+      complete_empty(job);
+      return job->id;
+    }
 
     // Dynamic write-queue limit (EMA of write duration): when the backlog
     // exceeds threads * max_queued_seconds / avg_write_seconds, the whole
@@ -284,6 +288,10 @@ class StorageOffloadEngine {
 
   int64_t async_load(std::vector<FileTransfer> files) {
     auto job = new_job(files.size(), /*is_store=*/false);
+    if (files.empty()) {
+      complete_empty(job);
+      return job->id;
+    }
     stats_inc([&](EngineStats& s) { s.loads_submitted += files.size(); });
     for (auto& ft : files) {
       validate(ft);
@@ -364,6 +372,13 @@ class StorageOffloadEngine {
               "block id " + std::to_string(id) + " out of range [0, " +
               std::to_string(nb) + ")");
     }
+  }
+
+  void complete_empty(const std::shared_ptr<Job>& job) {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    finished_.push_back({job->id, true, false});
+    jobs_.erase(job->id);
+    done_cv_.notify_all();
   }
 
   std::shared_ptr<Job> new_job(size_t n_tasks, bool) {

@@ -110,46 +110,86 @@ def bench_control_plane():
 
 def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     """Measure cross-GPU block-pull bandwidth over RCCL/xGMI (or gloo on
-    CPU). Each rank pulls 16-block chunks from its ring neighbor."""
+    CPU). Each rank pulls 16-block chunks from its ring neighbor.
+
+    Hang-proof by construction: every rank executes the SAME collective
+    sequence regardless of local failures — local errors are caught and
+    agreed on via all-reduce(MIN) sync points, never by skipping a
+    barrier (a skipped barrier would stall the whole scaling run until
+    the NCCL watchdog)."""
     from llm_d_kv_cache_amd.peer import PeerMigrationService
 
     gpu = torch.cuda.is_available()
+    dev = "cuda" if gpu else "cpu"
     ctrl_pg = dist.new_group(backend="gloo")
     data_pg = dist.new_group(backend="nccl" if gpu else "gloo")
-    svc = PeerMigrationService([group], data_group=data_pg,
-                               control_group=ctrl_pg,
-                               device=local_rank if gpu else 0)
+
+    def agree(ok_local):
+        t = torch.tensor([1 if ok_local else 0], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        return bool(t.item())
+
+    svc = None
+    err = None
+    try:
+        svc = PeerMigrationService([group], data_group=data_pg,
+                                   control_group=ctrl_pg,
+                                   device=local_rank if gpu else 0)
+    except Exception as e:
+        err = f"service init: {e}"
+    if not agree(svc is not None):
+        if svc is not None:
+            svc.close()
+        return {"ok": False, "error": err or "peer rank failed init"}
+
     n_chunks = 32
     bpf = BLOCKS_PER_FILE
-    for c in range(n_chunks):
-        svc.register_blocks(0xE000 + rank * 1000 + c, 0,
-                            list(range(c * bpf, (c + 1) * bpf)))
-    dist.barrier()
     src = (rank + 1) % world
-    # warmup
-    svc.pull(0xE000 + src * 1000, 0, list(range(1024, 1024 + bpf)),
-             src_rank=src).result(timeout=60)
-    dist.barrier()
+    ok_local = True
+    try:
+        for c in range(n_chunks):
+            svc.register_blocks(0xE000 + rank * 1000 + c, 0,
+                                list(range(c * bpf, (c + 1) * bpf)))
+    except Exception as e:
+        ok_local, err = False, f"register: {e}"
+    if not agree(ok_local):
+        svc.close()
+        return {"ok": False, "error": err or "peer rank failed register"}
+
+    try:
+        svc.pull(0xE000 + src * 1000, 0, list(range(1024, 1024 + bpf)),
+                 src_rank=src).result(timeout=120)  # warmup
+    except Exception as e:
+        ok_local, err = False, f"warmup pull: {e}"
+    if not agree(ok_local):
+        svc.close()
+        return {"ok": False, "error": err or "peer rank failed warmup"}
+
     if gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    futs = [
-        svc.pull(0xE000 + src * 1000 + c, 0,
-                 list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
-                 src_rank=src, timeout=120)
-        for c in range(n_chunks)
-    ]
-    ok = all(f.result(timeout=120) for f in futs)
+    try:
+        futs = [
+            svc.pull(0xE000 + src * 1000 + c, 0,
+                     list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
+                     src_rank=src, timeout=120)
+            for c in range(n_chunks)
+        ]
+        ok_local = all(f.result(timeout=150) for f in futs)
+    except Exception as e:
+        ok_local, err = False, f"pull: {e}"
     if gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    dist.barrier()
+    ok = agree(ok_local)
     pulled = svc.stats().bytes_received
-    t = torch.tensor([dt], device="cuda" if gpu else "cpu")
+    t = torch.tensor([dt], device=dev)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     svc.close()
+    if not ok:
+        return {"ok": False, "error": err or "peer pull failed on a rank"}
     return {
-        "ok": bool(ok),
+        "ok": True,
         "pull_GBps_per_gpu": round(pulled / dt / 1e9, 2),
         "pull_GBps_aggregate": round(pulled * world / float(t.item()) / 1e9, 2),
         "chunk_bytes": pulled // n_chunks,

@@ -1,0 +1,84 @@
+"""Tiered KV resolution: one call walks the whole hierarchy.
+
+A chunk requested for a request's prefix is resolved in cost order:
+
+  1. local filesystem/DRAM tier (the offload engine's load path — a
+     pinned-cache hit never touches the filesystem),
+  2. peer GPUs over RCCL/xGMI (cheaper than any host round trip when a
+     same-node rank still holds the blocks in HBM),
+  3. miss -> the caller recomputes (prefill).
+
+This is the node-side consumer of the global index: the scheduler's
+scorer says WHICH pod likely holds a prefix; TieredKVLoader turns that
+into bytes in local KV pages.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Sequence
+
+log = logging.getLogger(__name__)
+
+
+class TieredKVLoader:
+    def __init__(self, load_handler=None, peer_service=None,
+                 peer_ranks: Sequence[int] = (), pull_timeout_s: float = 30.0):
+        """load_handler: StorageToGPUHandler (or None when the node has no
+        storage tier). peer_service: PeerMigrationService (or None).
+        peer_ranks: same-node ranks to try, in preference order."""
+        self.load_handler = load_handler
+        self.peer = peer_service
+        self.peer_ranks = list(peer_ranks)
+        self.pull_timeout_s = pull_timeout_s
+
+    def resolve(self, chunk_hash: int, block_ids: Sequence[int],
+                group: int = 0,
+                preferred_ranks: Optional[Sequence[int]] = None) -> str:
+        """Fill local pages `block_ids` with the chunk's KV. Returns the
+        tier that served it: 'storage' (covers the DRAM cache), 'peer', or
+        'miss'."""
+        if self.load_handler is not None:
+            job = self.load_handler.transfer_async([chunk_hash],
+                                                   {group: list(block_ids)})
+            if self._wait(self.load_handler, job):
+                return "storage"
+        if self.peer is not None:
+            for rank in (preferred_ranks if preferred_ranks is not None
+                         else self.peer_ranks):
+                try:
+                    ok = self.peer.pull(chunk_hash, group, list(block_ids),
+                                        src_rank=rank,
+                                        timeout=self.pull_timeout_s
+                                        ).result(timeout=self.pull_timeout_s + 5)
+                except Exception as e:
+                    log.warning("peer pull from rank %d failed: %s", rank, e)
+                    continue
+                if ok:
+                    return "peer"
+        return "miss"
+
+    def resolve_prefix(self, chunk_hashes: Sequence[int],
+                       block_ids: Sequence[int], blocks_per_chunk: int,
+                       group: int = 0) -> int:
+        """Resolve a prefix chunk by chunk; stops at the first miss (a
+        prefix must be contiguous). Returns the number of chunks filled."""
+        filled = 0
+        for ci, h in enumerate(chunk_hashes):
+            ids = block_ids[ci * blocks_per_chunk:(ci + 1) * blocks_per_chunk]
+            if not ids:
+                break
+            if self.resolve(h, ids, group=group) == "miss":
+                break
+            filled += 1
+        return filled
+
+    def _wait(self, handler, job_id) -> bool:
+        import time
+
+        deadline = time.time() + self.pull_timeout_s
+        while time.time() < deadline:
+            for res in handler.get_finished():
+                if res.job_id == job_id:
+                    return res.success
+            time.sleep(0.002)
+        return False

@@ -133,6 +133,7 @@ def scenario_multi_group(rank, svc, group):
     "scenario_missing_chunk",
     "scenario_bidirectional",
     "scenario_multi_group",
+    "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):
     spawn2(scenario, tmp_path)
@@ -161,3 +162,55 @@ def test_block_copier_roundtrip():
     for bi, b in enumerate([0, 2, 4, 6]):
         for l in range(3):
             assert torch.equal(g2[l][b], view[bi, l])
+
+
+def scenario_tiered_loader(rank, svc, group):
+    """storage -> peer -> miss resolution across the tier hierarchy."""
+    import tempfile
+
+    import torch.distributed as dist
+
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        StorageToGPUHandler,
+        TorchOffloadEngine,
+    )
+    from llm_d_kv_cache_amd.peer.tiered import TieredKVLoader
+
+    eng = TorchOffloadEngine(
+        [group], OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=4,
+                                     copy_path="host"),
+    )
+    root = tempfile.mkdtemp(prefix=f"tiered_r{rank}_")
+    mapper = FileMapper(root, KVCacheLayoutConfig(model="tiered"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    load = StorageToGPUHandler(eng, mapper, [4])
+
+    CH_STORAGE, CH_PEER, CH_MISS = 0x51, 0x52, 0x53
+    if rank == 0:
+        # rank 0 holds CH_PEER in "HBM" (registry) only
+        svc.register_blocks(CH_PEER, 0, [0, 1, 2, 3])
+    else:
+        # rank 1 has CH_STORAGE on its local filesystem
+        store.transfer_async([CH_STORAGE], {0: [0, 1, 2, 3]})
+        import time
+
+        deadline = time.time() + 10
+        while not store.get_finished() and time.time() < deadline:
+            time.sleep(0.01)
+    dist.barrier()
+
+    if rank == 1:
+        loader = TieredKVLoader(load_handler=load, peer_service=svc,
+                                peer_ranks=[0])
+        assert loader.resolve(CH_STORAGE, [8, 9, 10, 11]) == "storage"
+        assert loader.resolve(CH_PEER, [12, 13, 14, 15]) == "peer"
+        assert loader.resolve(CH_MISS, [16, 17, 18, 19]) == "miss"
+        # prefix semantics: stop at the first gap
+        filled = loader.resolve_prefix([CH_STORAGE, CH_MISS, CH_PEER],
+                                       list(range(20, 32)), 4)
+        assert filled == 1
+    dist.barrier()

@@ -290,8 +290,9 @@ PYBIND11_MODULE(_kvcore, m) {
 
   py::class_<EventPool, std::shared_ptr<EventPool>>(m, "EventPool")
       .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<IndexBackend>,
-                    size_t>(),
-           py::arg("token_processor"), py::arg("index"), py::arg("concurrency") = 4)
+                    size_t, bool>(),
+           py::arg("token_processor"), py::arg("index"), py::arg("concurrency") = 4,
+           py::arg("dp_rank_routing") = false)
       .def("start", &EventPool::start, py::call_guard<py::gil_scoped_release>())
       .def("shutdown", &EventPool::shutdown, py::call_guard<py::gil_scoped_release>())
       .def(

@@ -68,10 +68,16 @@ struct PoolStats {
 
 class EventPool {
  public:
+  // dp_rank_routing: when a batch carries DataParallelRank, treat each DP
+  // rank as its own scheduling target ("<pod>-dp<r>") so the scorer can
+  // route to the rank that actually holds the blocks. The reference
+  // decodes the field but leaves routing as future work
+  // (docs/architecture.md:292, vllm_adapter.go:91-96).
   EventPool(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<IndexBackend> index,
-            size_t concurrency = 4)
+            size_t concurrency = 4, bool dp_rank_routing = false)
       : tp_(std::move(tp)), index_(std::move(index)),
-        queues_(std::max<size_t>(1, concurrency)) {}
+        queues_(std::max<size_t>(1, concurrency)),
+        dp_rank_routing_(dp_rank_routing) {}
 
   ~EventPool() { shutdown(); }
 
@@ -171,6 +177,8 @@ class EventPool {
       parse_failures_.fetch_add(1, std::memory_order_relaxed);
       return;
     }
+    if (dp_rank_routing_ && batch.dp_rank.has_value())
+      pod += "-dp" + std::to_string(*batch.dp_rank);
     for (const auto& ev : batch.events) {
       switch (ev.type) {
         case EventType::kBlockStored:
@@ -292,6 +300,7 @@ class EventPool {
   std::vector<Queue> queues_;
   std::vector<std::thread> workers_;
   std::mutex lifecycle_mu_;
+  bool dp_rank_routing_ = false;
   std::atomic<bool> running_{false};
   std::atomic<uint64_t> enqueued_{0}, processed_{0}, parse_failures_{0},
       dropped_parent_misses_{0};

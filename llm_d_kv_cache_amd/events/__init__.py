@@ -21,6 +21,8 @@ class EventPoolConfig:
     concurrency: int = 4
     engine_type: str = "vllm"  # wire layouts of vllm and sglang both decode
     discover_pods: bool = False  # False: bind fan-in; True: dial per pod
+    # route per DP rank ("<pod>-dp<r>") when batches carry DataParallelRank
+    dp_rank_routing: bool = False
 
 
 class KVEventsPool:
@@ -31,7 +33,8 @@ class KVEventsPool:
         k = ensure_native()
         self._k = k
         self._pool = k.EventPool(
-            indexer.token_processor, indexer.index, config.concurrency
+            indexer.token_processor, indexer.index, config.concurrency,
+            config.dp_rank_routing,
         )
         self._subscriber = None
 

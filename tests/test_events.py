@@ -243,3 +243,22 @@ def test_sglang_layout_decodes():
     ev = ["BlockStored", [5], None, list(range(16)), 16, None, None, None, None]
     pool.process(TOPIC, 0, encode_batch([ev]))
     assert ix.score_tokens(list(range(16)), MODEL) == {POD: 1.0}
+
+
+def test_dp_rank_routing():
+    """Opt-in DP-rank routing: each data-parallel rank becomes its own
+    scoring target (beyond the reference, which decodes but drops the
+    field)."""
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(dp_rank_routing=True), ix)
+    tokens = list(range(32))
+    pool.process(TOPIC, 0, encode_batch([stored(tokens, [1, 2])], dp_rank=0))
+    pool.process(TOPIC, 1,
+                 encode_batch([stored(tokens[:16], [3])], dp_rank=1))
+    scores = ix.score_tokens(tokens, MODEL)
+    assert scores == {f"{POD}-dp0": 2.0, f"{POD}-dp1": 1.0}
+    # default stays merged per pod
+    ix2 = KVCacheIndexer(IndexerConfig())
+    pool2 = KVEventsPool(EventPoolConfig(), ix2)
+    pool2.process(TOPIC, 0, encode_batch([stored(tokens, [1, 2])], dp_rank=3))
+    assert ix2.score_tokens(tokens, MODEL) == {POD: 2.0}

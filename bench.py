@@ -28,11 +28,15 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 MODEL = "meta-llama/Llama-3-8B"
 NUM_LAYERS = 32
+if os.environ.get("KVC_BENCH_TINY"):  # CPU rank-coordination shakeout only
+    NUM_LAYERS = 4
 KV_HEADS = 8
 HEAD_SIZE = 128
 BLOCK_TOKENS = 16
 BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2  # K+V, bf16 = 64 KiB
 FILES_PER_STEP = 64
+if os.environ.get("KVC_BENCH_TINY"):
+    FILES_PER_STEP = 16
 BLOCKS_PER_FILE = 16  # 256-token offload chunks
 
 

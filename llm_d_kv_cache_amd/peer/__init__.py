@@ -117,6 +117,7 @@ class PeerMigrationService:
         self._q_mu = threading.Lock()
         self._stats = PeerStats()
         self._stopping = False
+        self._byes_seen = 0
         self._next_req_id = self.rank + 1
         if self.gpu_mode:
             self._comm_stream = torch.cuda.Stream(device=self.device)
@@ -160,12 +161,17 @@ class PeerMigrationService:
         return fut
 
     def close(self) -> None:
+        """Collective: every rank must call close() (the shutdown protocol
+        synchronizes on the control group so no listener is left blocked
+        in a recv — a pending gloo recv at process-group destruction
+        aborts the process)."""
         if self._stopping:
             return
+        # Phase 1: stop the listener-exit flag BEFORE any BYE can arrive.
+        self._stopping = True
         with self._q_mu:
             self._cmd_q.append(("bye",))
-        self._thread.join(timeout=15.0)
-        self._stopping = True
+        self._thread.join(timeout=30.0)
 
     # ---- control listener (blocking gloo recv, any source) ------------------
 
@@ -177,11 +183,16 @@ class PeerMigrationService:
                 dist.recv(buf, src=None, group=self.control_group)
             except Exception:
                 return  # process group torn down
-            with self._q_mu:
-                self._ctrl_q.append(buf)
-            if self._stopping:
-                return
-            if int(buf[1]) == OP_BYE and int(buf[0]) == self.rank:
+            if int(buf[1]) == OP_BYE:
+                self._byes_seen += 1
+            else:
+                with self._q_mu:
+                    self._ctrl_q.append(buf)
+            # Exit only after draining every peer's BYE: unconsumed gloo
+            # buffers (or a still-pending recv) at process-group
+            # destruction abort the process. The close() barrier
+            # guarantees exactly world-1 BYEs arrive after _stopping.
+            if self._stopping and self._byes_seen >= self.world - 1:
                 return
 
     # ---- service loop -------------------------------------------------------
@@ -228,6 +239,15 @@ class PeerMigrationService:
                     try:
                         if kind == "cmd":
                             if payload[0] == "bye":
+                                # Two-phase shutdown: barrier first so every
+                                # rank's _stopping is set before ANY BYE is
+                                # sent; each listener's next recv then
+                                # observes _stopping and exits. Without
+                                # this, a listener whose peers closed early
+                                # blocks in recv forever and the pending op
+                                # aborts process-group destruction.
+                                with contextlib.suppress(Exception):
+                                    dist.barrier(group=self.control_group)
                                 for p in range(self.world):
                                     if p != self.rank:
                                         with contextlib.suppress(Exception):

@@ -3,6 +3,10 @@
 offload engine for N seconds, verifying the engine stays healthy and
 device/pinned memory stays flat (leak hunt).
 
+Disk usage is HARD-BOUNDED (DISK_BUDGET): old generations are unlinked
+synchronously once the budget is exceeded, so the soak can never exhaust
+the backing filesystem (tmpfs exhaustion kills the host).
+
 Run: python tools/chaos_soak.py [seconds]
 """
 import os
@@ -25,6 +29,7 @@ from llm_d_kv_cache_amd.offload import (
 )
 
 SECONDS = float(sys.argv[1]) if len(sys.argv) > 1 else 60.0
+DISK_BUDGET = 4 * 1024**3  # hard cap on bytes resident in the root
 NUM_LAYERS = 16
 BLOCK_BYTES = 64 * 1024
 NUM_BLOCKS = 1024
@@ -63,7 +68,23 @@ def main():
     load = StorageToGPUHandler(eng, mapper, [BPF])
 
     stored_hashes = []
+    live_bytes = [0]
+    file_bytes = {}
     outstanding = {"store": 0, "load": 0}
+
+    def account_store(h, n):
+        b = n * NUM_LAYERS * BLOCK_BYTES
+        file_bytes[h] = b
+        live_bytes[0] += b
+
+    def enforce_budget():
+        while live_bytes[0] > DISK_BUDGET and len(stored_hashes) > 8:
+            h, _ = stored_hashes.pop(0)
+            try:
+                os.unlink(mapper.file_name(h, 0))
+            except OSError:
+                pass
+            live_bytes[0] -= file_bytes.pop(h, 0)
     mem0 = mem_mb()
     t0 = time.time()
     ops = 0
@@ -77,6 +98,8 @@ def main():
             next_hash += 1
             job = store.transfer_async([h], {0: list(range(base, base + n))})
             stored_hashes.append((h, n))
+            account_store(h, n)
+            enforce_budget()
             outstanding["store"] += 1
             if rng.random() < 0.1:
                 store.wait_job(job)  # cancellation path
@@ -104,9 +127,7 @@ def main():
         outstanding["store"] -= len(store.get_finished())
         outstanding["load"] -= len(load.get_finished())
         ops += 1
-        if len(stored_hashes) > 4096:
-            stored_hashes = stored_hashes[-2048:]
-        if ops % 500 == 0:
+        if ops % 200 == 0:
             time.sleep(0.05)  # let the pool breathe
 
     # drain
@@ -118,6 +139,9 @@ def main():
         time.sleep(0.01)
     s = eng.stats()
     mem1 = mem_mb()
+    import shutil
+
+    shutil.rmtree(root, ignore_errors=True)
     print(f"chaos: {ops} ops in {SECONDS:.0f}s | written {s.files_written} "
           f"read {s.files_read} deduped {s.files_deduped} dropped "
           f"{s.writes_dropped} cancelled {s.tasks_cancelled} errors {s.errors} "

@@ -286,13 +286,14 @@ PYBIND11_MODULE(_kvcore, m) {
       .def_readonly("enqueued", &PoolStats::enqueued)
       .def_readonly("processed", &PoolStats::processed)
       .def_readonly("parse_failures", &PoolStats::parse_failures)
-      .def_readonly("dropped_parent_misses", &PoolStats::dropped_parent_misses);
+      .def_readonly("dropped_parent_misses", &PoolStats::dropped_parent_misses)
+      .def_readonly("dropped_backpressure", &PoolStats::dropped_backpressure);
 
   py::class_<EventPool, std::shared_ptr<EventPool>>(m, "EventPool")
       .def(py::init<std::shared_ptr<TokenProcessor>, std::shared_ptr<IndexBackend>,
-                    size_t, bool>(),
+                    size_t, bool, size_t>(),
            py::arg("token_processor"), py::arg("index"), py::arg("concurrency") = 4,
-           py::arg("dp_rank_routing") = false)
+           py::arg("dp_rank_routing") = false, py::arg("max_queue_depth") = 0)
       .def("start", &EventPool::start, py::call_guard<py::gil_scoped_release>())
       .def("shutdown", &EventPool::shutdown, py::call_guard<py::gil_scoped_release>())
       .def(

@@ -64,6 +64,7 @@ struct PoolStats {
   uint64_t processed = 0;
   uint64_t parse_failures = 0;
   uint64_t dropped_parent_misses = 0;
+  uint64_t dropped_backpressure = 0;
 };
 
 class EventPool {
@@ -73,11 +74,17 @@ class EventPool {
   // route to the rank that actually holds the blocks. The reference
   // decodes the field but leaves routing as future work
   // (docs/architecture.md:292, vllm_adapter.go:91-96).
+  // max_queue_depth bounds each shard's backlog (0 = unbounded): when a
+  // flood outruns the workers, the OLDEST queued message of that shard is
+  // dropped (the index converges from later full-prefix events; memory
+  // stays bounded — the workqueue-backpressure role of the reference's
+  // rate-limited queues, pool.go:37-86).
   EventPool(std::shared_ptr<TokenProcessor> tp, std::shared_ptr<IndexBackend> index,
-            size_t concurrency = 4, bool dp_rank_routing = false)
+            size_t concurrency = 4, bool dp_rank_routing = false,
+            size_t max_queue_depth = 0)
       : tp_(std::move(tp)), index_(std::move(index)),
         queues_(std::max<size_t>(1, concurrency)),
-        dp_rank_routing_(dp_rank_routing) {}
+        dp_rank_routing_(dp_rank_routing), max_queue_depth_(max_queue_depth) {}
 
   ~EventPool() { shutdown(); }
 
@@ -110,6 +117,10 @@ class EventPool {
     auto& q = queues_[shard];
     {
       std::lock_guard<std::mutex> g(q.mu);
+      if (max_queue_depth_ > 0 && q.items.size() >= max_queue_depth_) {
+        q.items.pop_front();
+        dropped_backpressure_.fetch_add(1, std::memory_order_relaxed);
+      }
       q.items.push_back(std::move(msg));
     }
     enqueued_.fetch_add(1, std::memory_order_relaxed);
@@ -133,6 +144,7 @@ class EventPool {
     s.processed = processed_.load(std::memory_order_relaxed);
     s.parse_failures = parse_failures_.load(std::memory_order_relaxed);
     s.dropped_parent_misses = dropped_parent_misses_.load(std::memory_order_relaxed);
+    s.dropped_backpressure = dropped_backpressure_.load(std::memory_order_relaxed);
     return s;
   }
 
@@ -301,9 +313,10 @@ class EventPool {
   std::vector<std::thread> workers_;
   std::mutex lifecycle_mu_;
   bool dp_rank_routing_ = false;
+  size_t max_queue_depth_ = 0;
   std::atomic<bool> running_{false};
   std::atomic<uint64_t> enqueued_{0}, processed_{0}, parse_failures_{0},
-      dropped_parent_misses_{0};
+      dropped_parent_misses_{0}, dropped_backpressure_{0};
 };
 
 }  // namespace kvc

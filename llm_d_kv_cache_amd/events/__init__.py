@@ -23,6 +23,8 @@ class EventPoolConfig:
     discover_pods: bool = False  # False: bind fan-in; True: dial per pod
     # route per DP rank ("<pod>-dp<r>") when batches carry DataParallelRank
     dp_rank_routing: bool = False
+    # per-shard backlog bound (0 = unbounded): floods drop oldest messages
+    max_queue_depth: int = 0
 
 
 class KVEventsPool:
@@ -34,7 +36,7 @@ class KVEventsPool:
         self._k = k
         self._pool = k.EventPool(
             indexer.token_processor, indexer.index, config.concurrency,
-            config.dp_rank_routing,
+            config.dp_rank_routing, config.max_queue_depth,
         )
         self._subscriber = None
 

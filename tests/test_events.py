@@ -262,3 +262,19 @@ def test_dp_rank_routing():
     pool2 = KVEventsPool(EventPoolConfig(), ix2)
     pool2.process(TOPIC, 0, encode_batch([stored(tokens, [1, 2])], dp_rank=3))
     assert ix2.score_tokens(tokens, MODEL) == {POD: 2.0}
+
+
+def test_backpressure_bounded_queue():
+    """max_queue_depth bounds memory under floods (oldest dropped; the
+    index converges from later events)."""
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(concurrency=1, max_queue_depth=16), ix)
+    # not started: messages pile up in the single shard, bounded at 16
+    for i in range(100):
+        pool.add_task(TOPIC, i, encode_batch([stored(list(range(16)), [i])]))
+    s = pool.stats()
+    assert s.dropped_backpressure == 84
+    pool._pool.start()
+    pool.drain()
+    assert pool.stats().processed == 16
+    pool.shutdown()

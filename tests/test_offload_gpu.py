@@ -165,7 +165,8 @@ def test_prefix_hash_kernel_matches_cpu():
         assert list(got[s * n_chunks:(s + 1) * n_chunks]) == want
 
 
-def test_gpu_fp8_serialize_roundtrip(tmp_path):
+@pytest.mark.parametrize("copy_path", ["staged", "zero_copy"])
+def test_gpu_fp8_serialize_roundtrip(tmp_path, copy_path):
     """CDNA4 fused gather+quantize / dequantize+scatter kernels: round-trip
     within e4m3 tolerance; compare against plain PyTorch fp32 reference of
     the same quantization."""
@@ -179,9 +180,10 @@ def test_gpu_fp8_serialize_roundtrip(tmp_path):
     eng = TorchOffloadEngine(
         [group],
         OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BPF,
-                            serialize="fp8_e4m3"),
+                            copy_path=copy_path, serialize="fp8_e4m3"),
     )
-    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gpu-fp8"))
+    mapper = FileMapper(str(tmp_path),
+                        KVCacheLayoutConfig(model=f"gpu-fp8-{copy_path}"))
     store = GPUToStorageHandler(eng, mapper, [BPF])
     load = StorageToGPUHandler(eng, mapper, [BPF])
     ids = list(range(BPF))

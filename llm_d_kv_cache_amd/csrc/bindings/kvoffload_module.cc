@@ -52,7 +52,7 @@ PYBIND11_MODULE(_kvoffload, m) {
                        double read_preferring_ratio, double max_write_queued_seconds,
                        bool gpu_mode, int device, const std::string& copy_path,
                        const std::string& serialize, size_t host_cache_bytes,
-                       const std::string& write_policy) {
+                       const std::string& write_policy, bool direct_io) {
              EngineConfig cfg;
              cfg.io_threads = io_threads;
              cfg.gpu_blocks_per_file = gpu_blocks_per_file;
@@ -82,6 +82,7 @@ PYBIND11_MODULE(_kvoffload, m) {
                cfg.write_policy = WritePolicy::kBack;
              else
                throw std::invalid_argument("write_policy must be through|back");
+             cfg.direct_io = direct_io;
              std::vector<GroupDesc> gs;
              for (auto& [ptrs, strides, block_bytes, num_blocks] : groups) {
                GroupDesc g;
@@ -100,7 +101,7 @@ PYBIND11_MODULE(_kvoffload, m) {
            py::arg("max_write_queued_seconds") = 30.0, py::arg("gpu_mode") = false,
            py::arg("device") = 0, py::arg("copy_path") = "staged",
            py::arg("serialize") = "raw", py::arg("host_cache_bytes") = 0,
-           py::arg("write_policy") = "through")
+           py::arg("write_policy") = "through", py::arg("direct_io") = false)
       .def(
           "async_store",
           [](StorageOffloadEngine& e,

@@ -96,6 +96,10 @@ struct EngineConfig {
   // cache-hit loads skip the filesystem (host_cache.h).
   size_t host_cache_bytes = 0;
   WritePolicy write_policy = WritePolicy::kThrough;
+  // O_DIRECT file I/O (page-cache bypass for local NVMe; auto-fallback on
+  // unsupported filesystems/unaligned transfers) — the GDS-substitute
+  // mode (no cuFile exists on ROCm; see docs/architecture.md).
+  bool direct_io = false;
 };
 
 struct FileTransfer {
@@ -516,7 +520,9 @@ class StorageOffloadEngine {
                        [this, cache, slot, path, bytes](WorkerCtx&) {
                          double w0 = now_s();
                          try {
-                           write_file_atomic(path, slot->buf->host(), bytes);
+                           AtomicFileWriter w(path, cfg_.direct_io);
+                           w.write_at(0, slot->buf->host(), bytes);
+                           w.commit();
                            stats_inc([&](EngineStats& s) {
                              s.files_written++;
                              s.writeback_flushes++;
@@ -546,7 +552,7 @@ class StorageOffloadEngine {
         double t_copy = 0, t_io = 0;
         try {
           const size_t chunk = pipeline_chunk(bytes);
-          AtomicFileWriter writer(ft.path);
+          AtomicFileWriter writer(ft.path, cfg_.direct_io);
           std::vector<std::future<void>> futs;
           for (size_t off = 0; off < bytes; off += chunk) {
             size_t n = std::min(chunk, bytes - off);
@@ -657,7 +663,7 @@ class StorageOffloadEngine {
                                          bytes));
       } else {
         check_file_span();
-        FileReader reader(ft.path);
+        FileReader reader(ft.path, cfg_.direct_io);
         double r0 = now_s();
         for (size_t off = 0; off < bytes; off += chunk) {
           size_t n = std::min(chunk, bytes - off);

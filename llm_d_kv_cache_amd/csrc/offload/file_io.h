@@ -55,6 +55,13 @@ inline void make_parent_dirs(const std::string& path) {
   }
 }
 
+constexpr size_t kDirectIoAlign = 4096;
+
+inline bool direct_io_eligible(const void* ptr, uint64_t off, size_t len) {
+  return (reinterpret_cast<uintptr_t>(ptr) % kDirectIoAlign == 0) &&
+         (off % kDirectIoAlign == 0) && (len % kDirectIoAlign == 0);
+}
+
 // Atomic whole-buffer write: <path>.<rand>.tmp then rename.
 inline void write_file_atomic(const std::string& path, const uint8_t* data,
                               size_t len) {
@@ -92,14 +99,20 @@ inline void write_file_atomic(const std::string& path, const uint8_t* data,
 // writes with the PCIe pipeline inside one transfer.
 class AtomicFileWriter {
  public:
-  explicit AtomicFileWriter(const std::string& path) : path_(path) {
+  // direct_io: request O_DIRECT (page-cache bypass — the GDS-substitute
+  // mode for local NVMe; reference GdsFileIO role). Falls back to buffered
+  // I/O when the filesystem refuses O_DIRECT (tmpfs) or when a write's
+  // buffer/offset/length is not 4 KiB-aligned (e.g. fp8 tile records).
+  explicit AtomicFileWriter(const std::string& path, bool direct_io = false)
+      : path_(path) {
     static thread_local std::mt19937_64 rng{std::random_device{}()};
     tmp_ = path + "." + std::to_string(rng()) + ".tmp";
-    fd_ = ::open(tmp_.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0644);
-    if (fd_ < 0 && errno == ENOENT) {
-      make_parent_dirs(tmp_);
-      fd_ = ::open(tmp_.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0644);
+    int flags = O_WRONLY | O_CREAT | O_EXCL;
+    if (direct_io) {
+      fd_ = open_with_dirs(flags | O_DIRECT);
+      direct_ = fd_ >= 0;
     }
+    if (fd_ < 0) fd_ = open_with_dirs(flags);
     if (fd_ < 0)
       throw FileIoError("open " + tmp_ + ": " + std::strerror(errno));
   }
@@ -113,12 +126,23 @@ class AtomicFileWriter {
   AtomicFileWriter(const AtomicFileWriter&) = delete;
 
   void write_at(uint64_t offset, const uint8_t* data, size_t len) {
+    if (direct_ && !direct_io_eligible(data, offset, len)) {
+      // drop O_DIRECT for the rest of this file: mixed modes on one fd
+      // risk short-write semantics differences across filesystems
+      ::fcntl(fd_, F_SETFL, ::fcntl(fd_, F_GETFL) & ~O_DIRECT);
+      direct_ = false;
+    }
     size_t off = 0;
     while (off < len) {
       ssize_t w = ::pwrite(fd_, data + off, len - off,
                            static_cast<off_t>(offset + off));
       if (w < 0) {
         if (errno == EINTR) continue;
+        if (direct_ && (errno == EINVAL)) {
+          ::fcntl(fd_, F_SETFL, ::fcntl(fd_, F_GETFL) & ~O_DIRECT);
+          direct_ = false;
+          continue;
+        }
         throw FileIoError("pwrite " + tmp_ + ": " + std::strerror(errno));
       }
       off += static_cast<size_t>(w);
@@ -136,16 +160,31 @@ class AtomicFileWriter {
   }
 
  private:
+  int open_with_dirs(int flags) {
+    int fd = ::open(tmp_.c_str(), flags, 0644);
+    if (fd < 0 && errno == ENOENT) {
+      make_parent_dirs(tmp_);
+      fd = ::open(tmp_.c_str(), flags, 0644);
+    }
+    return fd;
+  }
+
   std::string path_;
   std::string tmp_;
   int fd_ = -1;
+  bool direct_ = false;
 };
 
 // RAII reader for chunked loads: one open per transfer.
 class FileReader {
  public:
-  explicit FileReader(const std::string& path) : path_(path) {
-    fd_ = ::open(path.c_str(), O_RDONLY);
+  explicit FileReader(const std::string& path, bool direct_io = false)
+      : path_(path) {
+    if (direct_io) {
+      fd_ = ::open(path.c_str(), O_RDONLY | O_DIRECT);
+      direct_ = fd_ >= 0;
+    }
+    if (fd_ < 0) fd_ = ::open(path.c_str(), O_RDONLY);
     if (fd_ < 0)
       throw FileIoError("open " + path + ": " + std::strerror(errno));
   }
@@ -155,12 +194,21 @@ class FileReader {
   FileReader(const FileReader&) = delete;
 
   void read_at(uint64_t offset, uint8_t* buf, size_t len) {
+    if (direct_ && !direct_io_eligible(buf, offset, len)) {
+      ::fcntl(fd_, F_SETFL, ::fcntl(fd_, F_GETFL) & ~O_DIRECT);
+      direct_ = false;
+    }
     size_t got = 0;
     while (got < len) {
       ssize_t r = ::pread(fd_, buf + got, len - got,
                           static_cast<off_t>(offset + got));
       if (r < 0) {
         if (errno == EINTR) continue;
+        if (direct_ && errno == EINVAL) {
+          ::fcntl(fd_, F_SETFL, ::fcntl(fd_, F_GETFL) & ~O_DIRECT);
+          direct_ = false;
+          continue;
+        }
         throw FileIoError("pread " + path_ + ": " + std::strerror(errno));
       }
       if (r == 0) throw FileIoError("short read from " + path_);
@@ -171,6 +219,7 @@ class FileReader {
  private:
   std::string path_;
   int fd_ = -1;
+  bool direct_ = false;
 };
 
 // Read [offset, offset+len) into buf; the file may be a head-partial

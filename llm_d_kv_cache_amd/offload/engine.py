@@ -24,6 +24,7 @@ class OffloadEngineConfig:
     serialize: str = "raw"     # raw | fp8_e4m3 (bf16 pages -> fp8 + scale)
     host_cache_bytes: int = 0  # pinned-DRAM cache tier (0 = off)
     write_policy: str = "through"  # through | back (flush async via DRAM tier)
+    direct_io: bool = False    # O_DIRECT for NVMe (auto-fallback elsewhere)
     device: int = 0
     staging_budget_bytes: int = DEFAULT_STAGING_BUDGET_BYTES
 
@@ -111,6 +112,7 @@ class TorchOffloadEngine:
             serialize=config.serialize,
             host_cache_bytes=config.host_cache_bytes,
             write_policy=config.write_policy,
+            direct_io=config.direct_io,
         )
         del stream
         # keep tensor refs: the native engine holds raw pointers

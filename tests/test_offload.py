@@ -440,3 +440,28 @@ def test_writeback_without_cache_falls_through(tmp_path):
     assert wait_finished(store)[0].success
     assert os.path.exists(mapper.file_name(0xE2, 0))
     assert eng.stats().writeback_flushes == 0
+
+
+def test_direct_io_fallback(tmp_path):
+    """direct_io=True degrades gracefully on filesystems without O_DIRECT
+    (tmp dirs here) and on unaligned fp8 records — round trips stay
+    correct."""
+    group = make_group(seed=14)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host", direct_io=True),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="direct"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    ids = list(range(8))
+    store.transfer_async([0xD10], {0: ids})
+    assert wait_finished(store)[0].success
+    orig = [t[:8].clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xD10], {0: ids})
+    assert wait_finished(load)[0].success
+    for t, o in zip(group, orig):
+        assert torch.equal(t[:8], o)

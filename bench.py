@@ -26,7 +26,15 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-MODEL = "meta-llama/Llama-3-8B"
+# KV geometry presets (--model). Values are the models' published configs;
+# 70B is the per-rank TP=8 shard (1 of 8 KV heads), the deployment
+# BASELINE.json names for the 128k-context fp8 config.
+MODEL_PRESETS = {
+    "llama-3-8b": {"model": "meta-llama/Llama-3-8B", "layers": 32, "kv_heads": 8},
+    "llama-3-70b-tp8": {"model": "meta-llama/Llama-3-70B", "layers": 80,
+                        "kv_heads": 1},
+}
+MODEL = MODEL_PRESETS["llama-3-8b"]["model"]
 NUM_LAYERS = 32
 if os.environ.get("KVC_BENCH_TINY"):  # CPU rank-coordination shakeout only
     NUM_LAYERS = 4
@@ -38,6 +46,15 @@ FILES_PER_STEP = 64
 if os.environ.get("KVC_BENCH_TINY"):
     FILES_PER_STEP = 16
 BLOCKS_PER_FILE = 16  # 256-token offload chunks
+
+
+def apply_model_preset(name):
+    global MODEL, NUM_LAYERS, KV_HEADS, BLOCK_BYTES
+    p = MODEL_PRESETS[name]
+    MODEL = p["model"]
+    NUM_LAYERS = p["layers"] if not os.environ.get("KVC_BENCH_TINY") else 4
+    KV_HEADS = p["kv_heads"]
+    BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2
 
 
 def log(msg):
@@ -213,6 +230,8 @@ def main():
                     choices=["raw", "fp8_e4m3"])
     ap.add_argument("--io-threads", type=int, default=16)
     ap.add_argument("--device-blocks", type=int, default=2048)
+    ap.add_argument("--model", type=str, default="llama-3-8b",
+                    choices=sorted(MODEL_PRESETS))
     ap.add_argument("--host-cache-gb", type=float, default=6.0,
                     help="pinned-DRAM cache tier size (0 disables)")
     ap.add_argument("--write-policy", type=str, default="through",
@@ -226,6 +245,7 @@ def main():
     ap.add_argument("--no-overlap", dest="overlap", action="store_false",
                     help="sequential store-then-load phases")
     args = ap.parse_args()
+    apply_model_preset(args.model)
 
     import torch
 

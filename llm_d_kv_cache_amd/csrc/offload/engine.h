@@ -41,6 +41,9 @@ extern "C" hipError_t kvc_launch_scatter(const void* const*, const uint64_t*, in
 extern "C" hipError_t kvc_launch_gather_fp8(const void* const*, const uint64_t*,
                                             int, uint64_t, const int32_t*, int,
                                             uint8_t*, hipStream_t);
+extern "C" hipError_t kvc_launch_gather_fp8_split(
+    const void* const*, const uint64_t*, int, uint64_t, const int32_t*, int,
+    uint8_t*, float*, hipStream_t);
 extern "C" hipError_t kvc_launch_scatter_fp8(const void* const*, const uint64_t*,
                                              int, uint64_t, const int32_t*, int,
                                              const uint8_t*, hipStream_t);
@@ -483,16 +486,32 @@ class StorageOffloadEngine {
       uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
                                 ? ctx.device_staging->ptr()
                                 : ctx.host_staging->device();
-      hipError_t err =
-          cfg_.serialize == Serialize::kFp8E4M3
-              ? kvc_launch_gather_fp8(
-                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
-                    dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, kernel_dst, ctx.stream)
-              : kvc_launch_gather(
-                    const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
-                    dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, kernel_dst, ctx.stream);
+      hipError_t err;
+      if (cfg_.serialize == Serialize::kFp8E4M3) {
+        if (cfg_.copy_path == CopyPath::kStaged) {
+          // split amax+quantize (chip-filling); the scale scratch lives in
+          // the tail of the device bounce (packed fp8 <= half of raw, so
+          // the raw-sized bounce always has room)
+          float* scratch = reinterpret_cast<float*>(
+              ctx.device_staging->ptr() + bytes);
+          err = kvc_launch_gather_fp8_split(
+              const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+              dev_layer_strides_[ft.group], nl, g.block_bytes,
+              ft.block_ids.data(), nb, kernel_dst, scratch, ctx.stream);
+        } else {
+          // zero-copy writes pinned host directly: atomics over PCIe are
+          // pathological, keep the fused single-workgroup variant
+          err = kvc_launch_gather_fp8(
+              const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+              dev_layer_strides_[ft.group], nl, g.block_bytes,
+              ft.block_ids.data(), nb, kernel_dst, ctx.stream);
+        }
+      } else {
+        err = kvc_launch_gather(
+            const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
+            dev_layer_strides_[ft.group], nl, g.block_bytes,
+            ft.block_ids.data(), nb, kernel_dst, ctx.stream);
+      }
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       double t1 = now_s();
       stats_inc([&](EngineStats& s) { s.t_gather_ms += (t1 - t0) * 1e3; });

@@ -218,6 +218,101 @@ __device__ __forceinline__ float fp8_e4m3_to_f32(uint8_t b) {
 
 }  // namespace
 
+// Pass 1 of the split fp8 gather: per-(tile, wg-slice) partial amax,
+// combined with a device-scope float atomicMax (monotonic re-interpreted
+// uint order works for non-negative floats). Grid: x = slices per tile,
+// y = tile. scales[] must be zeroed before launch.
+__global__ __launch_bounds__(256) void kvc_fp8_amax(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    uint64_t block_bytes, BlockList blocks, float* __restrict__ scales) {
+  const uint32_t tile = blockIdx.y;
+  const int l = tile % num_layers;
+  const int bi = tile / num_layers;
+  const uint64_t n_elems = block_bytes / 2;
+  const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
+      static_cast<const uint8_t*>(layer_ptrs[l]) +
+      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+  const uint4* __restrict__ vsrc = reinterpret_cast<const uint4*>(src);
+  const uint64_t nvec = n_elems / 8;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
+  float amax = 0.0f;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       v < nvec; v += stride) {
+    uint4 w = vsrc[v];
+    const uint32_t* dw = reinterpret_cast<const uint32_t*>(&w);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      amax = fmaxf(amax, fabsf(bf16_to_f32(static_cast<uint16_t>(dw[j]))));
+      amax = fmaxf(amax, fabsf(bf16_to_f32(static_cast<uint16_t>(dw[j] >> 16))));
+    }
+  }
+  __shared__ float lds_max[4];
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_down(amax, off, 64));
+  if ((threadIdx.x & 63) == 0) lds_max[threadIdx.x >> 6] = amax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = fmaxf(fmaxf(lds_max[0], lds_max[1]), fmaxf(lds_max[2], lds_max[3]));
+    // non-negative floats compare correctly as their uint bit patterns
+    atomicMax(reinterpret_cast<unsigned int*>(&scales[tile]),
+              __float_as_uint(m));
+  }
+}
+
+// Pass 2: quantize with full chip parallelism (grid-stride slices per
+// tile), reading the tile amax from scales[] and rewriting it as the
+// actual scale in the output record.
+__global__ __launch_bounds__(256) void kvc_fp8_quant(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    uint64_t block_bytes, BlockList blocks,
+    const float* __restrict__ scales, uint8_t* __restrict__ dst) {
+  const uint32_t tile = blockIdx.y;
+  const int l = tile % num_layers;
+  const int bi = tile / num_layers;
+  const uint64_t n_elems = block_bytes / 2;
+  const uint64_t record = n_elems + 4;
+  const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
+      static_cast<const uint8_t*>(layer_ptrs[l]) +
+      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+  const uint4* __restrict__ vsrc = reinterpret_cast<const uint4*>(src);
+  uint8_t* __restrict__ payload = dst + static_cast<uint64_t>(tile) * record;
+  float amax = __uint_as_float(
+      *reinterpret_cast<const unsigned int*>(&scales[tile]));
+  if (amax <= 0.0f) amax = 1.0f;
+  const float inv_scale = kFp8Max / amax;
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    *reinterpret_cast<float*>(payload + n_elems) = amax / kFp8Max;
+  uint2* __restrict__ vout = reinterpret_cast<uint2*>(payload);
+  const uint64_t nvec = n_elems / 8;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       v < nvec; v += stride) {
+    uint4 w = vsrc[v];
+    const uint32_t* dw = reinterpret_cast<const uint32_t*>(&w);
+    uint2 out;
+    uint32_t lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(
+        bf16_to_f32(static_cast<uint16_t>(dw[0])) * inv_scale,
+        bf16_to_f32(static_cast<uint16_t>(dw[0] >> 16)) * inv_scale, lo, 0);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(
+        bf16_to_f32(static_cast<uint16_t>(dw[1])) * inv_scale,
+        bf16_to_f32(static_cast<uint16_t>(dw[1] >> 16)) * inv_scale, lo, 1);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(
+        bf16_to_f32(static_cast<uint16_t>(dw[2])) * inv_scale,
+        bf16_to_f32(static_cast<uint16_t>(dw[2] >> 16)) * inv_scale, hi, 0);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(
+        bf16_to_f32(static_cast<uint16_t>(dw[3])) * inv_scale,
+        bf16_to_f32(static_cast<uint16_t>(dw[3] >> 16)) * inv_scale, hi, 1);
+    out.x = lo;
+    out.y = hi;
+    vout[v] = out;
+  }
+}
+
+// Legacy single-workgroup-per-tile fused variant (kept for reference and
+// as the fallback when no scratch scale buffer is available).
 __global__ __launch_bounds__(256) void kvc_gather_fp8(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
@@ -337,6 +432,27 @@ extern "C" hipError_t kvc_launch_gather_fp8(
   hipLaunchKernelGGL(kvc_gather_fp8, dim3(tiles), dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
                      bl, tiles, dst);
+  return hipGetLastError();
+}
+
+// Split fp8 gather: amax pass + quantize pass, both chip-filling. scales
+// is caller-provided device scratch of `num_blocks * num_layers` floats.
+extern "C" hipError_t kvc_launch_gather_fp8_split(
+    const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
+    int num_layers, uint64_t block_bytes, const int32_t* block_ids,
+    int num_blocks, uint8_t* dst, float* scales_scratch, hipStream_t stream) {
+  BlockList bl;
+  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
+  hipError_t err = hipMemsetAsync(scales_scratch, 0, tiles * sizeof(float), stream);
+  if (err != hipSuccess) return err;
+  dim3 grid = copy_grid(tiles, block_bytes);
+  hipLaunchKernelGGL(kvc_fp8_amax, grid, dim3(256), 0, stream,
+                     layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
+                     bl, scales_scratch);
+  hipLaunchKernelGGL(kvc_fp8_quant, grid, dim3(256), 0, stream,
+                     layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
+                     bl, scales_scratch, dst);
   return hipGetLastError();
 }
 

@@ -489,9 +489,11 @@ class StorageOffloadEngine {
       hipError_t err;
       if (cfg_.serialize == Serialize::kFp8E4M3) {
         if (cfg_.copy_path == CopyPath::kStaged) {
-          // split amax+quantize (chip-filling); the scale scratch lives in
-          // the tail of the device bounce (packed fp8 <= half of raw, so
-          // the raw-sized bounce always has room)
+          // split amax+quantize (chip-filling); the partial-max scratch
+          // (tiles x slices floats, slices <= 2048/tiles + 1) lives in the
+          // tail of the device bounce — packed fp8 is half of raw, so the
+          // raw-sized bounce leaves bytes/1 >= tiles*slices*4 of room
+          // (worst case 2048+tiles floats ~ 16 KiB)
           float* scratch = reinterpret_cast<float*>(
               ctx.device_staging->ptr() + bytes);
           err = kvc_launch_gather_fp8_split(

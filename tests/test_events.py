@@ -278,3 +278,33 @@ def test_backpressure_bounded_queue():
     pool.drain()
     assert pool.stats().processed == 16
     pool.shutdown()
+
+
+def test_hybrid_swa_eviction_keeps_full_attention_chain(setup):
+    """HMA entry model: a hybrid (full-attention + sliding-window) pod
+    that evicted old SWA-group blocks still scores full prefix credit
+    through its full-attention entries — the per-(entry, group) model
+    keeps the chain intact (the forgiving-absent-window-blocks refinement
+    the reference lists as future work falls out of the design)."""
+    ix, pool = setup
+    tokens = list(range(64))  # 4 canonical blocks
+    # group 0 = full attention, group 1 = sliding window 32 tokens
+    pool.process(TOPIC, 0, encode_batch([
+        block_stored_payload([1, 2, 3, 4], None, tokens, 16, group_idx=0,
+                             spec_kind="full_attention"),
+        block_stored_payload([11, 12, 13, 14], None, tokens, 16, group_idx=1,
+                             spec_kind="sliding_window", sliding_window=32),
+    ]))
+    assert ix.score_tokens(tokens, MODEL) == {POD: 4.0}
+    # the engine drops the SWA blocks outside the window (first two chunks)
+    pool.process(TOPIC, 1, encode_batch([
+        block_removed_payload([11, 12], group_idx=1),
+    ]))
+    scores = ix.score_tokens(tokens, MODEL)
+    assert scores == {POD: 4.0}, \
+        "full-attention entries must keep the prefix chain intact"
+    # removing the full-attention blocks does break the chain
+    pool.process(TOPIC, 2, encode_batch([
+        block_removed_payload([2], group_idx=0),
+    ]))
+    assert ix.score_tokens(tokens, MODEL) == {POD: 1.0}

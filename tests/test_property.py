@@ -1,0 +1,86 @@
+"""Property-based tests (hypothesis): hash-chain contract vs the pure
+oracle, wire decode against arbitrary msgpack-built events, and index
+invariants under arbitrary op sequences."""
+import msgpack
+from hypothesis import given, settings, strategies as st
+
+import reference_impl as ref
+from llm_d_kv_cache_amd import ensure_native
+from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+
+k = ensure_native()
+
+tokens_st = st.lists(st.integers(0, 2**32 - 1), min_size=0, max_size=130)
+model_st = st.text(min_size=0, max_size=24)
+
+
+@settings(max_examples=200, deadline=None)
+@given(tokens=tokens_st, model=model_st,
+       block_size=st.sampled_from([1, 4, 16, 64]),
+       seed=st.text(max_size=8))
+def test_hash_chain_matches_oracle(tokens, model, block_size, seed):
+    tp = k.TokenProcessor(block_size, seed)
+    got = tp.tokens_to_block_keys(tokens, model)
+    want = ref.block_keys(tokens, model, block_size=block_size, hash_seed=seed)
+    assert got == want
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    hashes=st.lists(st.one_of(st.integers(0, 2**64 - 1),
+                              st.binary(min_size=1, max_size=16)),
+                    min_size=0, max_size=8),
+    tokens=st.lists(st.integers(0, 2**31 - 1), max_size=64),
+    block_size=st.integers(1, 64),
+    medium=st.one_of(st.none(), st.text(max_size=8)),
+    lora=st.one_of(st.none(), st.text(min_size=1, max_size=8)),
+    extra=st.one_of(
+        st.none(),
+        st.lists(st.one_of(st.none(), st.lists(st.text(max_size=6), max_size=3)),
+                 max_size=6),
+    ),
+)
+def test_arbitrary_block_stored_never_crashes(hashes, tokens, block_size,
+                                              medium, lora, extra):
+    """Any structurally-valid BlockStored the msgpack library can encode
+    either applies or is skipped — never crashes, and the pool stays
+    functional."""
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    ev = ["BlockStored", hashes, None, tokens, block_size, None, medium,
+          lora, extra]
+    payload = msgpack.packb([0.0, [ev]], use_bin_type=True)
+    pool.process("kv@pod@m", 0, payload)
+    s = pool.stats()
+    assert s.processed + s.parse_failures == 1
+
+
+@settings(max_examples=100, deadline=None)
+@given(ops=st.lists(
+    st.tuples(st.sampled_from(["add", "evict", "lookup", "clear"]),
+              st.integers(0, 15),  # key space
+              st.integers(0, 3)),  # pod space
+    min_size=1, max_size=60))
+def test_index_invariants_under_arbitrary_ops(ops):
+    """After any op sequence: every looked-up entry names a pod that was
+    added and not since cleared/evicted past it; lookups never return
+    empty lists for present keys."""
+    idx = k.InMemoryIndex(shards=4, pods_per_key=4)
+    for op, key, pod in ops:
+        e = [k.PodEntry(f"p{pod}", "gpu")]
+        if op == "add":
+            idx.add([key + 100], [key], e)
+        elif op == "evict":
+            idx.evict(key + 100, "engine", e)
+        elif op == "lookup":
+            got = idx.lookup([key] if key else [1])
+            for kk, entries in got.items():
+                assert entries, "present key must never have an empty list"
+        elif op == "clear":
+            idx.clear(f"p{pod}")
+    got = idx.lookup(list(range(16)))
+    for kk, entries in got.items():
+        assert entries
+        for ent in entries:
+            assert ent.pod.startswith("p")

@@ -61,6 +61,21 @@ class KVCacheAwareScorer:
         scores = self.indexer.score_tokens(tokens, model, candidate_pods)
         return {p: scores.get(p, 0.0) for p in candidate_pods}
 
+    def mark_scheduled(self, tokens: Sequence[int], model: str,
+                       pod: str) -> None:
+        """Speculative stickiness (reference scorer parity): after routing
+        a request to `pod`, predictively index its prefix as speculative
+        entries so follow-up requests with the same prefix route sticky
+        before the engine's confirming KVEvents arrive. The real events
+        later overwrite the speculative entries in place."""
+        from llm_d_kv_cache_amd import ensure_native
+
+        k = ensure_native()
+        keys = self.indexer.compute_block_keys(tokens, model)
+        if keys:
+            self.indexer.index.add(
+                [], keys, [k.PodEntry(pod, "gpu", speculative=True)])
+
     def shutdown(self):
         self.manager.shutdown()
         self.pool.shutdown()
@@ -85,6 +100,15 @@ if __name__ == "__main__":
                                   "other-pod": "tcp://127.0.0.1:1"})
     print("scores:", out)
     assert out["127.0.0.1"] == 2.0 and out["other-pod"] == 0.0
+    # speculative stickiness: a fresh prefix routed to other-pod scores
+    # there before any engine event confirms it
+    fresh = list(range(500, 532))
+    scorer.mark_scheduled(fresh, "m", "other-pod")
+    out2 = scorer.score(fresh, "m", ["127.0.0.1", "other-pod"],
+                        endpoints={"127.0.0.1": f"tcp://127.0.0.1:{pub.port}",
+                                   "other-pod": "tcp://127.0.0.1:1"})
+    assert out2["other-pod"] == 2.0
+    print("speculative stickiness ok")
     scorer.shutdown()
     pub.close()
     print("ok")

@@ -82,6 +82,10 @@ class GPUToStorageHandler(_BaseHandler):
         for group, block_ids in block_ids_per_group.items():
             bpf = self.blocks_per_file[group]
             block_ids = list(block_ids)
+            if len(block_ids) > len(chunk_hashes) * bpf:
+                raise ValueError(
+                    f"group {group}: {len(block_ids)} blocks exceed "
+                    f"{len(chunk_hashes)} chunks x {bpf} blocks/chunk")
             for ci, chunk_hash in enumerate(chunk_hashes):
                 ids = block_ids[ci * bpf:(ci + 1) * bpf]
                 if not ids:

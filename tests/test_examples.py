@@ -157,3 +157,29 @@ def test_valkey_example_runs():
         capture_output=True, text=True, timeout=120,
     )
     assert out.returncode == 0, out.stderr
+
+
+def test_kv_cache_manager_facade():
+    from llm_d_kv_cache_amd.events.publisher import EventPublisher, block_stored_payload
+    from llm_d_kv_cache_amd.manager import KVCacheManager, KVCacheManagerConfig
+    from llm_d_kv_cache_amd.events import EventPoolConfig
+
+    mgr = KVCacheManager(KVCacheManagerConfig(
+        events=EventPoolConfig(zmq_endpoint="tcp://127.0.0.1:0"),
+        register_metrics=False,
+    )).start()
+    try:
+        pub = EventPublisher(f"tcp://127.0.0.1:{mgr.events_port}", "pod-f", "m",
+                             bind=False)
+        deadline = time.time() + 10
+        while pub._pub.peer_count < 1 and time.time() < deadline:
+            time.sleep(0.01)
+        time.sleep(0.2)
+        pub.publish_events([block_stored_payload([1, 2], None, list(range(32)), 16)])
+        deadline = time.time() + 10
+        while mgr.events.stats().processed < 1 and time.time() < deadline:
+            time.sleep(0.01)
+        assert mgr.score_tokens(list(range(32)), "m") == {"pod-f": 2.0}
+        pub.close()
+    finally:
+        mgr.shutdown()

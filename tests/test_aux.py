@@ -199,3 +199,34 @@ def test_evictor_restarts_dead_children(tmp_path):
         assert ev._procs["activator"].is_alive()
     finally:
         ev.shutdown()
+
+
+def test_prometheus_offload_engine_metrics(tmp_path):
+    import torch
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        TorchOffloadEngine,
+    )
+    from llm_d_kv_cache_amd.utils.metrics import KVCacheMetricsCollector
+
+    group = [torch.zeros(16, 1024, dtype=torch.uint8)]
+    eng = TorchOffloadEngine(
+        [group], OffloadEngineConfig(io_threads=1, gpu_blocks_per_file=4,
+                                     copy_path="host"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="metrics"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    store.transfer_async([1], {0: [0, 1, 2, 3]})
+    deadline = time.time() + 10
+    while not store.get_finished() and time.time() < deadline:
+        time.sleep(0.01)
+    reg = CollectorRegistry()
+    reg.register(KVCacheMetricsCollector(offload_engine=eng))
+    text = generate_latest(reg).decode()
+    assert "kv_offload_files_written_total 1.0" in text
+    assert "kv_offload_bytes_stored_total 4096.0" in text

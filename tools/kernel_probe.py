@@ -16,11 +16,15 @@ import torch
 
 from llm_d_kv_cache_amd import ensure_offload_native  # noqa: E402
 
-NUM_LAYERS = 32
-BLOCK_BYTES = 64 * 1024
+# default: Llama-3-8B geometry; --small switches to the 70B TP=8 shard
+# (8 KiB tiles x 80 layers — the small-tile regime)
+SMALL = "--small" in sys.argv
+NUM_LAYERS = 80 if SMALL else 32
+BLOCK_BYTES = (8 if SMALL else 64) * 1024
 NUM_BLOCKS = 2048
 BPF = 16
-ITERS = int(sys.argv[1]) if len(sys.argv) > 1 else 50
+args = [a for a in sys.argv[1:] if not a.startswith("-")]
+ITERS = int(args[0]) if args else 50
 
 
 def main():

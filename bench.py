@@ -328,6 +328,7 @@ def main():
     step_bytes = FILES_PER_STEP * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
 
     phase_wall = {"store": 0.0, "load": 0.0, "unlink": 0.0}
+    moved = {"bytes": 0}  # actual store+load volume issued in timed steps
 
     # Background deleters: steady-state disk management runs concurrently
     # with serving (the evictor's job in production), not on the hot path.
@@ -375,6 +376,8 @@ def main():
             if ds < n_store or dl < n_load:
                 time.sleep(0.0002)
         p1 = time.perf_counter()
+        if step_id >= 0:
+            moved["bytes"] += step_bytes + (step_bytes if do_load else 0)
         old = (step_id - 2) * FILES_PER_STEP + 1
         if step_id - 2 >= -args.warmup:
             for h in range(old, old + FILES_PER_STEP):
@@ -407,6 +410,8 @@ def main():
             if done < n_jobs:
                 time.sleep(0.0002)
         p2 = time.perf_counter()
+        if step_id >= 0:
+            moved["bytes"] += 2 * step_bytes
         # drop the previous generation in the background (evictor's role)
         prev = (step_id - 1) * FILES_PER_STEP + 1
         if step_id > 0:
@@ -459,9 +464,9 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    moved_bytes = 2 * step_bytes * args.steps  # store + load, per rank
-    # (overlap mode: every timed step stores gen N and loads gen N-1 — the
-    # same total volume)
+    # actual store + load volume per rank (overlap mode's first step has
+    # no generation to load when warmup == 0)
+    moved_bytes = moved["bytes"]
     total_gbps = moved_bytes * world / elapsed / 1e9
     stats = eng.stats()
 

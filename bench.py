@@ -33,6 +33,10 @@ MODEL_PRESETS = {
     "llama-3-8b": {"model": "meta-llama/Llama-3-8B", "layers": 32, "kv_heads": 8},
     "llama-3-70b-tp8": {"model": "meta-llama/Llama-3-70B", "layers": 80,
                         "kv_heads": 1},
+    # MLA: one compressed latent per token (kv_lora_rank 512 + rope 64),
+    # not K+V heads — block_bytes computed from the latent width
+    "deepseek-v3-mla": {"model": "deepseek-ai/DeepSeek-V3", "layers": 61,
+                        "latent": 576},
 }
 MODEL = MODEL_PRESETS["llama-3-8b"]["model"]
 NUM_LAYERS = 32
@@ -53,8 +57,12 @@ def apply_model_preset(name):
     p = MODEL_PRESETS[name]
     MODEL = p["model"]
     NUM_LAYERS = p["layers"] if not os.environ.get("KVC_BENCH_TINY") else 4
-    KV_HEADS = p["kv_heads"]
-    BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2
+    if "latent" in p:  # MLA: latent vector per token, bf16
+        KV_HEADS = 0
+        BLOCK_BYTES = BLOCK_TOKENS * p["latent"] * 2
+    else:
+        KV_HEADS = p["kv_heads"]
+        BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2
 
 
 def log(msg):

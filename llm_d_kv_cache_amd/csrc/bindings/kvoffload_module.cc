@@ -174,6 +174,30 @@ PYBIND11_MODULE(_kvoffload, m) {
            py::arg("groups"), py::arg("gpu_mode") = false, py::arg("device") = 0)
       .def("packed_bytes", &BlockCopier::packed_bytes, py::arg("group"),
            py::arg("n_blocks"))
+      .def("packed_bytes_fp8", &BlockCopier::packed_bytes_fp8, py::arg("group"),
+           py::arg("n_blocks"))
+      .def("fp8_scratch_bytes", &BlockCopier::fp8_scratch_bytes,
+           py::arg("group"), py::arg("n_blocks"))
+      .def(
+          "gather_fp8",
+          [](BlockCopier& c, int group, std::vector<int32_t> ids, uintptr_t dst,
+             uintptr_t scratch, uintptr_t stream) {
+            py::gil_scoped_release rel;
+            c.gather_fp8(group, ids, reinterpret_cast<void*>(dst),
+                         reinterpret_cast<void*>(scratch), stream);
+          },
+          py::arg("group"), py::arg("block_ids"), py::arg("dst"),
+          py::arg("scratch") = 0, py::arg("stream") = 0)
+      .def(
+          "scatter_fp8",
+          [](BlockCopier& c, int group, std::vector<int32_t> ids, uintptr_t src,
+             uintptr_t stream) {
+            py::gil_scoped_release rel;
+            c.scatter_fp8(group, ids, reinterpret_cast<const void*>(src),
+                          stream);
+          },
+          py::arg("group"), py::arg("block_ids"), py::arg("src"),
+          py::arg("stream") = 0)
       .def(
           "gather",
           [](BlockCopier& c, int group, std::vector<int32_t> ids, uintptr_t dst,

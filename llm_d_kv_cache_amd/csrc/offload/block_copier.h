@@ -8,6 +8,10 @@
 #include "common.h"
 #include "engine.h"  // GroupDesc, kernel launcher decls
 
+extern "C" hipError_t kvc_launch_gather_fp8_split(
+    const void* const*, const uint64_t*, int, uint64_t, const int32_t*, int,
+    uint8_t*, float*, hipStream_t);
+
 namespace kvo {
 
 class BlockCopier {
@@ -41,6 +45,58 @@ class BlockCopier {
   size_t packed_bytes(int group, size_t n_blocks) const {
     const GroupDesc& g = groups_.at(group);
     return n_blocks * g.layer_ptrs.size() * g.block_bytes;
+  }
+
+  // fp8 e4m3 record size: block_bytes/2 payload + f32 scale per tile.
+  size_t packed_bytes_fp8(int group, size_t n_blocks) const {
+    const GroupDesc& g = groups_.at(group);
+    return n_blocks * g.layer_ptrs.size() * (g.block_bytes / 2 + 4);
+  }
+
+  // Scratch floats the fp8 gather needs beyond the packed output
+  // (partial-max slab; copy_grid caps slices at 2048/tiles + 1).
+  size_t fp8_scratch_bytes(int group, size_t n_blocks) const {
+    const GroupDesc& g = groups_.at(group);
+    size_t tiles = n_blocks * g.layer_ptrs.size();
+    return (tiles + 4096) * sizeof(float);
+  }
+
+  // Gather + quantize into dst (device ptr, GPU mode); scratch must hold
+  // fp8_scratch_bytes(). Host mode uses the software codec.
+  void gather_fp8(int group, const std::vector<int32_t>& ids, void* dst,
+                  void* scratch, uintptr_t stream) {
+    const GroupDesc& g = groups_.at(group);
+    check_ids(ids, g);
+    if (!gpu_mode_) {
+      host_fp8(g, ids, static_cast<uint8_t*>(dst), /*quantize=*/true);
+      return;
+    }
+    hipError_t err = kvc_launch_gather_fp8_split(
+        const_cast<const void* const*>(dev_layer_ptrs_[group]),
+        dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        static_cast<uint8_t*>(dst), static_cast<float*>(scratch),
+        reinterpret_cast<hipStream_t>(stream));
+    if (err != hipSuccess) throw HipError(hipGetErrorString(err));
+  }
+
+  // Dequantize + scatter from an fp8 record slab.
+  void scatter_fp8(int group, const std::vector<int32_t>& ids, const void* src,
+                   uintptr_t stream) {
+    const GroupDesc& g = groups_.at(group);
+    check_ids(ids, g);
+    if (!gpu_mode_) {
+      host_fp8(g, ids,
+               const_cast<uint8_t*>(static_cast<const uint8_t*>(src)),
+               /*quantize=*/false);
+      return;
+    }
+    hipError_t err = kvc_launch_scatter_fp8(
+        const_cast<const void* const*>(dev_layer_ptrs_[group]),
+        dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        static_cast<const uint8_t*>(src), reinterpret_cast<hipStream_t>(stream));
+    if (err != hipSuccess) throw HipError(hipGetErrorString(err));
   }
 
   // Gather block_ids of `group` into contiguous dst (device ptr in GPU
@@ -103,6 +159,12 @@ class BlockCopier {
           std::memcpy(page, slab, g.block_bytes);
       }
     }
+  }
+
+  void host_fp8(const GroupDesc& g, const std::vector<int32_t>& ids,
+                uint8_t* packed, bool quantize) const {
+    // software codec shared with the engine's host mode
+    StorageOffloadEngine::host_fp8_copy(g, ids, packed, quantize);
   }
 
   std::vector<GroupDesc> groups_;

@@ -869,8 +869,28 @@ class StorageOffloadEngine {
     return sign ? -v : v;
   }
 
-  void fp8_quantize_tile(const uint8_t* src, size_t block_bytes,
-                         uint8_t* out) const {
+  public:
+  // Shared software fp8 path (BlockCopier's host mode reuses it).
+  static void host_fp8_copy(const GroupDesc& g, const std::vector<int32_t>& ids,
+                            uint8_t* packed, bool quantize) {
+    const size_t nl = g.layer_ptrs.size();
+    const size_t rec = g.block_bytes / 2 + 4;
+    for (size_t bi = 0; bi < ids.size(); ++bi) {
+      for (size_t l = 0; l < nl; ++l) {
+        uint8_t* page = static_cast<uint8_t*>(g.layer_ptrs[l]) +
+                        static_cast<uint64_t>(ids[bi]) * g.layer_strides[l];
+        uint8_t* slab = packed + (bi * nl + l) * rec;
+        if (quantize)
+          fp8_quantize_tile(page, g.block_bytes, slab);
+        else
+          fp8_dequantize_tile(slab, g.block_bytes, page);
+      }
+    }
+  }
+
+  private:
+  static void fp8_quantize_tile(const uint8_t* src, size_t block_bytes,
+                         uint8_t* out) {
     const size_t n = block_bytes / 2;
     const uint16_t* in = reinterpret_cast<const uint16_t*>(src);
     float amax = 0.0f;
@@ -884,8 +904,8 @@ class StorageOffloadEngine {
     std::memcpy(out + n, &scale, 4);
   }
 
-  void fp8_dequantize_tile(const uint8_t* in, size_t block_bytes,
-                           uint8_t* dst) const {
+  static void fp8_dequantize_tile(const uint8_t* in, size_t block_bytes,
+                           uint8_t* dst) {
     const size_t n = block_bytes / 2;
     float scale;
     std::memcpy(&scale, in + n, 4);

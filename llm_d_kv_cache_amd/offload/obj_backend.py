@@ -95,9 +95,7 @@ class ObjStorageEngine:
                 {"num_layers": len(g), "block_bytes": bb, "record_bytes": record,
                  "num_device_blocks": g[0].shape[0]}
             )
-        if config.serialize == "fp8_e4m3":
-            raise NotImplementedError(
-                "fp8 object offload rides the filesystem engine for now")
+        self._fp8 = config.serialize == "fp8_e4m3"
         device = groups[0][0].device.index or 0 if self.gpu_mode else 0
         self._copier = ko.BlockCopier(native_groups, self.gpu_mode, device)
         self._tensors = [list(g) for g in groups]
@@ -112,28 +110,44 @@ class ObjStorageEngine:
 
     # ---- engine surface -----------------------------------------------------
 
+    def _packed(self, group: int, n_blocks: int) -> int:
+        if self._fp8:
+            return self._copier.packed_bytes_fp8(group, n_blocks)
+        return self._copier.packed_bytes(group, n_blocks)
+
     def _slab(self, group: int, n_blocks: int):
-        nb = self._copier.packed_bytes(group, n_blocks)
-        t = self._torch.empty(nb, dtype=self._torch.uint8,
+        t = self._torch.empty(self._packed(group, n_blocks),
+                              dtype=self._torch.uint8,
                               pin_memory=self.gpu_mode)
         return t
 
     def _store_one(self, group: int, key: str, ids: List[int]) -> None:
         if self.client.head(key):
             return  # dedupe
+        nb = self._packed(group, len(ids))
         if self.gpu_mode:
-            dev = self._torch.empty(self._copier.packed_bytes(group, len(ids)),
-                                    dtype=self._torch.uint8, device="cuda")
+            scratch_b = (self._copier.fp8_scratch_bytes(group, len(ids))
+                         if self._fp8 else 0)
+            dev = self._torch.empty(nb + scratch_b, dtype=self._torch.uint8,
+                                    device="cuda")
             stream = self._torch.cuda.Stream()
             with self._torch.cuda.stream(stream):
-                self._copier.gather(group, ids, dev.data_ptr(),
-                                    stream.cuda_stream)
+                if self._fp8:
+                    self._copier.gather_fp8(group, ids, dev.data_ptr(),
+                                            dev.data_ptr() + nb,
+                                            stream.cuda_stream)
+                else:
+                    self._copier.gather(group, ids, dev.data_ptr(),
+                                        stream.cuda_stream)
                 host = self._slab(group, len(ids))
-                host.copy_(dev, non_blocking=True)
+                host.copy_(dev[:nb], non_blocking=True)
             stream.synchronize()
         else:
             host = self._slab(group, len(ids))
-            self._copier.gather(group, ids, host.data_ptr(), 0)
+            if self._fp8:
+                self._copier.gather_fp8(group, ids, host.data_ptr(), 0, 0)
+            else:
+                self._copier.gather(group, ids, host.data_ptr(), 0)
         self.client.put(key, host.numpy().tobytes())
 
     def _load_one(self, group: int, key: str, ids: List[int],
@@ -145,14 +159,16 @@ class ObjStorageEngine:
         if len(data) != len(ids) * rec:
             raise IOError(f"short object read for {key}")
         host = self._torch.frombuffer(bytearray(data), dtype=self._torch.uint8)
+        scatter = (self._copier.scatter_fp8 if self._fp8
+                   else self._copier.scatter)
         if self.gpu_mode:
             dev = host.cuda()
             self._torch.cuda.synchronize()
-            self._copier.scatter(group, ids, dev.data_ptr(),
-                                 self._torch.cuda.current_stream().cuda_stream)
+            scatter(group, ids, dev.data_ptr(),
+                    self._torch.cuda.current_stream().cuda_stream)
             self._torch.cuda.synchronize()
         else:
-            self._copier.scatter(group, ids, host.data_ptr(), 0)
+            scatter(group, ids, host.data_ptr(), 0)
 
     def async_store(self, files, stream: int = None) -> int:
         if self.gpu_mode:

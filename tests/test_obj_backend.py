@@ -135,3 +135,26 @@ def test_obj_dedupe_and_missing(s3):
     assert len(_FakeS3.store) == n_objects
     load.transfer_async([0x999], {0: [0, 1]})
     assert not wait_finished(load)[0].success
+
+
+def test_obj_fp8_roundtrip(s3):
+    torch.manual_seed(6)
+    group = [(torch.randn(16, 1024) * 3).to(torch.bfloat16) for _ in range(2)]
+    eng = ObjStorageEngine([group], ObjStorageConfig(endpoint=s3,
+                                                     serialize="fp8_e4m3"))
+    mapper = ObjKeyMapper(FileMapper("/kv", KVCacheLayoutConfig(model="obj8")))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    load = StorageToGPUHandler(eng, mapper, [4])
+    store.transfer_async([0xF0], {0: [0, 1, 2, 3]})
+    assert wait_finished(store)[0].success
+    # object is ~half the raw bytes
+    raw = 4 * 2 * 2048
+    assert sum(len(v) for v in _FakeS3.store.values()) < raw * 0.6
+    orig = [t[:4].float().clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xF0], {0: [0, 1, 2, 3]})
+    assert wait_finished(load)[0].success
+    for t, o in zip(group, orig):
+        amax = o.abs().amax()
+        assert (t[:4].float() - o).abs().max() <= 0.07 * amax

@@ -266,3 +266,48 @@ def test_gpu_reads_overtake_write_storm(tmp_path):
     deadline = time.time() + 60
     while eng.native.pending_writes > 0 and time.time() < deadline:
         time.sleep(0.05)
+
+
+def test_gpu_obj_fp8_roundtrip():
+    """Object-tier fp8: BlockCopier.gather_fp8/scatter_fp8 on device via the
+    S3 engine (embedded fake server), e4m3 tolerance vs fp32 original."""
+    import sys
+    sys.path.insert(0, os.path.dirname(__file__))
+    from test_obj_backend import _FakeS3, ThreadingHTTPServer
+    import threading
+
+    from llm_d_kv_cache_amd.offload.obj_backend import (
+        ObjKeyMapper, ObjStorageConfig, ObjStorageEngine)
+
+    _FakeS3.store = {}
+    srv = ThreadingHTTPServer(("127.0.0.1", 0), _FakeS3)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        torch.manual_seed(7)
+        group = [(torch.randn(16, 2048, device="cuda") * 2).to(torch.bfloat16)
+                 for _ in range(2)]
+        eng = ObjStorageEngine(
+            [group],
+            ObjStorageConfig(endpoint=f"http://127.0.0.1:{srv.server_port}",
+                             serialize="fp8_e4m3"))
+        mapper = ObjKeyMapper(
+            FileMapper("/kv", KVCacheLayoutConfig(model="gobj8")))
+        store = GPUToStorageHandler(eng, mapper, [8])
+        load = StorageToGPUHandler(eng, mapper, [8])
+        store.transfer_async([0xE8], {0: list(range(8))})
+        assert wait_finished(store)[0].success
+        assert sum(len(v) for v in _FakeS3.store.values()) == \
+            8 * 2 * (2048 * 2 // 2 + 4)
+        orig = [t[:8].float().cpu() for t in group]
+        for t in group:
+            t.zero_()
+        torch.cuda.synchronize()
+        load.transfer_async([0xE8], {0: list(range(8))})
+        assert wait_finished(load)[0].success
+        torch.cuda.synchronize()
+        for t, want in zip(group, orig):
+            amax = want.abs().amax()
+            got = t[:8].float().cpu()
+            assert (got - want).abs().max() <= 0.07 * amax
+    finally:
+        srv.shutdown()

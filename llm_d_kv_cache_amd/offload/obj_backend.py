@@ -6,17 +6,22 @@ handler protocol as the filesystem engine, but chunks live in an object
 store. The data path is GPU pages -> CDNA4 gather into a packed slab ->
 pinned staging -> HTTP PUT (and GET -> staging -> scatter on load); lookup
 is a HEAD per chunk. Works against any S3-compatible endpoint (MinIO,
-Ceph RGW, S3 with anonymous/bucket-policy auth; request signing is left
-to a fronting proxy, the common in-cluster deployment).
+Ceph RGW, S3). Authenticated buckets are supported via AWS Signature V4
+(set access_key/secret_key in ObjStorageConfig); anonymous/proxy-fronted
+endpoints work with no credentials configured.
 """
 from __future__ import annotations
 
 import concurrent.futures
+import hashlib
+import hmac as hmaclib
 import threading
+import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Tuple
 from urllib import request as urlrequest
 from urllib.error import HTTPError, URLError
+from urllib.parse import parse_qsl, quote, urlparse
 
 
 @dataclass
@@ -26,6 +31,65 @@ class ObjStorageConfig:
     io_threads: int = 8
     timeout_s: float = 30.0
     serialize: str = "raw"  # raw | fp8_e4m3 (via the native codec kernels)
+    # AWS SigV4 credentials; leave access_key empty for anonymous access.
+    access_key: str = ""
+    secret_key: str = ""
+    session_token: str = ""
+    region: str = "us-east-1"
+
+
+_EMPTY_SHA256 = hashlib.sha256(b"").hexdigest()
+
+
+def sigv4_headers(method: str, url: str, *, access_key: str, secret_key: str,
+                  region: str, payload_hash: str = _EMPTY_SHA256,
+                  service: str = "s3", amz_date: Optional[str] = None,
+                  session_token: str = "",
+                  extra_headers: Optional[Dict[str, str]] = None,
+                  sign_content_sha256: bool = True) -> Dict[str, str]:
+    """AWS Signature Version 4 headers for one request.
+
+    Pure-Python implementation of the SigV4 canonicalization + HMAC chain
+    (validated against the AWS SigV4 test-suite `get-vanilla` vector in
+    tests/test_obj_backend.py). Signs host, x-amz-date,
+    x-amz-content-sha256 (S3 requires it), the session token if present,
+    and any extra headers passed in (e.g. Range is deliberately NOT
+    signed — S3 treats non-amz headers as optional in the signature).
+    """
+    parsed = urlparse(url)
+    amz_date = amz_date or time.strftime("%Y%m%dT%H%M%SZ", time.gmtime())
+    datestamp = amz_date[:8]
+    canonical_uri = quote(parsed.path or "/", safe="/-_.~")
+    qs = sorted(parse_qsl(parsed.query, keep_blank_values=True))
+    canonical_query = "&".join(
+        f"{quote(k, safe='-_.~')}={quote(v, safe='-_.~')}" for k, v in qs)
+    headers = {"host": parsed.netloc, "x-amz-date": amz_date}
+    if sign_content_sha256:
+        headers["x-amz-content-sha256"] = payload_hash
+    if session_token:
+        headers["x-amz-security-token"] = session_token
+    for k, v in (extra_headers or {}).items():
+        headers[k.lower()] = v.strip()
+    names = sorted(headers)
+    signed = ";".join(names)
+    canonical_headers = "".join(f"{k}:{headers[k]}\n" for k in names)
+    canonical_request = "\n".join(
+        [method, canonical_uri, canonical_query, canonical_headers, signed,
+         payload_hash])
+    scope = f"{datestamp}/{region}/{service}/aws4_request"
+    string_to_sign = "\n".join(
+        ["AWS4-HMAC-SHA256", amz_date, scope,
+         hashlib.sha256(canonical_request.encode()).hexdigest()])
+    key = f"AWS4{secret_key}".encode()
+    for part in (datestamp, region, service, "aws4_request"):
+        key = hmaclib.new(key, part.encode(), hashlib.sha256).digest()
+    sig = hmaclib.new(key, string_to_sign.encode(), hashlib.sha256).hexdigest()
+    out = dict(headers)
+    out.pop("host")  # urllib sets Host itself
+    out["Authorization"] = (
+        f"AWS4-HMAC-SHA256 Credential={access_key}/{scope}, "
+        f"SignedHeaders={signed}, Signature={sig}")
+    return out
 
 
 class ObjClient:
@@ -37,9 +101,21 @@ class ObjClient:
     def _url(self, key: str) -> str:
         return f"{self.cfg.endpoint}/{self.cfg.bucket}/{key}"
 
+    def _sign(self, req: urlrequest.Request, payload: bytes = b"") -> None:
+        if not self.cfg.access_key:
+            return
+        for k, v in sigv4_headers(
+                req.get_method(), req.full_url,
+                access_key=self.cfg.access_key,
+                secret_key=self.cfg.secret_key, region=self.cfg.region,
+                payload_hash=hashlib.sha256(payload).hexdigest(),
+                session_token=self.cfg.session_token).items():
+            req.add_header(k, v)
+
     def put(self, key: str, data: bytes) -> None:
         req = urlrequest.Request(self._url(key), data=data, method="PUT")
         req.add_header("Content-Type", "application/octet-stream")
+        self._sign(req, data)
         with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
             if resp.status not in (200, 201, 204):
                 raise IOError(f"PUT {key}: HTTP {resp.status}")
@@ -49,11 +125,13 @@ class ObjClient:
         if offset or length is not None:
             end = "" if length is None else str(offset + length - 1)
             req.add_header("Range", f"bytes={offset}-{end}")
+        self._sign(req)
         with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
             return resp.read()
 
     def head(self, key: str) -> bool:
         req = urlrequest.Request(self._url(key), method="HEAD")
+        self._sign(req)
         try:
             with urlrequest.urlopen(req, timeout=self.cfg.timeout_s) as resp:
                 return resp.status == 200
@@ -64,6 +142,7 @@ class ObjClient:
 
     def delete(self, key: str) -> None:
         req = urlrequest.Request(self._url(key), method="DELETE")
+        self._sign(req)
         try:
             urlrequest.urlopen(req, timeout=self.cfg.timeout_s)
         except HTTPError:

@@ -158,3 +158,61 @@ def test_obj_fp8_roundtrip(s3):
     for t, o in zip(group, orig):
         amax = o.abs().amax()
         assert (t[:4].float() - o).abs().max() <= 0.07 * amax
+
+
+def test_sigv4_aws_test_vector():
+    """AWS SigV4 test-suite `get-vanilla` vector: known inputs must produce
+    the published signature exactly."""
+    from llm_d_kv_cache_amd.offload.obj_backend import sigv4_headers
+
+    h = sigv4_headers(
+        "GET", "https://example.amazonaws.com/",
+        access_key="AKIDEXAMPLE",
+        secret_key="wJalrXUtnFEMI/K7MDENG+bPxRfiCYEXAMPLEKEY",
+        region="us-east-1", service="service",
+        amz_date="20150830T123600Z", sign_content_sha256=False)
+    assert h["Authorization"] == (
+        "AWS4-HMAC-SHA256 "
+        "Credential=AKIDEXAMPLE/20150830/us-east-1/service/aws4_request, "
+        "SignedHeaders=host;x-amz-date, "
+        "Signature="
+        "5fa00fa31553b73ebf1942676e86291e8372ff2a2260956d9b8aae1d763fbf31")
+
+
+def test_sigv4_client_attaches_auth(s3):
+    """Configured credentials -> every request carries a SigV4 Authorization
+    header (captured server-side) and the data path still round-trips."""
+    from llm_d_kv_cache_amd.offload.obj_backend import ObjClient
+
+    seen = {}
+    orig_put = _FakeS3.do_PUT
+    orig_get = _FakeS3.do_GET
+
+    def spy_put(self):
+        seen["put_auth"] = self.headers.get("Authorization", "")
+        seen["put_sha"] = self.headers.get("x-amz-content-sha256", "")
+        orig_put(self)
+
+    def spy_get(self):
+        seen["get_auth"] = self.headers.get("Authorization", "")
+        orig_get(self)
+
+    _FakeS3.do_PUT = spy_put
+    _FakeS3.do_GET = spy_get
+    try:
+        cli = ObjClient(ObjStorageConfig(endpoint=s3, access_key="AK",
+                                         secret_key="SK", region="eu-west-1",
+                                         session_token="TOK"))
+        cli.put("k1", b"hello")
+        assert cli.get("k1") == b"hello"
+        assert cli.head("k1")
+    finally:
+        _FakeS3.do_PUT = orig_put
+        _FakeS3.do_GET = orig_get
+    import hashlib
+    assert seen["put_auth"].startswith(
+        "AWS4-HMAC-SHA256 Credential=AK/")
+    assert "/eu-west-1/s3/aws4_request" in seen["put_auth"]
+    assert "x-amz-security-token" in seen["put_auth"]
+    assert seen["put_sha"] == hashlib.sha256(b"hello").hexdigest()
+    assert seen["get_auth"].startswith("AWS4-HMAC-SHA256 ")

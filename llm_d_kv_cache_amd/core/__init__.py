@@ -76,6 +76,11 @@ class IndexerConfig:
     redis_index: Optional[RedisIndexConfig] = None
     index: InMemoryIndexConfig = field(default_factory=InMemoryIndexConfig)
     backends: List[KVCacheBackendConfig] = field(default_factory=default_backend_configs)
+    # Window-aware scoring (reference marks hybrid-aware scoring WIP): pods
+    # whose learned KV-cache groups are all sliding-window get credit for a
+    # cached tail covering the window even after out-of-window leading blocks
+    # were evicted. Hints come from the attached events pool's GroupCatalog.
+    window_aware_scoring: bool = True
 
 
 class KVCacheIndexer:
@@ -109,6 +114,21 @@ class KVCacheIndexer:
             )
         weights = {b.name: b.weight for b in self.config.backends}
         self._indexer = k.Indexer(self.token_processor, self.index, weights)
+        self._events_pool = None  # set by KVEventsPool for window hints
+
+    def _attach_pool(self, pool) -> None:
+        self._events_pool = pool
+
+    def _window_hints(self, pods: Sequence[str]) -> Dict[str, int]:
+        if not self.config.window_aware_scoring or self._events_pool is None:
+            return {}
+        bs = self.token_processor.block_size
+        hints: Dict[str, int] = {}
+        for p in pods:
+            w = self._events_pool.native.sliding_window_tokens(p)
+            if w > 0:
+                hints[p] = -(-w // bs)  # ceil(window_tokens / block_tokens)
+        return hints
 
     @property
     def block_size(self) -> int:
@@ -123,7 +143,8 @@ class KVCacheIndexer:
     ) -> Dict[str, float]:
         """Score pods by longest cached prefix for this token stream."""
         res = self._indexer.score_tokens(
-            tokens, model_name, list(pod_identifiers), extra_features
+            tokens, model_name, list(pod_identifiers), extra_features,
+            self._window_hints(pod_identifiers),
         )
         return dict(res.scores)
 
@@ -136,7 +157,8 @@ class KVCacheIndexer:
     ):
         """Like score_tokens but also returns (total_blocks, hit_blocks)."""
         res = self._indexer.score_tokens(
-            tokens, model_name, list(pod_identifiers), extra_features
+            tokens, model_name, list(pod_identifiers), extra_features,
+            self._window_hints(pod_identifiers),
         )
         return dict(res.scores), res.total_blocks, res.hit_blocks
 

@@ -264,15 +264,18 @@ PYBIND11_MODULE(_kvcore, m) {
       .def(
           "score_tokens",
           [](Indexer& ix, py::handle tokens, const std::string& model,
-             const std::vector<std::string>& pods, py::handle extra) {
+             const std::vector<std::string>& pods, py::handle extra,
+             std::unordered_map<std::string, int32_t> window_hints) {
             auto toks = as_tokens(tokens);
             auto ex = as_extra(extra);
             py::gil_scoped_release rel;
             return ix.score_tokens(toks.data(), toks.size(), model, pods,
-                                   ex.empty() ? nullptr : &ex);
+                                   ex.empty() ? nullptr : &ex,
+                                   window_hints.empty() ? nullptr : &window_hints);
           },
           py::arg("tokens"), py::arg("model"),
-          py::arg("pods") = std::vector<std::string>{}, py::arg("extra") = py::none())
+          py::arg("pods") = std::vector<std::string>{}, py::arg("extra") = py::none(),
+          py::arg("window_hints") = std::unordered_map<std::string, int32_t>{})
       .def(
           "compute_block_keys",
           [](Indexer& ix, py::handle tokens, const std::string& model) {
@@ -330,7 +333,12 @@ PYBIND11_MODULE(_kvcore, m) {
         d["sliding_window"] =
             md->sliding_window.has_value() ? py::cast(*md->sliding_window) : py::none();
         return d;
-      });
+      })
+      .def("sliding_window_tokens",
+           [](EventPool& p, const std::string& pod) {
+             return p.group_catalog().sliding_window_tokens(pod);
+           },
+           py::arg("pod"));
 
   py::class_<ZmtpPublisher>(m, "Publisher")
       .def(py::init<const std::string&, bool>(), py::arg("endpoint"),

@@ -33,6 +33,17 @@ struct PodEntry {
 
   bool speculative() const { return flags & 1; }
   bool has_group() const { return flags & 2; }
+
+  // Eviction match: BlockRemoved events often omit group (and never carry
+  // the speculative bit), so eviction matches on pod + tier, constraining
+  // group only when the probe names one. Strict operator== stays for add()
+  // dedupe.
+  static bool evict_match(const PodEntry& stored, const PodEntry& probe) {
+    if (stored.pod != probe.pod || stored.tier != probe.tier) return false;
+    if (probe.has_group())
+      return stored.has_group() && stored.group == probe.group;
+    return true;
+  }
   bool operator==(const PodEntry& o) const {
     return pod == o.pod && tier == o.tier && flags == o.flags && group == o.group;
   }
@@ -362,7 +373,7 @@ class InMemoryIndex : public IndexBackend {
     auto& pods = it->second.pods;
     for (const auto& e : entries) {
       for (size_t i = 0; i < pods.size(); ++i) {
-        if (pods[i] == e) {
+        if (PodEntry::evict_match(pods[i], e)) {
           pods.erase(pods.begin() + i);
           sh.bytes -= sizeof(PodEntry);
           evictions_.fetch_add(1, std::memory_order_relaxed);

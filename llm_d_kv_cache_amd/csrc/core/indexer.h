@@ -44,7 +44,9 @@ class Indexer {
 
   ScoreResult score_tokens(const uint32_t* tokens, size_t n, const std::string& model,
                            const std::vector<std::string>& pods,
-                           const std::vector<BlockExtra>* extra = nullptr) {
+                           const std::vector<BlockExtra>* extra = nullptr,
+                           const std::unordered_map<std::string, int32_t>*
+                               window_hints = nullptr) {
     ScoreResult res;
     auto keys = tp_->tokens_to_block_keys(0, tokens, n, model, extra);
     res.total_blocks = keys.size();
@@ -62,7 +64,14 @@ class Indexer {
 
     auto hits = index_->lookup(keys, filter);
     res.hit_blocks = hits.size();
-    auto scores = scorer_->score(keys, hits);
+    std::unordered_map<uint32_t, int32_t> wh;
+    if (window_hints != nullptr) {
+      for (const auto& [pod, blocks] : *window_hints) {
+        uint32_t id = index_->strings().find(pod);
+        if (id != StringTable::kInvalid) wh.emplace(id, blocks);
+      }
+    }
+    auto scores = scorer_->score(keys, hits, wh.empty() ? nullptr : &wh);
     for (const auto& [pod_id, s] : scores)
       res.scores.emplace(index_->strings().get(pod_id), s);
     return res;

@@ -124,13 +124,32 @@ class RedisIndex : public IndexBackend {
       rks = parse_rk_list(reply.str());
       if (rks.empty()) return;
     }
+    // BlockRemoved may omit group/flags the stored field encodes, so the
+    // exact HDEL field is not constructible client-side: read the fields,
+    // apply the relaxed evict_match, delete the exact matches.
+    std::vector<std::vector<std::string>> reads;
+    for (uint64_t rk : rks) reads.push_back({"HKEYS", rkey(rk)});
+    auto fields = pool_.with(
+        [&](RespConnection& c) { return c.pipeline(reads); });
     std::vector<std::vector<std::string>> cmds;
-    for (uint64_t rk : rks) {
-      std::vector<std::string> cmd{"HDEL", rkey(rk)};
-      for (const auto& e : entries) cmd.push_back(encode_entry(e));
-      cmds.push_back(std::move(cmd));
+    for (size_t ri = 0; ri < rks.size(); ++ri) {
+      std::vector<std::string> cmd{"HDEL", rkey(rks[ri])};
+      if (!fields[ri].is_error && !fields[ri].is_nil()) {
+        for (const auto& f : fields[ri].array()) {
+          PodEntry stored;
+          if (!decode_entry(f.str(), &stored)) continue;
+          for (const auto& e : entries) {
+            if (PodEntry::evict_match(stored, e)) {
+              cmd.push_back(f.str());
+              break;
+            }
+          }
+        }
+      }
+      if (cmd.size() > 2) cmds.push_back(std::move(cmd));
     }
     for (uint64_t rk : rks) cmds.push_back({"HLEN", rkey(rk)});
+    const size_t n_hdel = cmds.size() - rks.size();
     auto replies = pool_.with(
         [&](RespConnection& c) { return c.pipeline(cmds); });
     evictions_.fetch_add(1, std::memory_order_relaxed);
@@ -138,7 +157,7 @@ class RedisIndex : public IndexBackend {
       bool all_empty = true;
       std::vector<std::vector<std::string>> cleanup;
       for (size_t i = 0; i < rks.size(); ++i) {
-        const auto& r = replies[rks.size() + i];
+        const auto& r = replies[n_hdel + i];
         int64_t len = (!r.is_error && !r.is_nil()) ? r.integer() : 0;
         if (len > 0) all_empty = false;
         if (len == 0) cleanup.push_back({"DEL", rkey(rks[i])});

@@ -18,6 +18,8 @@
 #include <unordered_map>
 #include <vector>
 
+#include <algorithm>
+
 #include "../common/fnv.h"
 #include "../core/index.h"
 #include "../core/token_processor.h"
@@ -46,6 +48,23 @@ class GroupCatalog {
     auto jt = it->second.find(group);
     if (jt == it->second.end()) return std::nullopt;
     return jt->second;
+  }
+
+  // Sliding-window hint for window-aware scoring: the pod's window size in
+  // TOKENS if every learned KV-cache group on the pod uses sliding-window
+  // attention (engines then evict out-of-window leading blocks), else 0
+  // (full/hybrid attention needs the whole prefix; with several sliding
+  // groups the MAX window is the binding requirement).
+  int32_t sliding_window_tokens(const std::string& pod) const {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = catalog_.find(pod);
+    if (it == catalog_.end() || it->second.empty()) return 0;
+    int32_t w = 0;
+    for (const auto& [gid, md] : it->second) {
+      if (!md.sliding_window.has_value() || *md.sliding_window <= 0) return 0;
+      w = std::max(w, *md.sliding_window);
+    }
+    return w;
   }
 
  private:

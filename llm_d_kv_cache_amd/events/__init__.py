@@ -38,6 +38,9 @@ class KVEventsPool:
             indexer.token_processor, indexer.index, config.concurrency,
             config.dp_rank_routing, config.max_queue_depth,
         )
+        attach = getattr(indexer, "_attach_pool", None)
+        if attach is not None:
+            attach(self)  # enables window-aware scoring hints
         self._subscriber = None
 
     @property

@@ -125,6 +125,9 @@ class FakeRedis:
                     items.append(self._bulk(f))
                     items.append(self._bulk(v))
                 return self._arr(items)
+            if cmd == b"HKEYS":
+                h = self.hashes.get(args[1], {})
+                return self._arr([self._bulk(f) for f in h])
             if cmd == b"HDEL":
                 h = self.hashes.get(args[1], {})
                 n = 0

@@ -145,3 +145,74 @@ def test_mm_features_read_write_agree():
     # diverges at the tainted block
     assert ix.score_tokens(tokens, MODEL) == {"pod-mm": 1.0}
     assert ix.score_tokens(tokens, MODEL, extra_features=extra) == {"pod-mm": 3.0}
+
+
+def test_window_aware_scoring_sliding_window_pod():
+    """Reference WIP feature (docs/architecture.md:291), implemented here:
+    a pure sliding-window pod keeps a high score after its engine evicts
+    out-of-window leading blocks — the vanilla prefix walk would score 0."""
+    from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_removed_payload,
+        block_stored_payload,
+        encode_batch,
+    )
+
+    ix = make_indexer()
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    tokens = list(range(128))  # 8 blocks of 16
+    eng = list(range(1, 9))
+    pool.process("kv@pod-swa@m", 0, encode_batch([
+        block_stored_payload(eng, None, tokens, 16, group_idx=0,
+                             spec_kind="sliding_window", sliding_window=32)
+    ]))
+    assert pool.native.sliding_window_tokens("pod-swa") == 32
+    assert ix.score_tokens(tokens, MODEL) == {"pod-swa": 8.0}
+
+    # engine evicts the 6 out-of-window leading blocks
+    pool.process("kv@pod-swa@m", 1, encode_batch([
+        block_removed_payload(eng[:6])
+    ]))
+    scores, total, hits = ix.score_tokens_detailed(
+        tokens, MODEL, ["pod-swa"])
+    assert total == 8 and hits == 2
+    # trailing run of 2 blocks covers the 32-token window at P=8
+    assert scores == {"pod-swa": 8.0}
+
+    # with window-aware scoring disabled the pod drops to 0 (prefix broken)
+    ix.config.window_aware_scoring = False
+    assert ix.score_tokens(tokens, MODEL, ["pod-swa"]) == {}
+
+
+def test_window_hints_only_for_pure_sliding_pods():
+    """A hybrid pod (any learned group without a window) gets no hint, and a
+    partial tail shorter than the window earns nothing."""
+    from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_removed_payload,
+        block_stored_payload,
+        encode_batch,
+    )
+
+    ix = make_indexer()
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    tokens = list(range(128))
+    # hybrid pod: sliding group 0 + full-attention group 1
+    pool.process("kv@pod-hy@m", 0, encode_batch([
+        block_stored_payload(list(range(1, 9)), None, tokens, 16, group_idx=0,
+                             spec_kind="sliding_window", sliding_window=32),
+        block_stored_payload(list(range(11, 19)), None, tokens, 16,
+                             group_idx=1, spec_kind="full_attention"),
+    ]))
+    assert pool.native.sliding_window_tokens("pod-hy") == 0
+
+    # pure sliding pod with window 64 (4 blocks) holding only 2 tail blocks:
+    # the window is not covered anywhere -> no windowed credit
+    pool.process("kv@pod-sw2@m", 0, encode_batch([
+        block_stored_payload(list(range(21, 29)), None, tokens, 16,
+                             group_idx=0, spec_kind="sliding_window",
+                             sliding_window=64),
+        block_removed_payload(list(range(21, 27))),
+    ]))
+    assert pool.native.sliding_window_tokens("pod-sw2") == 64
+    assert ix.score_tokens(tokens, MODEL, ["pod-sw2"]) == {}

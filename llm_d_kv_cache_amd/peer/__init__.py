@@ -35,8 +35,16 @@ from typing import Dict, List, Optional, Sequence, Tuple
 OP_PULL_REQ = 1
 OP_PULL_ACK = 2
 OP_BYE = 3
+OP_PULL_REQ_MULTI = 4
+OP_PULL_ACK_MULTI = 5
 
-MSG_LEN = 8  # [sender, op, req_id, group, chunk_hash, n_blocks, ok, pad]
+# Single-chunk ops use the first 8 slots
+# [sender, op, req_id, group, chunk_hash, n_blocks, ok, pad]; batched ops
+# put n_chunks in slot 3, a granted bitmask in slot 6 (ACK), and per-chunk
+# (chunk_hash, group, n_blocks) triples from slot 8. All control messages
+# are the same fixed size so the any-source listener recv stays simple.
+MAX_BATCH = 16
+MSG_LEN = 8 + 3 * MAX_BATCH
 
 
 @dataclass
@@ -160,6 +168,24 @@ class PeerMigrationService:
                                 time.time() + timeout, fut))
         return fut
 
+    def pull_many(self, chunks: Sequence[Tuple[int, int, Sequence[int]]],
+                  src_rank: int,
+                  timeout: float = 30.0) -> concurrent.futures.Future:
+        """Batched pull: up to MAX_BATCH (chunk_hash, group, dst_block_ids)
+        from ONE peer with one control round trip and one xGMI data
+        transfer (small chunks amortize the handshake and the per-transfer
+        launch cost). The Future resolves to a list[bool] per chunk (True =
+        pulled); raises on timeout/shutdown."""
+        if not chunks or len(chunks) > MAX_BATCH:
+            raise ValueError(f"pull_many takes 1..{MAX_BATCH} chunks")
+        fut: concurrent.futures.Future = concurrent.futures.Future()
+        norm = [(int(h), int(g), [int(b) for b in ids])
+                for (h, g, ids) in chunks]
+        with self._q_mu:
+            self._cmd_q.append(("pull_many", norm, src_rank,
+                                time.time() + timeout, fut))
+        return fut
+
     def close(self) -> None:
         """Collective: every rank must call close() (the shutdown protocol
         synchronizes on the control group so no listener is left blocked
@@ -254,6 +280,8 @@ class PeerMigrationService:
                                             dist.send(self._msg(OP_BYE), dst=p,
                                                       group=self.control_group)
                                 run = False
+                            elif payload[0] == "pull_many":
+                                self._start_pull_many(payload, pending_pulls)
                             else:
                                 self._start_pull(payload, pending_pulls)
                         else:
@@ -273,13 +301,27 @@ class PeerMigrationService:
                     st = pending_pulls[req_id]
                     if st["tracker"] is not None and st["tracker"].done():
                         made_progress = True
-                        self._copier.scatter(
-                            st["group"], st["dst_ids"], st["buf"].data_ptr(),
-                            self._comm_stream.cuda_stream if self.gpu_mode else 0)
+                        stream = (self._comm_stream.cuda_stream
+                                  if self.gpu_mode else 0)
+                        if "chunks" in st:  # batched pull
+                            base = st["buf"].data_ptr()
+                            off = 0
+                            for ci, (h, g, ids) in enumerate(st["chunks"]):
+                                if not st["granted"][ci]:
+                                    continue
+                                self._copier.scatter(g, ids, base + off,
+                                                     stream)
+                                off += self._copier.packed_bytes(g, len(ids))
+                            result = st["granted"]
+                        else:
+                            self._copier.scatter(
+                                st["group"], st["dst_ids"],
+                                st["buf"].data_ptr(), stream)
+                            result = True
                         if self.gpu_mode:
                             self._comm_stream.synchronize()
                         self._stats.bytes_received += st["buf"].numel()
-                        st["fut"].set_result(True)
+                        st["fut"].set_result(result)
                         del pending_pulls[req_id]
                     elif time.time() > st["deadline"]:
                         st["fut"].set_exception(
@@ -312,6 +354,24 @@ class PeerMigrationService:
             "deadline": deadline, "fut": fut, "tracker": None, "buf": None,
         }
         self._stats.pulls_requested += 1
+
+    def _start_pull_many(self, payload, pending_pulls) -> None:
+        _, chunks, src, deadline, fut = payload
+        req_id = self._next_req_id
+        self._next_req_id += self.world
+        t = self._msg(OP_PULL_REQ_MULTI, req_id)
+        t[3] = len(chunks)
+        for i, (h, g, ids) in enumerate(chunks):
+            u = h & ((1 << 64) - 1)
+            t[8 + 3 * i] = u - (1 << 64) if u >= (1 << 63) else u
+            t[9 + 3 * i] = g
+            t[10 + 3 * i] = len(ids)
+        self._dist.send(t, dst=src, group=self.control_group)
+        pending_pulls[req_id] = {
+            "chunks": chunks, "src": src, "deadline": deadline, "fut": fut,
+            "tracker": None, "buf": None, "granted": None,
+        }
+        self._stats.pulls_requested += len(chunks)
 
     def _handle_ctrl(self, buf, pending_pulls, pending_sends) -> None:
         dist = self._dist
@@ -349,4 +409,62 @@ class PeerMigrationService:
             else:
                 st["fut"].set_result(False)
                 self._stats.pulls_failed += 1
+                del pending_pulls[req_id]
+        elif op == OP_PULL_REQ_MULTI:
+            n_chunks = int(buf[3])
+            reqs = []
+            for i in range(n_chunks):
+                h = int(buf[8 + 3 * i]) & ((1 << 64) - 1)
+                g = int(buf[9 + 3 * i])
+                nb = int(buf[10 + 3 * i])
+                reqs.append((h, g, nb))
+            mask = 0
+            total = 0
+            served = []
+            for i, (h, g, nb) in enumerate(reqs):
+                ids = self.lookup_local(h, g)
+                if ids is not None and len(ids) == nb:
+                    mask |= 1 << i
+                    total += self._copier.packed_bytes(g, nb)
+                    served.append((g, ids))
+            ack = self._msg(OP_PULL_ACK_MULTI, req_id)
+            ack[3] = n_chunks
+            ack[6] = mask
+            dist.send(ack, dst=sender, group=self.control_group)
+            if mask:
+                slab = (self._torch.empty(total, dtype=self._torch.uint8,
+                                          device="cuda") if self.gpu_mode
+                        else self._torch.empty(total,
+                                               dtype=self._torch.uint8))
+                off = 0
+                stream = (self._comm_stream.cuda_stream
+                          if self.gpu_mode else 0)
+                for g, ids in served:
+                    self._copier.gather(g, ids, slab.data_ptr() + off, stream)
+                    off += self._copier.packed_bytes(g, len(ids))
+                w = dist.isend(slab, dst=sender, group=self.data_group)
+                pending_sends.append((self._track(w), slab))
+                self._stats.pulls_served += len(served)
+                self._stats.bytes_sent += slab.numel()
+        elif op == OP_PULL_ACK_MULTI:
+            st = pending_pulls.get(req_id)
+            if st is None:
+                return
+            mask = ok
+            granted = [bool(mask >> i & 1) for i in range(len(st["chunks"]))]
+            st["granted"] = granted
+            if mask:
+                total = sum(self._copier.packed_bytes(g, len(ids))
+                            for gi, (h, g, ids) in enumerate(st["chunks"])
+                            if granted[gi])
+                slab = (self._torch.empty(total, dtype=self._torch.uint8,
+                                          device="cuda") if self.gpu_mode
+                        else self._torch.empty(total,
+                                               dtype=self._torch.uint8))
+                st["buf"] = slab
+                w = dist.irecv(slab, src=st["src"], group=self.data_group)
+                st["tracker"] = self._track(w)
+            else:
+                st["fut"].set_result([False] * len(st["chunks"]))
+                self._stats.pulls_failed += len(st["chunks"])
                 del pending_pulls[req_id]

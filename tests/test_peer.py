@@ -114,6 +114,40 @@ def scenario_bidirectional(rank, svc, group):
     dist.barrier()
 
 
+def scenario_batched_pull(rank, svc, group):
+    """pull_many: one control round trip + one data transfer for several
+    chunks across groups, with partial grants (missing chunk -> False)."""
+    import torch.distributed as dist
+
+    if rank == 0:
+        svc.register_blocks(0x21, 0, [1, 2])
+        svc.register_blocks(0x22, 1, [5])
+        svc.register_blocks(0x23, 0, [9])
+        golden = [group[0][[1, 2]].clone(), group[1][[1, 2]].clone()]
+        dist.broadcast(golden[0], src=0)
+        dist.broadcast(golden[1], src=0)
+        dist.barrier()
+    else:
+        golden = [torch.zeros(2, 4096, dtype=torch.uint8) for _ in range(2)]
+        dist.broadcast(golden[0], src=0)
+        dist.broadcast(golden[1], src=0)
+        res = svc.pull_many(
+            [(0x21, 0, [14, 15]),      # granted
+             (0x9999, 0, [16]),        # missing -> False
+             (0x23, 0, [17])],         # granted
+            src_rank=0).result(timeout=60)
+        assert res == [True, False, True]
+        assert torch.equal(group[0][14], golden[0][0])
+        assert torch.equal(group[0][15], golden[0][1])
+        assert torch.equal(group[1][14], golden[1][0])
+        assert torch.equal(group[1][15], golden[1][1])
+        # all-missing batch resolves all-False without a data transfer
+        res2 = svc.pull_many([(0x77, 0, [18]), (0x78, 1, [3])],
+                             src_rank=0).result(timeout=60)
+        assert res2 == [False, False]
+        dist.barrier()
+
+
 def scenario_multi_group(rank, svc, group):
     import torch.distributed as dist
 
@@ -133,6 +167,7 @@ def scenario_multi_group(rank, svc, group):
     "scenario_missing_chunk",
     "scenario_bidirectional",
     "scenario_multi_group",
+    "scenario_batched_pull",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):

@@ -58,6 +58,17 @@ def _run_rank(rank, world, init_file, q):
         dist.all_gather(recv, peer_blocks)
         assert torch.equal(mine.cpu(), recv[other].cpu())
         dist.barrier()
+        # batched pull: one handshake + one xGMI transfer, partial grant
+        svc.register_blocks(0xB00 + rank, 0, [2, 3])
+        dist.barrier()
+        res = svc.pull_many([(0xB00 + other, 0, [50, 51]),
+                             (0xBAD, 0, [52])],
+                            src_rank=other).result(timeout=60)
+        assert res == [True, False]
+        for i, src_blk in enumerate((2, 3)):
+            got = torch.stack([group[l][50 + i] for l in range(4)])
+            assert torch.equal(got.cpu(), recv[other][src_blk].cpu())
+        dist.barrier()
         svc.close()
         dist.barrier()
         dist.destroy_process_group()

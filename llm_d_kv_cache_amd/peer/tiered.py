@@ -61,15 +61,63 @@ class TieredKVLoader:
                        block_ids: Sequence[int], blocks_per_chunk: int,
                        group: int = 0) -> int:
         """Resolve a prefix chunk by chunk; stops at the first miss (a
-        prefix must be contiguous). Returns the number of chunks filled."""
+        prefix must be contiguous). Returns the number of chunks filled.
+
+        Storage is tried per chunk; on a storage miss the peer tier is
+        asked for the next run of chunks in ONE batched pull_many (one
+        control round trip + one xGMI transfer), and the granted prefix of
+        the batch extends the fill. The first chunk no tier serves ends
+        the walk."""
+        from . import MAX_BATCH
+
         filled = 0
-        for ci, h in enumerate(chunk_hashes):
-            ids = block_ids[ci * blocks_per_chunk:(ci + 1) * blocks_per_chunk]
+        n = len(chunk_hashes)
+
+        def ids_of(ci):
+            return list(
+                block_ids[ci * blocks_per_chunk:(ci + 1) * blocks_per_chunk])
+
+        while filled < n:
+            ids = ids_of(filled)
             if not ids:
                 break
-            if self.resolve(h, ids, group=group) == "miss":
+            if self.load_handler is not None:
+                job = self.load_handler.transfer_async(
+                    [chunk_hashes[filled]], {group: ids})
+                if self._wait(self.load_handler, job):
+                    filled += 1
+                    continue
+            if self.peer is None:
                 break
-            filled += 1
+            batch = []
+            for ci in range(filled, min(n, filled + MAX_BATCH)):
+                bids = ids_of(ci)
+                if not bids:
+                    break
+                batch.append((chunk_hashes[ci], group, bids))
+            if not batch:
+                break
+            granted_prefix = 0
+            for rank in self.peer_ranks:
+                try:
+                    res = self.peer.pull_many(
+                        batch, src_rank=rank, timeout=self.pull_timeout_s
+                    ).result(timeout=self.pull_timeout_s + 5)
+                except Exception as e:
+                    log.warning("peer batch pull from rank %d failed: %s",
+                                rank, e)
+                    continue
+                k = 0
+                while k < len(res) and res[k]:
+                    k += 1
+                if k:
+                    granted_prefix = k
+                    break
+            if granted_prefix == 0:
+                break
+            filled += granted_prefix
+            # a partial grant loops back: the next chunk retries storage
+            # before the peers are asked again
         return filled
 
     def _wait(self, handler, job_id) -> bool:

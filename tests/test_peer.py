@@ -248,4 +248,10 @@ def scenario_tiered_loader(rank, svc, group):
         filled = loader.resolve_prefix([CH_STORAGE, CH_MISS, CH_PEER],
                                        list(range(20, 32)), 4)
         assert filled == 1
+        # mixed tiers resolve across one batched peer attempt:
+        # storage, then a peer run, then storage again, then miss
+        filled = loader.resolve_prefix(
+            [CH_STORAGE, CH_PEER, CH_STORAGE, CH_MISS],
+            list(range(8, 24)), 4)
+        assert filled == 3
     dist.barrier()

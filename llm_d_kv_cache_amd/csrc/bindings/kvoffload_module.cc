@@ -151,6 +151,15 @@ PYBIND11_MODULE(_kvoffload, m) {
            })
       .def("wait_job", &StorageOffloadEngine::wait_job,
            py::call_guard<py::gil_scoped_release>(), py::arg("job_id"))
+      .def(
+          "host_cache_read",
+          [](kvo::StorageOffloadEngine& e, const std::string& path,
+             uintptr_t dst, size_t cap) {
+            py::gil_scoped_release rel;
+            return e.host_cache_read(path, reinterpret_cast<uint8_t*>(dst),
+                                     cap);
+          },
+          py::arg("path"), py::arg("dst"), py::arg("cap"))
       .def("stats", &StorageOffloadEngine::stats,
            py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("pending_writes", &StorageOffloadEngine::pending_writes);

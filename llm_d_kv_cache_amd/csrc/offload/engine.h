@@ -213,6 +213,25 @@ class StorageOffloadEngine {
 
   const EngineConfig& config() const { return cfg_; }
 
+  // Peer/DRAM-tier bridge: copy a resident host-cache entry for `path` into
+  // caller memory so a peer GPU can be served over xGMI even after the
+  // blocks left HBM. Returns payload bytes on hit, -1 on miss, -2 when cap
+  // is too small. The copy (vs handing out the slot) decouples the pinned
+  // slab's lifetime from the caller's async send.
+  int64_t host_cache_read(const std::string& path, uint8_t* dst, size_t cap) {
+    if (!cache_) return -1;
+    HostPinnedCache::Slot* s = cache_->lookup(path);
+    if (s == nullptr) return -1;
+    const size_t n = s->bytes_used;
+    if (n > cap) {
+      cache_->release(s);
+      return -2;
+    }
+    std::memcpy(dst, s->buf->host(), n);
+    cache_->release(s);
+    return static_cast<int64_t>(n);
+  }
+
   // ---- store ----------------------------------------------------------------
 
   int64_t async_store(std::vector<FileTransfer> files, uintptr_t caller_stream) {

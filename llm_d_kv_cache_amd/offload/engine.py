@@ -163,3 +163,21 @@ class TorchOffloadEngine:
 
     def stats(self):
         return self._engine.stats()
+
+    def dram_chunk(self, path: str, n_blocks: int, group: int = 0):
+        """Read a chunk file's payload out of the pinned host-DRAM cache
+        (peer/DRAM-tier bridge). Returns a host uint8 tensor of exactly the
+        first n_blocks' packed bytes, or None on a cache miss / short
+        entry. The payload layout matches the file layout (raw gather
+        layout, or fp8 tile records when serialize="fp8_e4m3")."""
+        import torch
+
+        geo = self.group_geometry[group]
+        want = n_blocks * geo["num_layers"] * geo["record_bytes"]
+        cap = (self.config.gpu_blocks_per_file * geo["num_layers"]
+               * geo["record_bytes"])
+        buf = torch.empty(cap, dtype=torch.uint8)
+        got = self._engine.host_cache_read(path, buf.data_ptr(), cap)
+        if got < want:
+            return None
+        return buf[:want]

@@ -51,6 +51,7 @@ MSG_LEN = 8 + 3 * MAX_BATCH
 class PeerStats:
     pulls_requested: int = 0
     pulls_served: int = 0
+    pulls_served_dram: int = 0
     pulls_failed: int = 0
     bytes_sent: int = 0
     bytes_received: int = 0
@@ -84,10 +85,14 @@ class _OpTracker:
 class PeerMigrationService:
     def __init__(self, groups: Sequence[Sequence], data_group=None,
                  control_group=None, device: Optional[int] = None,
-                 poll_interval_s: float = 0.0002):
+                 poll_interval_s: float = 0.0002, dram_lookup=None):
         """groups: per KV-cache group, a list of per-layer page tensors
         (the offload engine's layout). data_group: nccl(=RCCL) on GPU /
-        gloo on CPU. control_group: gloo. Both must be dedicated groups."""
+        gloo on CPU. control_group: gloo. Both must be dedicated groups.
+        dram_lookup: optional (chunk_hash, group, n_blocks) ->
+        (host_uint8_tensor, is_fp8) | None — lets this rank serve pulls
+        from its pinned host-DRAM cache after HBM eviction (see
+        peer.tiered.make_dram_lookup)."""
         import torch
         import torch.distributed as dist
 
@@ -105,6 +110,7 @@ class PeerMigrationService:
         self.device = device if device is not None else (
             groups[0][0].device.index or 0 if self.gpu_mode else 0)
         self._poll = poll_interval_s
+        self._dram_lookup = dram_lookup
         self._data_needs_wait_thread = not self.gpu_mode  # gloo data plane
 
         native_groups = []
@@ -309,14 +315,27 @@ class PeerMigrationService:
                             for ci, (h, g, ids) in enumerate(st["chunks"]):
                                 if not st["granted"][ci]:
                                     continue
-                                self._copier.scatter(g, ids, base + off,
-                                                     stream)
-                                off += self._copier.packed_bytes(g, len(ids))
+                                if st["fp8_chunks"][ci]:
+                                    self._copier.scatter_fp8(g, ids,
+                                                             base + off,
+                                                             stream)
+                                    off += self._copier.packed_bytes_fp8(
+                                        g, len(ids))
+                                else:
+                                    self._copier.scatter(g, ids, base + off,
+                                                         stream)
+                                    off += self._copier.packed_bytes(
+                                        g, len(ids))
                             result = st["granted"]
                         else:
-                            self._copier.scatter(
-                                st["group"], st["dst_ids"],
-                                st["buf"].data_ptr(), stream)
+                            if st.get("fp8"):
+                                self._copier.scatter_fp8(
+                                    st["group"], st["dst_ids"],
+                                    st["buf"].data_ptr(), stream)
+                            else:
+                                self._copier.scatter(
+                                    st["group"], st["dst_ids"],
+                                    st["buf"].data_ptr(), stream)
                             result = True
                         if self.gpu_mode:
                             self._comm_stream.synchronize()
@@ -355,6 +374,25 @@ class PeerMigrationService:
         }
         self._stats.pulls_requested += 1
 
+    def _serve_dram(self, chunk_hash, group, n_blocks):
+        """DRAM-tier lookup with error isolation: a broken callback must
+        not kill the service loop."""
+        try:
+            got = self._dram_lookup(chunk_hash, group, n_blocks)
+        except Exception as e:
+            import logging
+
+            logging.getLogger(__name__).warning("dram_lookup failed: %s", e)
+            return None
+        if got is None:
+            return None
+        host, fp8 = got
+        want = (self._copier.packed_bytes_fp8(group, n_blocks) if fp8
+                else self._copier.packed_bytes(group, n_blocks))
+        if host.numel() < want:
+            return None
+        return host[:want].contiguous(), fp8
+
     def _start_pull_many(self, payload, pending_pulls) -> None:
         _, chunks, src, deadline, fut = payload
         req_id = self._next_req_id
@@ -385,8 +423,13 @@ class PeerMigrationService:
         if op == OP_PULL_REQ:
             ids = self.lookup_local(chunk_hash, group)
             grant = ids is not None and len(ids) == n_blocks
+            dram = None
+            if not grant and self._dram_lookup is not None:
+                dram = self._serve_dram(chunk_hash, group, n_blocks)
+            ok_code = 1 if grant else (0 if dram is None
+                                       else (2 if dram[1] else 1))
             dist.send(self._msg(OP_PULL_ACK, req_id, group, chunk_hash,
-                                n_blocks, 1 if grant else 0),
+                                n_blocks, ok_code),
                       dst=sender, group=self.control_group)
             if grant:
                 slab = self._data_tensor(group, n_blocks)
@@ -397,12 +440,28 @@ class PeerMigrationService:
                 pending_sends.append((self._track(w), slab))
                 self._stats.pulls_served += 1
                 self._stats.bytes_sent += slab.numel()
+            elif dram is not None:
+                slab = dram[0].cuda() if self.gpu_mode else dram[0]
+                w = dist.isend(slab, dst=sender, group=self.data_group)
+                pending_sends.append((self._track(w), slab))
+                self._stats.pulls_served += 1
+                self._stats.pulls_served_dram += 1
+                self._stats.bytes_sent += slab.numel()
         elif op == OP_PULL_ACK:
             st = pending_pulls.get(req_id)
             if st is None:
                 return
             if ok:
-                slab = self._data_tensor(st["group"], len(st["dst_ids"]))
+                st["fp8"] = ok == 2
+                if st["fp8"]:
+                    nb = self._copier.packed_bytes_fp8(st["group"],
+                                                       len(st["dst_ids"]))
+                    slab = (self._torch.empty(nb, dtype=self._torch.uint8,
+                                              device="cuda")
+                            if self.gpu_mode else
+                            self._torch.empty(nb, dtype=self._torch.uint8))
+                else:
+                    slab = self._data_tensor(st["group"], len(st["dst_ids"]))
                 st["buf"] = slab
                 w = dist.irecv(slab, src=st["src"], group=self.data_group)
                 st["tracker"] = self._track(w)
@@ -419,17 +478,30 @@ class PeerMigrationService:
                 nb = int(buf[10 + 3 * i])
                 reqs.append((h, g, nb))
             mask = 0
+            fp8_mask = 0
             total = 0
-            served = []
+            served = []  # ("hbm", g, ids) | ("dram", host_tensor)
+            n_dram = 0
             for i, (h, g, nb) in enumerate(reqs):
                 ids = self.lookup_local(h, g)
                 if ids is not None and len(ids) == nb:
                     mask |= 1 << i
                     total += self._copier.packed_bytes(g, nb)
-                    served.append((g, ids))
+                    served.append(("hbm", g, ids))
+                    continue
+                if self._dram_lookup is not None:
+                    dram = self._serve_dram(h, g, nb)
+                    if dram is not None:
+                        mask |= 1 << i
+                        if dram[1]:
+                            fp8_mask |= 1 << i
+                        total += dram[0].numel()
+                        served.append(("dram", dram[0], None))
+                        n_dram += 1
             ack = self._msg(OP_PULL_ACK_MULTI, req_id)
             ack[3] = n_chunks
             ack[6] = mask
+            ack[7] = fp8_mask
             dist.send(ack, dst=sender, group=self.control_group)
             if mask:
                 slab = (self._torch.empty(total, dtype=self._torch.uint8,
@@ -439,24 +511,39 @@ class PeerMigrationService:
                 off = 0
                 stream = (self._comm_stream.cuda_stream
                           if self.gpu_mode else 0)
-                for g, ids in served:
-                    self._copier.gather(g, ids, slab.data_ptr() + off, stream)
-                    off += self._copier.packed_bytes(g, len(ids))
+                for item in served:
+                    if item[0] == "hbm":
+                        _, g, ids = item
+                        self._copier.gather(g, ids, slab.data_ptr() + off,
+                                            stream)
+                        off += self._copier.packed_bytes(g, len(ids))
+                    else:
+                        host = item[1]
+                        slab[off:off + host.numel()].copy_(host)
+                        off += host.numel()
                 w = dist.isend(slab, dst=sender, group=self.data_group)
                 pending_sends.append((self._track(w), slab))
                 self._stats.pulls_served += len(served)
+                self._stats.pulls_served_dram += n_dram
                 self._stats.bytes_sent += slab.numel()
         elif op == OP_PULL_ACK_MULTI:
             st = pending_pulls.get(req_id)
             if st is None:
                 return
             mask = ok
+            fp8_mask = int(buf[7])
             granted = [bool(mask >> i & 1) for i in range(len(st["chunks"]))]
+            fp8_chunks = [bool(fp8_mask >> i & 1)
+                          for i in range(len(st["chunks"]))]
             st["granted"] = granted
+            st["fp8_chunks"] = fp8_chunks
             if mask:
-                total = sum(self._copier.packed_bytes(g, len(ids))
-                            for gi, (h, g, ids) in enumerate(st["chunks"])
-                            if granted[gi])
+                total = sum(
+                    (self._copier.packed_bytes_fp8(g, len(ids))
+                     if fp8_chunks[gi] else
+                     self._copier.packed_bytes(g, len(ids)))
+                    for gi, (h, g, ids) in enumerate(st["chunks"])
+                    if granted[gi])
                 slab = (self._torch.empty(total, dtype=self._torch.uint8,
                                           device="cuda") if self.gpu_mode
                         else self._torch.empty(total,

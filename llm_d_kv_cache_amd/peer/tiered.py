@@ -20,6 +20,23 @@ from typing import Dict, List, Optional, Sequence
 log = logging.getLogger(__name__)
 
 
+def make_dram_lookup(engine, mapper):
+    """Build a PeerMigrationService dram_lookup callback from an offload
+    engine + file mapper: serves PULL requests out of the pinned host-DRAM
+    cache when the blocks are no longer registered in HBM. Returns
+    (host_tensor, is_fp8) or None."""
+    fp8 = engine.config.serialize == "fp8_e4m3"
+
+    def lookup(chunk_hash, group, n_blocks):
+        path = mapper.file_name(chunk_hash, group)
+        t = engine.dram_chunk(path, n_blocks, group)
+        if t is None:
+            return None
+        return t, fp8
+
+    return lookup
+
+
 class TieredKVLoader:
     def __init__(self, load_handler=None, peer_service=None,
                  peer_ranks: Sequence[int] = (), pull_timeout_s: float = 30.0):

@@ -160,6 +160,90 @@ def scenario_multi_group(rank, svc, group):
     dist.barrier()
 
 
+def scenario_dram_tier(rank, svc, group):
+    """Peer serving from the pinned host-DRAM cache: rank 0 offloads chunks
+    (write-through populates the cache), never registers them in HBM, and
+    deletes the files — rank 1's pulls are served from DRAM over the data
+    plane, including an fp8-serialized group in a mixed batched pull."""
+    import os as _os
+    import tempfile
+    import time
+
+    import torch.distributed as dist
+
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        TorchOffloadEngine,
+    )
+    from llm_d_kv_cache_amd.peer.tiered import make_dram_lookup
+
+    # make group_b rows valid bf16 payloads for the fp8 codec
+    torch.manual_seed(900 + rank)
+    group_b = svc._tensors[1]
+    vals = (torch.randn(32, 1024) * 2).to(torch.bfloat16)
+    group_b[0].copy_(vals.view(torch.uint8).reshape(32, 2048))
+
+    CH_RAW, CH_FP8, CH_MISS = 0x61, 0x62, 0xBAD
+    if rank == 0:
+        root = tempfile.mkdtemp(prefix="dramtier0_")
+        groups = [svc._tensors[0], group_b]
+        eng_raw = TorchOffloadEngine(groups, OffloadEngineConfig(
+            io_threads=2, gpu_blocks_per_file=4, copy_path="host",
+            host_cache_bytes=64 << 20))
+        eng_fp8 = TorchOffloadEngine(groups, OffloadEngineConfig(
+            io_threads=2, gpu_blocks_per_file=4, copy_path="host",
+            host_cache_bytes=64 << 20, serialize="fp8_e4m3"))
+        m_raw = FileMapper(root + "/raw", KVCacheLayoutConfig(model="dt"))
+        m_fp8 = FileMapper(root + "/fp8", KVCacheLayoutConfig(model="dt8"))
+        st_raw = GPUToStorageHandler(eng_raw, m_raw, [4, 4])
+        st_fp8 = GPUToStorageHandler(eng_fp8, m_fp8, [4, 4])
+        st_raw.transfer_async([CH_RAW], {0: [0, 1, 2, 3]})
+        st_fp8.transfer_async([CH_FP8], {1: [0, 1, 2, 3]})
+        for h in (st_raw, st_fp8):
+            deadline = time.time() + 10
+            while not h.get_finished() and time.time() < deadline:
+                time.sleep(0.01)
+        # prove DRAM (not fs) serves: remove the files
+        _os.remove(m_raw.file_name(CH_RAW, 0))
+        _os.remove(m_fp8.file_name(CH_FP8, 1))
+        lk_raw = make_dram_lookup(eng_raw, m_raw)
+        lk_fp8 = make_dram_lookup(eng_fp8, m_fp8)
+        svc._dram_lookup = (lambda h, g, n:
+                            lk_raw(h, g, n) if g == 0 else lk_fp8(h, g, n))
+        golden = svc._tensors[0][0][[0, 1, 2, 3]].clone()
+        dist.broadcast(golden, src=0)
+        golden_b = vals[[0, 1, 2, 3]].float()
+        dist.broadcast(golden_b, src=0)
+        dist.barrier()
+        dist.barrier()
+        assert svc.stats().pulls_served_dram >= 3
+    else:
+        golden = torch.zeros(4, 4096, dtype=torch.uint8)
+        dist.broadcast(golden, src=0)
+        golden_b = torch.zeros(4, 1024)
+        dist.broadcast(golden_b, src=0)
+        # single pull from DRAM
+        ok = svc.pull(CH_RAW, 0, [10, 11, 12, 13], src_rank=0).result(
+            timeout=60)
+        assert ok is True
+        assert torch.equal(svc._tensors[0][0][[10, 11, 12, 13]], golden)
+        # mixed batched pull: raw DRAM + fp8 DRAM + miss
+        res = svc.pull_many(
+            [(CH_RAW, 0, [20, 21, 22, 23]),
+             (CH_FP8, 1, [8, 9, 10, 11]),
+             (CH_MISS, 0, [24])], src_rank=0).result(timeout=60)
+        assert res == [True, True, False]
+        assert torch.equal(svc._tensors[0][0][[20, 21, 22, 23]], golden)
+        got = group_b[0][[8, 9, 10, 11]].view(torch.bfloat16).float()
+        amax = golden_b.abs().amax()
+        assert (got - golden_b).abs().max() <= 0.07 * amax
+        dist.barrier()
+        dist.barrier()
+
+
 # ---- tests ------------------------------------------------------------------
 
 @pytest.mark.parametrize("scenario", [
@@ -168,6 +252,7 @@ def scenario_multi_group(rank, svc, group):
     "scenario_bidirectional",
     "scenario_multi_group",
     "scenario_batched_pull",
+    "scenario_dram_tier",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):

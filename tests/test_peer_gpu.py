@@ -69,6 +69,43 @@ def _run_rank(rank, world, init_file, q):
             got = torch.stack([group[l][50 + i] for l in range(4)])
             assert torch.equal(got.cpu(), recv[other][src_blk].cpu())
         dist.barrier()
+        # DRAM-tier serving: offload a chunk (write-through populates the
+        # pinned host cache), delete the file, never register it in HBM —
+        # the pull must be served from the peer's DRAM cache.
+        import tempfile
+        import time as _time
+
+        from llm_d_kv_cache_amd.offload import (
+            FileMapper,
+            GPUToStorageHandler,
+            KVCacheLayoutConfig,
+            OffloadEngineConfig,
+            TorchOffloadEngine,
+        )
+        from llm_d_kv_cache_amd.peer.tiered import make_dram_lookup
+
+        eng = TorchOffloadEngine([group], OffloadEngineConfig(
+            io_threads=2, gpu_blocks_per_file=4, copy_path="staged",
+            host_cache_bytes=256 << 20, device=rank))
+        root = tempfile.mkdtemp(prefix=f"dram_gpu_{rank}_")
+        mapper = FileMapper(root, KVCacheLayoutConfig(model="dtg"))
+        store = GPUToStorageHandler(eng, mapper, [4])
+        CH_DRAM = 0xD0 + rank
+        store.transfer_async([CH_DRAM], {0: [4, 5, 6, 7]})
+        deadline = _time.time() + 20
+        while not store.get_finished() and _time.time() < deadline:
+            _time.sleep(0.01)
+        os.remove(mapper.file_name(CH_DRAM, 0))
+        svc._dram_lookup = make_dram_lookup(eng, mapper)
+        dist.barrier()
+        ok = svc.pull(0xD0 + other, 0, [56, 57, 58, 59],
+                      src_rank=other).result(timeout=60)
+        assert ok is True
+        for i, src_blk in enumerate((4, 5, 6, 7)):
+            got = torch.stack([group[l][56 + i] for l in range(4)])
+            assert torch.equal(got.cpu(), recv[other][src_blk].cpu())
+        assert svc.stats().pulls_served_dram >= 1
+        dist.barrier()
         svc.close()
         dist.barrier()
         dist.destroy_process_group()

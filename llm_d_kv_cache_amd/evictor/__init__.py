@@ -6,8 +6,9 @@ hash-prefix shard directories walk the content-addressed layout and queue
 cold files (atime threshold), one activator toggles deletion on a
 filesystem-utilization hysteresis (default 85% on / 70% off), and one
 deleter batch-unlinks queued files, optionally publishing BlockRemoved
-storage events so the global indexer drops the entries. The coordinator
-restarts failed children.
+storage events so the global indexer drops the entries; a folder cleaner
+prunes directories the deleter emptied. The coordinator restarts failed
+children.
 """
 from __future__ import annotations
 
@@ -31,6 +32,7 @@ class EvictorConfig:
     deactivate_utilization: float = 0.70   # stop below this
     check_interval_s: float = 5.0
     delete_batch: int = 64
+    clean_interval_s: float = 300.0        # empty-dir pruning cadence
     # ZMQ endpoint to publish BlockRemoved storage events (optional)
     events_endpoint: Optional[str] = None
     events_model: str = ""
@@ -152,6 +154,37 @@ def deleter_proc(cfg: EvictorConfig, candidates: mp.Queue, active: "mp.Event",
             publisher.publish_block_removed(removed_hashes)
 
 
+def clean_empty_dirs(root: str) -> int:
+    """Remove empty shard/sub directories left behind by deletions
+    (reference folder_cleaner process). Run directories themselves stay:
+    they hold the layout manifest (config.json). Returns dirs removed."""
+    removed = 0
+    for run_dir in _list_dirs(root):
+        for shard_dir in _list_dirs(run_dir):
+            for sub in _list_dirs(shard_dir):
+                try:
+                    os.rmdir(sub)  # fails (correctly) unless empty
+                    removed += 1
+                except OSError:
+                    pass
+            try:
+                os.rmdir(shard_dir)
+                removed += 1
+            except OSError:
+                pass
+    return removed
+
+
+def folder_cleaner_proc(cfg: EvictorConfig, stop: "mp.Event") -> None:
+    while not stop.is_set():
+        stop.wait(cfg.clean_interval_s)
+        if stop.is_set():
+            return
+        n = clean_empty_dirs(cfg.root)
+        if n:
+            log.info("evictor: pruned %d empty directories", n)
+
+
 def _list_dirs(path):
     try:
         return [os.path.join(path, d) for d in sorted(os.listdir(path))
@@ -196,6 +229,8 @@ class PvcEvictor:
         spec["deleter"] = (deleter_proc,
                            (self.cfg, self.candidates, self.active,
                             self.stop_event, self.deleted))
+        spec["folder-cleaner"] = (folder_cleaner_proc,
+                                  (self.cfg, self.stop_event))
         return spec
 
     def start(self) -> None:

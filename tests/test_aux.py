@@ -230,3 +230,18 @@ def test_prometheus_offload_engine_metrics(tmp_path):
     text = generate_latest(reg).decode()
     assert "kv_offload_files_written_total 1.0" in text
     assert "kv_offload_bytes_stored_total 4096.0" in text
+
+
+def test_evictor_folder_cleaner(tmp_path):
+    from llm_d_kv_cache_amd.evictor import clean_empty_dirs
+
+    run = tmp_path / "m_abc_r0"
+    (run / "0aa" / "bb_g0").mkdir(parents=True)      # empty chain -> pruned
+    (run / "0cc" / "dd_g0").mkdir(parents=True)
+    (run / "0cc" / "dd_g0" / "x.bin").write_bytes(b"1")  # non-empty -> kept
+    (run / "config.json").write_text("{}")
+    removed = clean_empty_dirs(str(tmp_path))
+    assert removed == 2  # 0aa/bb_g0 and then 0aa itself
+    assert not (run / "0aa").exists()
+    assert (run / "0cc" / "dd_g0" / "x.bin").exists()
+    assert (run / "config.json").exists()

@@ -361,8 +361,44 @@ class PeerMigrationService:
             if not st["fut"].done():
                 st["fut"].set_exception(RuntimeError("peer service closed"))
 
+    def _local_serve(self, chunk_hash, group, dst_ids):
+        """Self-pull short circuit (RCCL cannot send to self): resolve from
+        local HBM registry or the DRAM cache straight into the dst pages."""
+        stream = self._comm_stream.cuda_stream if self.gpu_mode else 0
+        ids = self.lookup_local(chunk_hash, group)
+        if ids is not None and len(ids) == len(dst_ids):
+            slab = self._data_tensor(group, len(dst_ids))
+            self._copier.gather(group, ids, slab.data_ptr(), stream)
+            self._copier.scatter(group, dst_ids, slab.data_ptr(), stream)
+            self._stats.pulls_served += 1
+            return True
+        if self._dram_lookup is not None:
+            dram = self._serve_dram(chunk_hash, group, len(dst_ids))
+            if dram is not None:
+                host, fp8 = dram
+                buf = host.cuda() if self.gpu_mode else host
+                if fp8:
+                    self._copier.scatter_fp8(group, dst_ids, buf.data_ptr(),
+                                             stream)
+                else:
+                    self._copier.scatter(group, dst_ids, buf.data_ptr(),
+                                         stream)
+                self._stats.pulls_served += 1
+                self._stats.pulls_served_dram += 1
+                return True
+        return False
+
     def _start_pull(self, payload, pending_pulls) -> None:
         _, chunk_hash, group, dst_ids, src, deadline, fut = payload
+        if src == self.rank:
+            ok = self._local_serve(chunk_hash, group, dst_ids)
+            if self.gpu_mode:
+                self._comm_stream.synchronize()
+            if not ok:
+                self._stats.pulls_failed += 1
+            self._stats.pulls_requested += 1
+            fut.set_result(ok)
+            return
         req_id = self._next_req_id
         self._next_req_id += self.world
         self._dist.send(self._msg(OP_PULL_REQ, req_id, group, chunk_hash,
@@ -395,6 +431,14 @@ class PeerMigrationService:
 
     def _start_pull_many(self, payload, pending_pulls) -> None:
         _, chunks, src, deadline, fut = payload
+        if src == self.rank:
+            res = [self._local_serve(h, g, ids) for (h, g, ids) in chunks]
+            if self.gpu_mode:
+                self._comm_stream.synchronize()
+            self._stats.pulls_requested += len(chunks)
+            self._stats.pulls_failed += sum(1 for r in res if not r)
+            fut.set_result(res)
+            return
         req_id = self._next_req_id
         self._next_req_id += self.world
         t = self._msg(OP_PULL_REQ_MULTI, req_id)

@@ -148,6 +148,24 @@ def scenario_batched_pull(rank, svc, group):
         dist.barrier()
 
 
+def scenario_self_pull(rank, svc, group):
+    """src_rank == own rank takes the local short circuit (RCCL cannot
+    send to self): HBM hit, miss, and batched variants."""
+    import torch.distributed as dist
+
+    svc.register_blocks(0x31 + rank, 0, [1, 2])
+    golden = [group[0][[1, 2]].clone(), group[1][[1, 2]].clone()]
+    ok = svc.pull(0x31 + rank, 0, [28, 29], src_rank=rank).result(timeout=60)
+    assert ok is True
+    assert torch.equal(group[0][[28, 29]], golden[0])
+    assert torch.equal(group[1][[28, 29]], golden[1])
+    assert svc.pull(0x777, 0, [30], src_rank=rank).result(timeout=60) is False
+    res = svc.pull_many([(0x31 + rank, 0, [24, 25]), (0x778, 0, [26])],
+                        src_rank=rank).result(timeout=60)
+    assert res == [True, False]
+    dist.barrier()
+
+
 def scenario_multi_group(rank, svc, group):
     import torch.distributed as dist
 
@@ -253,6 +271,7 @@ def scenario_dram_tier(rank, svc, group):
     "scenario_multi_group",
     "scenario_batched_pull",
     "scenario_dram_tier",
+    "scenario_self_pull",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):

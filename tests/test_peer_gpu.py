@@ -1,20 +1,16 @@
 """GPU twin of test_peer.py: RCCL(=nccl) data plane over xGMI.
 
-Needs >= 2 GPUs; skipped on single-GPU boxes (the driver's scaling tier
-exercises the multi-GPU path via bench.py --gpus N)."""
+On a single-GPU box this runs world=1 self-pull (RCCL send/recv to self,
+same code path minus the xGMI hop); >= 2 GPUs exercises the real
+cross-device transfer. The driver's scaling tier also covers the
+multi-GPU path via bench.py --gpus N."""
 import multiprocessing as mp
 import os
 
 import pytest
 import torch
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(
-        not torch.cuda.is_available() or torch.cuda.device_count() < 2,
-        reason="needs >= 2 GPUs",
-    ),
-]
+pytestmark = pytest.mark.gpu
 
 
 def _run_rank(rank, world, init_file, q):

@@ -52,15 +52,12 @@ class Indexer {
     res.total_blocks = keys.size();
     if (keys.empty()) return res;
 
+    // Intern (not find): in a multi-replica deployment this process may
+    // score pods whose entries were ingested by ANOTHER process into a
+    // shared backend (Redis/Valkey) — an unknown-here name must still
+    // filter-match entries the backend lookup interns on decode.
     std::unordered_set<uint32_t> filter;
-    for (const auto& p : pods) {
-      uint32_t id = index_->strings().find(p);
-      if (id != StringTable::kInvalid) filter.insert(id);
-      // A named pod the index has never seen can match nothing; when *every*
-      // pod is unknown the filter stays non-empty-intent but empty-content —
-      // represent that as a sentinel below.
-    }
-    if (!pods.empty() && filter.empty()) return res;  // no known pods
+    for (const auto& p : pods) filter.insert(index_->strings().intern(p));
 
     auto hits = index_->lookup(keys, filter);
     res.hit_blocks = hits.size();

@@ -128,3 +128,25 @@ def test_redis_backend_drives_indexer_and_pool(redis_pair):
                  encode_batch([block_stored_payload([1, 2], None, tokens, 16)]))
     res = ix.score_tokens(tokens, "m", [])
     assert res.scores == {"pod-r": 2.0}
+
+
+def test_redis_cross_process_pod_filter(redis_pair):
+    """Multi-replica shape: one indexer process ingests into the shared
+    backend, a DIFFERENT process (fresh string table) scores with an
+    explicit candidate list — the filter must match entries it has never
+    interned locally."""
+    from llm_d_kv_cache_amd.core import (
+        IndexerConfig,
+        KVCacheIndexer,
+        RedisIndexConfig,
+    )
+
+    srv, _ = redis_pair
+    cfg = lambda: IndexerConfig(  # noqa: E731
+        redis_index=RedisIndexConfig(host="127.0.0.1", port=srv.port))
+    writer = KVCacheIndexer(cfg())
+    keys = writer.compute_block_keys(list(range(64)), "m")
+    writer.index.add([], keys, [entry("pod-x")])
+    reader = KVCacheIndexer(cfg())
+    assert reader.score_tokens(list(range(64)), "m",
+                               ["pod-x", "pod-absent"]) == {"pod-x": 4.0}

@@ -219,8 +219,10 @@ class EventPool {
           handle_removed(pod, ev.removed);
           break;
         case EventType::kAllBlocksCleared: {
-          uint32_t pid = index_->strings().find(pod);
-          if (pid != StringTable::kInvalid) index_->clear(pid);
+          // Intern (not find): with a shared backend (Redis/Valkey) this
+          // replica may be asked to clear a pod whose stores another
+          // process wrote.
+          index_->clear(index_->strings().intern(pod));
           break;
         }
       }

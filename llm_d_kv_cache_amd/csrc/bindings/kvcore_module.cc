@@ -150,17 +150,12 @@ PYBIND11_MODULE(_kvcore, m) {
           "lookup",
           [](IndexBackend& idx, const std::vector<uint64_t>& keys,
              const std::vector<std::string>& pods) {
+            // Intern (not find): shared backends hold entries written by
+            // other processes whose pod names this one never saw.
             std::unordered_set<uint32_t> filter;
-            bool any_unknown_only = !pods.empty();
-            for (const auto& p : pods) {
-              uint32_t id = idx.strings().find(p);
-              if (id != StringTable::kInvalid) {
-                filter.insert(id);
-                any_unknown_only = false;
-              }
-            }
+            for (const auto& p : pods) filter.insert(idx.strings().intern(p));
             std::vector<std::pair<uint64_t, std::vector<PodEntry>>> hits;
-            if (!any_unknown_only) {
+            {
               py::gil_scoped_release rel;
               hits = idx.lookup(keys, filter);
             }
@@ -215,8 +210,7 @@ PYBIND11_MODULE(_kvcore, m) {
            })
       .def("clear",
            [](IndexBackend& idx, const std::string& pod) {
-             uint32_t id = idx.strings().find(pod);
-             if (id == StringTable::kInvalid) return;
+             uint32_t id = idx.strings().intern(pod);
              py::gil_scoped_release rel;
              idx.clear(id);
            })

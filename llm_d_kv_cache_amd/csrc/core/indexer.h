@@ -63,10 +63,8 @@ class Indexer {
     res.hit_blocks = hits.size();
     std::unordered_map<uint32_t, int32_t> wh;
     if (window_hints != nullptr) {
-      for (const auto& [pod, blocks] : *window_hints) {
-        uint32_t id = index_->strings().find(pod);
-        if (id != StringTable::kInvalid) wh.emplace(id, blocks);
-      }
+      for (const auto& [pod, blocks] : *window_hints)
+        wh.emplace(index_->strings().intern(pod), blocks);
     }
     auto scores = scorer_->score(keys, hits, wh.empty() ? nullptr : &wh);
     for (const auto& [pod_id, s] : scores)

@@ -245,3 +245,26 @@ def test_evictor_folder_cleaner(tmp_path):
     assert not (run / "0aa").exists()
     assert (run / "0cc" / "dd_g0" / "x.bin").exists()
     assert (run / "config.json").exists()
+
+
+def test_routing_sim_precise_beats_baselines():
+    """The routing-value experiment (reference benchmarking shape) driven
+    by the real native scoring + event-ingestion paths: precise scheduling
+    must cut TTFT and raise the cached-prefix fraction vs load/random."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parents[1] / "tools"))
+    import routing_sim
+
+    out = routing_sim.main([
+        "--requests", "300", "--qps", "18", "--groups", "12",
+        "--prefix-tokens", "1024", "--question-tokens", "128",
+        "--output-tokens", "30", "--prefill-tok-s", "3000",
+        "--capacity-blocks", "512",
+    ])
+    res = {r["scheduler"]: r for r in out["results"]}
+    assert res["precise"]["cached_prefix_frac"] > \
+        1.5 * res["random"]["cached_prefix_frac"]
+    assert res["precise"]["ttft_mean_s"] < res["load"]["ttft_mean_s"] / 1.5
+    assert res["precise"]["ttft_mean_s"] < res["random"]["ttft_mean_s"] / 3

@@ -24,7 +24,14 @@ def _msg(fdp, name, fields):
         f.name = fname
         f.number = num
         f.type = ftype
-        f.label = label
+        if label == "POPT":  # proto3 `optional` (explicit presence)
+            f.label = _F.LABEL_OPTIONAL
+            f.proto3_optional = True
+            od = m.oneof_decl.add()
+            od.name = "_" + fname
+            f.oneof_index = len(m.oneof_decl) - 1
+        else:
+            f.label = label
         if type_name:
             f.type_name = type_name
     return m
@@ -91,6 +98,7 @@ def _build():
     _msg(tok, "ChatMessage", [
         ("role", 1, S, OPT, None),
         ("content", 2, S, OPT, None),
+        ("content_parts", 3, M, REP, ".tokenizerpb.ContentPart"),
     ])
     _msg(tok, "PlaceholderRange", [
         ("offset", 1, U64, OPT, None),
@@ -111,6 +119,38 @@ def _build():
         ("rendered", 2, S, OPT, None),
         ("mm_features", 3, M, OPT, ".tokenizerpb.MultiModalFeatures"),
         ("error", 4, S, OPT, None),
+    ])
+    _msg(tok, "ImageUrl", [
+        ("url", 1, S, OPT, None),
+    ])
+    _msg(tok, "ContentPart", [
+        ("type", 1, S, OPT, None),
+        ("text", 2, S, OPT, None),
+        ("image_url", 3, M, OPT, ".tokenizerpb.ImageUrl"),
+    ])
+    _msg(tok, "RenderChatCompletionRequest", [
+        ("model_name", 1, S, OPT, None),
+        ("messages", 2, M, REP, ".tokenizerpb.ChatMessage"),
+        ("tools_json", 3, S, OPT, None),
+        ("chat_template", 4, S, OPT, None),
+        ("add_generation_prompt", 5, B, "POPT", None),
+        ("continue_final_message", 6, B, OPT, None),
+        ("chat_template_kwargs", 7, S, OPT, None),
+    ])
+    _msg(tok, "RenderChatCompletionResponse", [
+        ("request_id", 1, S, OPT, None),
+        ("token_ids", 2, U32, REP, None),
+        ("features", 3, M, OPT, ".tokenizerpb.MultiModalFeatures"),
+        ("error", 4, S, OPT, None),
+    ])
+    _msg(tok, "RenderCompletionRequest", [
+        ("model_name", 1, S, OPT, None),
+        ("prompt", 2, S, OPT, None),
+    ])
+    _msg(tok, "RenderCompletionResponse", [
+        ("request_id", 1, S, OPT, None),
+        ("token_ids", 2, U32, REP, None),
+        ("error", 3, S, OPT, None),
     ])
 
     for fdp in (idx, tok):

@@ -52,6 +52,20 @@ class UdsTokenizerClient:
             request_serializer=g("tokenizerpb.RenderChatRequest").SerializeToString,
             response_deserializer=g("tokenizerpb.RenderChatResponse").FromString,
         )
+        self._render_cc = self._channel.unary_unary(
+            f"/{svc}/RenderChatCompletion",
+            request_serializer=g(
+                "tokenizerpb.RenderChatCompletionRequest").SerializeToString,
+            response_deserializer=g(
+                "tokenizerpb.RenderChatCompletionResponse").FromString,
+        )
+        self._render_c = self._channel.unary_unary(
+            f"/{svc}/RenderCompletion",
+            request_serializer=g(
+                "tokenizerpb.RenderCompletionRequest").SerializeToString,
+            response_deserializer=g(
+                "tokenizerpb.RenderCompletionResponse").FromString,
+        )
 
     def initialize(self, model_name: str, tokenizer_path: str = "") -> bool:
         Req = proto.get("tokenizerpb.InitializeTokenizerRequest")
@@ -88,6 +102,50 @@ class UdsTokenizerClient:
         if resp.error:
             raise RuntimeError(resp.error)
         return list(resp.token_ids), resp.rendered
+
+    def render_chat_completion(self, model_name: str, messages,
+                               add_generation_prompt: Optional[bool] = None,
+                               continue_final_message: bool = False,
+                               chat_template: str = "",
+                               chat_template_kwargs: str = ""):
+        """OpenAI chat-completion render. messages: dicts with role +
+        content (str) or content_parts (list of {type, text|image_url}).
+        Returns (request_id, token_ids, mm_hashes, placeholder_ranges)."""
+        Req = proto.get("tokenizerpb.RenderChatCompletionRequest")
+        Msg = proto.get("tokenizerpb.ChatMessage")
+        Part = proto.get("tokenizerpb.ContentPart")
+        msgs = []
+        for m in messages:
+            pm = Msg(role=m.get("role", "user"), content=m.get("content", ""))
+            for part in m.get("content_parts", []) or []:
+                p = Part(type=part.get("type", "text"),
+                         text=part.get("text", "") or "")
+                if part.get("image_url"):
+                    p.image_url.url = part["image_url"].get("url", "")
+                pm.content_parts.append(p)
+            msgs.append(pm)
+        req = Req(model_name=model_name, messages=msgs,
+                  continue_final_message=continue_final_message,
+                  chat_template=chat_template,
+                  chat_template_kwargs=chat_template_kwargs)
+        if add_generation_prompt is not None:
+            req.add_generation_prompt = add_generation_prompt
+        resp = self._render_cc(req, timeout=max(self._timeout, 30.0))
+        if resp.error:
+            raise RuntimeError(resp.error)
+        hashes = list(resp.features.mm_hashes)
+        ranges = [(r.offset, r.length)
+                  for r in resp.features.placeholder_ranges]
+        return resp.request_id, list(resp.token_ids), hashes, ranges
+
+    def render_completion(self, model_name: str, prompt: str):
+        """OpenAI completion render: (request_id, token_ids)."""
+        Req = proto.get("tokenizerpb.RenderCompletionRequest")
+        resp = self._render_c(Req(model_name=model_name, prompt=prompt),
+                              timeout=self._timeout)
+        if resp.error:
+            raise RuntimeError(resp.error)
+        return resp.request_id, list(resp.token_ids)
 
     def close(self):
         self._channel.close()

@@ -62,6 +62,75 @@ class TokenizerManager:
         ids = tok(rendered, add_special_tokens=False)["input_ids"]
         return ids, rendered
 
+    IMAGE_MARKER = "<image>"
+
+    def render_chat_completion(self, model_name: str, messages,
+                               add_generation_prompt: bool = True,
+                               continue_final_message: bool = False,
+                               chat_template: str = "",
+                               chat_template_kwargs: str = ""):
+        """OpenAI chat-completion render (reference RenderChatCompletion,
+        tokenizer.proto:152-171): multimodal content parts are flattened —
+        each image_url part becomes an IMAGE_MARKER in the text whose
+        token span is reported as its placeholder range, and its mm hash
+        is the sha256 of the image reference (content-addressed, matching
+        what the engine's extra-keys carry). Text-only requests behave
+        like render_chat."""
+        import hashlib
+        import json as _json
+
+        tok = self._load(model_name)
+        flat = []
+        mm_hashes = []
+        for m in messages:
+            parts = m.get("content_parts")
+            if parts:
+                pieces = []
+                for part in parts:
+                    if part.get("type") == "image_url":
+                        url = (part.get("image_url") or {}).get("url", "")
+                        mm_hashes.append(
+                            hashlib.sha256(url.encode()).hexdigest()[:32])
+                        pieces.append(self.IMAGE_MARKER)
+                    else:
+                        pieces.append(part.get("text") or "")
+                flat.append({"role": m["role"], "content": "".join(pieces)})
+            else:
+                flat.append({"role": m["role"],
+                             "content": m.get("content") or ""})
+        kwargs = {}
+        if chat_template_kwargs:
+            kwargs = _json.loads(chat_template_kwargs)
+        rendered = tok.apply_chat_template(
+            flat, tokenize=False,
+            add_generation_prompt=add_generation_prompt,
+            continue_final_message=continue_final_message,
+            **({"chat_template": chat_template} if chat_template else {}),
+            **kwargs)
+        enc = tok(rendered, add_special_tokens=False,
+                  return_offsets_mapping=True)
+        ids = enc["input_ids"]
+        offsets = enc.get("offset_mapping") or []
+        ranges = []
+        start = 0
+        for _ in range(len(mm_hashes)):
+            pos = rendered.find(self.IMAGE_MARKER, start)
+            if pos < 0:
+                break
+            end = pos + len(self.IMAGE_MARKER)
+            span = [i for i, (a, b) in enumerate(offsets)
+                    if a < end and b > pos]
+            if span:
+                ranges.append((span[0], span[-1] - span[0] + 1))
+            start = end
+        return ids, mm_hashes, ranges
+
+    def render_completion(self, model_name: str, prompt: str):
+        """OpenAI completion render (reference RenderCompletion): plain
+        prompt validation + tokenization."""
+        tok = self._load(model_name)
+        return tok(prompt, add_special_tokens=True)["input_ids"]
+
     async def run(self, fn, *args):
         return await asyncio.get_running_loop().run_in_executor(
             self._pool, fn, *args)
@@ -112,6 +181,55 @@ class TokenizationServicer:
             log.warning("render_chat failed: %s", e)
             return Resp(error=str(e))
 
+    async def render_chat_completion(self, req, ctx):
+        import uuid
+
+        Resp = proto.get("tokenizerpb.RenderChatCompletionResponse")
+        MM = proto.get("tokenizerpb.MultiModalFeatures")
+        PR = proto.get("tokenizerpb.PlaceholderRange")
+        try:
+            messages = []
+            for m in req.messages:
+                d = {"role": m.role, "content": m.content}
+                if m.content_parts:
+                    d["content_parts"] = [
+                        {"type": p.type, "text": p.text,
+                         "image_url": {"url": p.image_url.url}}
+                        for p in m.content_parts
+                    ]
+                messages.append(d)
+            ids, hashes, ranges = await self.m.run(
+                lambda: self.m.render_chat_completion(
+                    req.model_name, messages,
+                    not req.HasField("add_generation_prompt")
+                    or req.add_generation_prompt,
+                    req.continue_final_message, req.chat_template,
+                    req.chat_template_kwargs))
+            resp = Resp(request_id=f"render-{uuid.uuid4().hex[:12]}",
+                        token_ids=ids)
+            if hashes:
+                resp.features.CopyFrom(MM(
+                    mm_hashes=hashes,
+                    placeholder_ranges=[PR(offset=o, length=n)
+                                        for o, n in ranges]))
+            return resp
+        except Exception as e:
+            log.warning("render_chat_completion failed: %s", e)
+            return Resp(error=str(e))
+
+    async def render_completion(self, req, ctx):
+        import uuid
+
+        Resp = proto.get("tokenizerpb.RenderCompletionResponse")
+        try:
+            ids = await self.m.run(self.m.render_completion, req.model_name,
+                                   req.prompt)
+            return Resp(request_id=f"render-{uuid.uuid4().hex[:12]}",
+                        token_ids=ids)
+        except Exception as e:
+            log.warning("render_completion failed: %s", e)
+            return Resp(error=str(e))
+
 
 def _handlers(servicer: TokenizationServicer):
     g = proto.get
@@ -130,6 +248,16 @@ def _handlers(servicer: TokenizationServicer):
             servicer.render_chat,
             g("tokenizerpb.RenderChatRequest"),
             g("tokenizerpb.RenderChatResponse"),
+        ),
+        "RenderChatCompletion": (
+            servicer.render_chat_completion,
+            g("tokenizerpb.RenderChatCompletionRequest"),
+            g("tokenizerpb.RenderChatCompletionResponse"),
+        ),
+        "RenderCompletion": (
+            servicer.render_completion,
+            g("tokenizerpb.RenderCompletionRequest"),
+            g("tokenizerpb.RenderCompletionResponse"),
         ),
     }
     method_handlers = {}

@@ -186,3 +186,46 @@ def test_indexer_service_prompt_path(uds_server):
     finally:
         server.stop(0.1)
         pool.shutdown()
+
+
+def test_render_chat_completion_and_completion(uds_server):
+    """OpenAI-shape render RPCs (reference RenderChatCompletion /
+    RenderCompletion): multimodal parts produce content-addressed mm
+    hashes + placeholder token ranges; plain completion tokenizes."""
+    from llm_d_kv_cache_amd.services.tokenizer_client import (
+        UdsTokenizerClient,
+    )
+
+    c = UdsTokenizerClient(uds_server)
+    rid, ids, hashes, ranges = c.render_chat_completion(
+        "toy-model",
+        [{"role": "user", "content_parts": [
+            {"type": "text", "text": "hello"},
+            {"type": "image_url", "image_url": {"url": "data:img-1"}},
+            {"type": "text", "text": "world"}]},
+         {"role": "assistant", "content": "how are you"}],
+    )
+    assert rid.startswith("render-")
+    assert len(ids) > 4
+    assert len(hashes) == 1 and len(hashes[0]) == 32
+    assert len(ranges) == 1 and ranges[0][1] >= 1
+    # same image -> same hash (content-addressed); different -> different
+    _, _, h2, _ = c.render_chat_completion(
+        "toy-model",
+        [{"role": "user", "content_parts": [
+            {"type": "image_url", "image_url": {"url": "data:img-1"}},
+            {"type": "image_url", "image_url": {"url": "data:img-2"}}]}],
+    )
+    assert h2[0] == hashes[0] and h2[1] != h2[0]
+
+    # text-only chat completion matches the plain chat-template render
+    ids_cc = c.render_chat_completion(
+        "toy-model", [{"role": "user", "content": "hello world"}])[1]
+    ids_rc = c.render_chat("toy-model", [("user", "hello world")],
+                           add_generation_prompt=True)[0]
+    assert ids_cc == ids_rc
+
+    rid2, ids2 = c.render_completion("toy-model", "the quick brown fox")
+    assert rid2.startswith("render-")
+    assert len(ids2) == 4
+    c.close()

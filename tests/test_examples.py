@@ -183,3 +183,42 @@ def test_kv_cache_manager_facade():
         pub.close()
     finally:
         mgr.shutdown()
+
+
+def test_kv_events_offline_example():
+    out = subprocess.run(
+        [sys.executable, str(REPO / "examples" / "kv_events_offline.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.strip().endswith("ok")
+
+
+def test_indexer_service_client_example():
+    """End to end: start the real gRPC service on a free port, score with
+    the example client."""
+    import time
+
+    from llm_d_kv_cache_amd import ensure_native
+    from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+    from llm_d_kv_cache_amd.services.indexer_service import create_server
+
+    k = ensure_native()
+    ix = KVCacheIndexer(IndexerConfig())
+    keys = ix.compute_block_keys(list(range(48)), "demo-model")
+    ix.index.add([], keys, [k.PodEntry("pod-a", "gpu")])
+    server, port = create_server(ix, "127.0.0.1:0")
+    server.start()
+    try:
+        sys.path.insert(0, str(REPO / "examples"))
+        import indexer_service_client
+
+        resp = indexer_service_client.main([
+            "--target", f"127.0.0.1:{port}", "--model", "demo-model",
+            "--tokens", ",".join(str(t) for t in range(48)),
+            "--pods", "pod-a,pod-b"])
+        scores = {s.pod_identifier: s.score for s in resp.scores}
+        assert scores == {"pod-a": 3.0}
+        assert resp.total_blocks == 3
+    finally:
+        server.stop(0)

@@ -163,20 +163,24 @@ class PeerMigrationService:
     # ---- API ----------------------------------------------------------------
 
     def pull(self, chunk_hash: int, group: int, dst_block_ids: Sequence[int],
-             src_rank: int, timeout: float = 30.0) -> concurrent.futures.Future:
+             src_rank: int, timeout: float = 30.0,
+             fp8: bool = False) -> concurrent.futures.Future:
         """Pull a peer's cached chunk into local pages. The Future resolves
         to True (pulled), False (peer does not hold it), or raises on
-        timeout/shutdown."""
+        timeout/shutdown. fp8=True asks the peer to quantize HBM-resident
+        blocks to fp8 e4m3 on the fly — half the xGMI bytes, e4m3 rounding
+        on the payload (DRAM-cached chunks always travel in their stored
+        codec; the ACK says which arrived)."""
         fut: concurrent.futures.Future = concurrent.futures.Future()
         with self._q_mu:
             self._cmd_q.append(("pull", chunk_hash, group,
                                 [int(b) for b in dst_block_ids], src_rank,
-                                time.time() + timeout, fut))
+                                time.time() + timeout, fut, bool(fp8)))
         return fut
 
     def pull_many(self, chunks: Sequence[Tuple[int, int, Sequence[int]]],
-                  src_rank: int,
-                  timeout: float = 30.0) -> concurrent.futures.Future:
+                  src_rank: int, timeout: float = 30.0,
+                  fp8: bool = False) -> concurrent.futures.Future:
         """Batched pull: up to MAX_BATCH (chunk_hash, group, dst_block_ids)
         from ONE peer with one control round trip and one xGMI data
         transfer (small chunks amortize the handshake and the per-transfer
@@ -189,7 +193,7 @@ class PeerMigrationService:
                 for (h, g, ids) in chunks]
         with self._q_mu:
             self._cmd_q.append(("pull_many", norm, src_rank,
-                                time.time() + timeout, fut))
+                                time.time() + timeout, fut, bool(fp8)))
         return fut
 
     def close(self) -> None:
@@ -389,8 +393,8 @@ class PeerMigrationService:
         return False
 
     def _start_pull(self, payload, pending_pulls) -> None:
-        _, chunk_hash, group, dst_ids, src, deadline, fut = payload
-        if src == self.rank:
+        _, chunk_hash, group, dst_ids, src, deadline, fut, want_fp8 = payload
+        if src == self.rank:  # local: no wire, fp8 request is moot
             ok = self._local_serve(chunk_hash, group, dst_ids)
             if self.gpu_mode:
                 self._comm_stream.synchronize()
@@ -401,14 +405,30 @@ class PeerMigrationService:
             return
         req_id = self._next_req_id
         self._next_req_id += self.world
-        self._dist.send(self._msg(OP_PULL_REQ, req_id, group, chunk_hash,
-                                  len(dst_ids)), dst=src,
-                        group=self.control_group)
+        t = self._msg(OP_PULL_REQ, req_id, group, chunk_hash, len(dst_ids))
+        t[7] = 1 if want_fp8 else 0
+        self._dist.send(t, dst=src, group=self.control_group)
         pending_pulls[req_id] = {
             "dst_ids": dst_ids, "group": group, "src": src,
             "deadline": deadline, "fut": fut, "tracker": None, "buf": None,
         }
         self._stats.pulls_requested += 1
+
+    def _gather_fp8_slab(self, group, ids, stream):
+        """Quantizing gather of HBM blocks into an exact-size fp8 slab."""
+        nb = self._copier.packed_bytes_fp8(group, len(ids))
+        if self.gpu_mode:
+            scratch_b = self._copier.fp8_scratch_bytes(group, len(ids))
+            scratch = self._torch.empty(scratch_b, dtype=self._torch.uint8,
+                                        device="cuda")
+            slab = self._torch.empty(nb, dtype=self._torch.uint8,
+                                     device="cuda")
+            self._copier.gather_fp8(group, ids, slab.data_ptr(),
+                                    scratch.data_ptr(), stream)
+        else:
+            slab = self._torch.empty(nb, dtype=self._torch.uint8)
+            self._copier.gather_fp8(group, ids, slab.data_ptr(), 0, 0)
+        return slab
 
     def _serve_dram(self, chunk_hash, group, n_blocks):
         """DRAM-tier lookup with error isolation: a broken callback must
@@ -430,8 +450,8 @@ class PeerMigrationService:
         return host[:want].contiguous(), fp8
 
     def _start_pull_many(self, payload, pending_pulls) -> None:
-        _, chunks, src, deadline, fut = payload
-        if src == self.rank:
+        _, chunks, src, deadline, fut, want_fp8 = payload
+        if src == self.rank:  # local: no wire, fp8 request is moot
             res = [self._local_serve(h, g, ids) for (h, g, ids) in chunks]
             if self.gpu_mode:
                 self._comm_stream.synchronize()
@@ -443,6 +463,7 @@ class PeerMigrationService:
         self._next_req_id += self.world
         t = self._msg(OP_PULL_REQ_MULTI, req_id)
         t[3] = len(chunks)
+        t[7] = 1 if want_fp8 else 0
         for i, (h, g, ids) in enumerate(chunks):
             u = h & ((1 << 64) - 1)
             t[8 + 3 * i] = u - (1 << 64) if u >= (1 << 63) else u
@@ -465,21 +486,27 @@ class PeerMigrationService:
         n_blocks = int(buf[5])
         ok = int(buf[6])
         if op == OP_PULL_REQ:
+            want_fp8 = bool(int(buf[7]))
             ids = self.lookup_local(chunk_hash, group)
             grant = ids is not None and len(ids) == n_blocks
             dram = None
             if not grant and self._dram_lookup is not None:
                 dram = self._serve_dram(chunk_hash, group, n_blocks)
-            ok_code = 1 if grant else (0 if dram is None
-                                       else (2 if dram[1] else 1))
+            if grant:
+                ok_code = 2 if want_fp8 else 1
+            else:
+                ok_code = (0 if dram is None else (2 if dram[1] else 1))
             dist.send(self._msg(OP_PULL_ACK, req_id, group, chunk_hash,
                                 n_blocks, ok_code),
                       dst=sender, group=self.control_group)
             if grant:
-                slab = self._data_tensor(group, n_blocks)
-                self._copier.gather(group, ids, slab.data_ptr(),
-                                    self._comm_stream.cuda_stream
-                                    if self.gpu_mode else 0)
+                stream = (self._comm_stream.cuda_stream
+                          if self.gpu_mode else 0)
+                if want_fp8:
+                    slab = self._gather_fp8_slab(group, ids, stream)
+                else:
+                    slab = self._data_tensor(group, n_blocks)
+                    self._copier.gather(group, ids, slab.data_ptr(), stream)
                 w = dist.isend(slab, dst=sender, group=self.data_group)
                 pending_sends.append((self._track(w), slab))
                 self._stats.pulls_served += 1
@@ -514,6 +541,7 @@ class PeerMigrationService:
                 self._stats.pulls_failed += 1
                 del pending_pulls[req_id]
         elif op == OP_PULL_REQ_MULTI:
+            want_fp8 = bool(int(buf[7]))
             n_chunks = int(buf[3])
             reqs = []
             for i in range(n_chunks):
@@ -530,7 +558,11 @@ class PeerMigrationService:
                 ids = self.lookup_local(h, g)
                 if ids is not None and len(ids) == nb:
                     mask |= 1 << i
-                    total += self._copier.packed_bytes(g, nb)
+                    if want_fp8:
+                        fp8_mask |= 1 << i
+                        total += self._copier.packed_bytes_fp8(g, nb)
+                    else:
+                        total += self._copier.packed_bytes(g, nb)
                     served.append(("hbm", g, ids))
                     continue
                 if self._dram_lookup is not None:
@@ -558,9 +590,15 @@ class PeerMigrationService:
                 for item in served:
                     if item[0] == "hbm":
                         _, g, ids = item
-                        self._copier.gather(g, ids, slab.data_ptr() + off,
-                                            stream)
-                        off += self._copier.packed_bytes(g, len(ids))
+                        if want_fp8:
+                            part = self._gather_fp8_slab(g, ids, stream)
+                            slab[off:off + part.numel()].copy_(part)
+                            off += part.numel()
+                        else:
+                            self._copier.gather(g, ids,
+                                                slab.data_ptr() + off,
+                                                stream)
+                            off += self._copier.packed_bytes(g, len(ids))
                     else:
                         host = item[1]
                         slab[off:off + host.numel()].copy_(host)

@@ -80,40 +80,57 @@ class TieredKVLoader:
         """Resolve a prefix chunk by chunk; stops at the first miss (a
         prefix must be contiguous). Returns the number of chunks filled.
 
-        Storage is tried per chunk; on a storage miss the peer tier is
-        asked for the next run of chunks in ONE batched pull_many (one
-        control round trip + one xGMI transfer), and the granted prefix of
-        the batch extends the fill. The first chunk no tier serves ends
-        the walk."""
+        All storage loads are issued CONCURRENTLY up front (the engine's
+        I/O pool parallelizes them); the longest contiguous success prefix
+        counts. Gaps are then offered to the peer tier in batched
+        pull_many calls (one control round trip + one xGMI transfer per
+        batch); the granted prefix of each batch extends the fill up to
+        the next storage-loaded chunk. The first chunk no tier serves
+        ends the walk."""
+        import time as _time
+
         from . import MAX_BATCH
 
-        filled = 0
         n = len(chunk_hashes)
 
         def ids_of(ci):
             return list(
                 block_ids[ci * blocks_per_chunk:(ci + 1) * blocks_per_chunk])
 
-        while filled < n:
-            ids = ids_of(filled)
-            if not ids:
-                break
-            if self.load_handler is not None:
-                job = self.load_handler.transfer_async(
-                    [chunk_hashes[filled]], {group: ids})
-                if self._wait(self.load_handler, job):
-                    filled += 1
-                    continue
-            if self.peer is None:
-                break
-            batch = []
-            for ci in range(filled, min(n, filled + MAX_BATCH)):
-                bids = ids_of(ci)
-                if not bids:
+        status = [None] * n  # True: storage-loaded, False/None: not
+        if self.load_handler is not None:
+            jobs = {}
+            for ci in range(n):
+                ids = ids_of(ci)
+                if not ids:
                     break
-                batch.append((chunk_hashes[ci], group, bids))
-            if not batch:
+                jid = self.load_handler.transfer_async(
+                    [chunk_hashes[ci]], {group: ids})
+                jobs[jid] = ci
+            deadline = _time.time() + self.pull_timeout_s
+            pending = set(jobs)
+            while pending and _time.time() < deadline:
+                for res in self.load_handler.get_finished():
+                    ci = jobs.get(res.job_id)
+                    if ci is not None:
+                        status[ci] = res.success
+                        pending.discard(res.job_id)
+                if pending:
+                    _time.sleep(0.002)
+
+        filled = 0
+        while filled < n:
+            if status[filled]:
+                filled += 1
+                continue
+            if not ids_of(filled) or self.peer is None:
                 break
+            end = filled
+            while (end < n and end - filled < MAX_BATCH
+                   and not status[end] and ids_of(end)):
+                end += 1
+            batch = [(chunk_hashes[ci], group, ids_of(ci))
+                     for ci in range(filled, end)]
             granted_prefix = 0
             for rank in self.peer_ranks:
                 try:
@@ -133,8 +150,6 @@ class TieredKVLoader:
             if granted_prefix == 0:
                 break
             filled += granted_prefix
-            # a partial grant loops back: the next chunk retries storage
-            # before the peers are asked again
         return filled
 
     def _wait(self, handler, job_id) -> bool:

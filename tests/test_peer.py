@@ -148,6 +148,41 @@ def scenario_batched_pull(rank, svc, group):
         dist.barrier()
 
 
+def scenario_fp8_wire(rank, svc, group):
+    """fp8-on-the-wire pulls of HBM-resident blocks: requester opts in,
+    peer quantizes on gather, half the data-plane bytes, e4m3 tolerance."""
+    import torch.distributed as dist
+
+    torch.manual_seed(300 + rank)
+    group_b = svc._tensors[1]
+    vals = (torch.randn(32, 1024) * 2).to(torch.bfloat16)
+    group_b[0].copy_(vals.view(torch.uint8).reshape(32, 2048))
+    svc.register_blocks(0x41 + rank, 1, [0, 1])
+    golden = vals[[0, 1]].float()
+    peer_golden = [torch.zeros(2, 1024) for _ in range(2)]
+    for r in range(2):
+        src = golden.clone() if r == rank else peer_golden[r]
+        dist.broadcast(src, src=r)
+        peer_golden[r] = src
+    other = 1 - rank
+    before = svc.stats().bytes_received
+    ok = svc.pull(0x41 + other, 1, [10, 11], src_rank=other,
+                  fp8=True).result(timeout=60)
+    assert ok is True
+    moved = svc.stats().bytes_received - before
+    assert moved == svc._copier.packed_bytes_fp8(1, 2)  # half + scales
+    got = group_b[0][[10, 11]].view(torch.bfloat16).float()
+    want = peer_golden[other]
+    assert (got - want).abs().max() <= 0.07 * want.abs().amax()
+    # batched fp8 + a miss
+    res = svc.pull_many([(0x41 + other, 1, [14, 15]), (0x9E9E, 1, [16])],
+                        src_rank=other, fp8=True).result(timeout=60)
+    assert res == [True, False]
+    got2 = group_b[0][[14, 15]].view(torch.bfloat16).float()
+    assert (got2 - want).abs().max() <= 0.07 * want.abs().amax()
+    dist.barrier()
+
+
 def scenario_self_pull(rank, svc, group):
     """src_rank == own rank takes the local short circuit (RCCL cannot
     send to self): HBM hit, miss, and batched variants."""
@@ -272,6 +307,7 @@ def scenario_dram_tier(rank, svc, group):
     "scenario_batched_pull",
     "scenario_dram_tier",
     "scenario_self_pull",
+    "scenario_fp8_wire",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):

@@ -214,16 +214,43 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     pulled = svc.stats().bytes_received
     t = torch.tensor([dt], device=dev)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
-    svc.close()
     if not ok:
+        svc.close()
         return {"ok": False, "error": err or "peer pull failed on a rank"}
-    return {
+
+    # second round: fp8 on the wire (peer quantizes during the gather) —
+    # same collective structure, half the data-plane bytes per pull
+    raw_bytes = svc._copier.packed_bytes(0, bpf) * n_chunks
+    t1 = time.perf_counter()
+    try:
+        futs = [
+            svc.pull(0xE000 + src * 1000 + c, 0,
+                     list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
+                     src_rank=src, timeout=120, fp8=True)
+            for c in range(n_chunks)
+        ]
+        ok_local = all(f.result(timeout=150) for f in futs)
+    except Exception as e:
+        ok_local, err = False, f"fp8 pull: {e}"
+    if gpu:
+        torch.cuda.synchronize()
+    dt8 = time.perf_counter() - t1
+    ok8 = agree(ok_local)
+    wire8 = svc.stats().bytes_received - pulled
+    t8 = torch.tensor([dt8], device=dev)
+    dist.all_reduce(t8, op=dist.ReduceOp.MAX)
+    svc.close()
+    out = {
         "ok": True,
         "pull_GBps_per_gpu": round(pulled / dt / 1e9, 2),
         "pull_GBps_aggregate": round(pulled * world / float(t.item()) / 1e9, 2),
         "chunk_bytes": pulled // n_chunks,
         "n_pulls": n_chunks,
     }
+    if ok8:
+        out["fp8_wire_GBps_per_gpu"] = round(wire8 / dt8 / 1e9, 2)
+        out["fp8_logical_GBps_per_gpu"] = round(raw_bytes / dt8 / 1e9, 2)
+    return out
 
 
 def main():

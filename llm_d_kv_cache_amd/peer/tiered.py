@@ -15,7 +15,7 @@ into bytes in local KV pages.
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional, Sequence
+from typing import Optional, Sequence
 
 log = logging.getLogger(__name__)
 

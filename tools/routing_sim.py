@@ -25,7 +25,7 @@ import json
 import os
 import random
 import sys
-from collections import OrderedDict, defaultdict
+from collections import OrderedDict
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 

@@ -119,15 +119,20 @@ class KVCacheIndexer:
     def _attach_pool(self, pool) -> None:
         self._events_pool = pool
 
-    def _window_hints(self, pods: Sequence[str]) -> Dict[str, int]:
+    def _window_hints(self, pods: Sequence[str]):
+        """Per-pod KV-cache group structure for hybrid-aware scoring:
+        {pod: [(group_idx, window_blocks)]} with window_blocks 0 for
+        full-attention groups; only pods whose structure the events pool
+        has learned get hints (others keep the vanilla prefix walk)."""
         if not self.config.window_aware_scoring or self._events_pool is None:
             return {}
         bs = self.token_processor.block_size
-        hints: Dict[str, int] = {}
+        hints = {}
         for p in pods:
-            w = self._events_pool.native.sliding_window_tokens(p)
-            if w > 0:
-                hints[p] = -(-w // bs)  # ceil(window_tokens / block_tokens)
+            gw = self._events_pool.native.group_windows(p)
+            if gw:
+                hints[p] = [(g, -(-w // bs) if w > 0 else 0)
+                            for g, w in sorted(gw.items())]
         return hints
 
     @property

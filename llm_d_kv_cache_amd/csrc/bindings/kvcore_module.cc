@@ -259,7 +259,9 @@ PYBIND11_MODULE(_kvcore, m) {
           "score_tokens",
           [](Indexer& ix, py::handle tokens, const std::string& model,
              const std::vector<std::string>& pods, py::handle extra,
-             std::unordered_map<std::string, int32_t> window_hints) {
+             std::unordered_map<std::string,
+                                LongestPrefixScorer::GroupWindows>
+                 window_hints) {
             auto toks = as_tokens(tokens);
             auto ex = as_extra(extra);
             py::gil_scoped_release rel;
@@ -269,7 +271,9 @@ PYBIND11_MODULE(_kvcore, m) {
           },
           py::arg("tokens"), py::arg("model"),
           py::arg("pods") = std::vector<std::string>{}, py::arg("extra") = py::none(),
-          py::arg("window_hints") = std::unordered_map<std::string, int32_t>{})
+          py::arg("window_hints") =
+              std::unordered_map<std::string,
+                                 LongestPrefixScorer::GroupWindows>{})
       .def(
           "compute_block_keys",
           [](Indexer& ix, py::handle tokens, const std::string& model) {
@@ -331,6 +335,11 @@ PYBIND11_MODULE(_kvcore, m) {
       .def("sliding_window_tokens",
            [](EventPool& p, const std::string& pod) {
              return p.group_catalog().sliding_window_tokens(pod);
+           },
+           py::arg("pod"))
+      .def("group_windows",
+           [](EventPool& p, const std::string& pod) {
+             return p.group_catalog().group_windows(pod);
            },
            py::arg("pod"));
 

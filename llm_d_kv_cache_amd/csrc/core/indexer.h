@@ -45,7 +45,8 @@ class Indexer {
   ScoreResult score_tokens(const uint32_t* tokens, size_t n, const std::string& model,
                            const std::vector<std::string>& pods,
                            const std::vector<BlockExtra>* extra = nullptr,
-                           const std::unordered_map<std::string, int32_t>*
+                           const std::unordered_map<
+                               std::string, LongestPrefixScorer::GroupWindows>*
                                window_hints = nullptr) {
     ScoreResult res;
     auto keys = tp_->tokens_to_block_keys(0, tokens, n, model, extra);
@@ -61,10 +62,10 @@ class Indexer {
 
     auto hits = index_->lookup(keys, filter);
     res.hit_blocks = hits.size();
-    std::unordered_map<uint32_t, int32_t> wh;
+    std::unordered_map<uint32_t, LongestPrefixScorer::GroupWindows> wh;
     if (window_hints != nullptr) {
-      for (const auto& [pod, blocks] : *window_hints)
-        wh.emplace(index_->strings().intern(pod), blocks);
+      for (const auto& [pod, groups] : *window_hints)
+        wh.emplace(index_->strings().intern(pod), groups);
     }
     auto scores = scorer_->score(keys, hits, wh.empty() ? nullptr : &wh);
     for (const auto& [pod_id, s] : scores)

@@ -6,15 +6,17 @@
 // first missing block. Operates on interned ids end-to-end; the active set
 // is a flat vector updated in place (no per-key allocation).
 //
-// Window-aware extension (the reference marks hybrid/sliding-window-aware
-// scoring WIP, docs/architecture.md:291): for pods whose model uses pure
-// sliding-window attention, engines evict out-of-window leading blocks, so
-// the vanilla prefix walk scores such pods 0 even though the engine can
-// fully reuse the prefix (vLLM sliding-window cache-hit lookup only needs
-// the last ceil(window/block) blocks of a prefix present). score() accepts
-// per-pod window hints and credits such pods max over positions P of
-// P x best-tier-weight where the pod holds a consecutive run ending at P of
-// length >= min(window_blocks, P).
+// Hybrid/window-aware extension (the reference marks this WIP,
+// docs/architecture.md:291): for pods whose KV-cache group structure is
+// known (learned from BlockStored HMA fields), the score is computed with
+// a per-group walk and the pod's reusable prefix is the MIN over its
+// groups — full-attention groups need a strict prefix, sliding-window
+// groups only need a consecutive run ending at P covering
+// min(ceil(window/block), P) blocks (engines evict out-of-window leading
+// blocks, and their cache-hit lookup tolerates exactly that). This both
+// credits sliding-window pods the vanilla walk would score 0 AND removes
+// the vanilla walk's over-optimism for hybrid pods that lost one group's
+// blocks but not the other's. Entries without a group act as wildcards.
 #pragma once
 
 #include <algorithm>
@@ -32,14 +34,18 @@ class LongestPrefixScorer {
   explicit LongestPrefixScorer(std::unordered_map<uint32_t, double> tier_weights = {})
       : tier_weights_(std::move(tier_weights)) {}
 
+  // (group id, window length in blocks; <=0 = full attention)
+  using GroupWindows = std::vector<std::pair<int32_t, int32_t>>;
+
   // keys: the full ordered key list; hits: Lookup output (ordered subset).
-  // window_hints: optional pod id -> window length in blocks for pods with
-  // pure sliding-window attention (see header comment).
-  // Returns pod id -> accumulated weight over its consecutive prefix.
+  // window_hints: optional pod id -> group/window structure; hinted pods
+  // get the per-group hybrid walk (replacing their vanilla score).
+  // Returns pod id -> accumulated weight over its reusable prefix.
   std::unordered_map<uint32_t, double> score(
       const std::vector<uint64_t>& keys,
       const std::vector<std::pair<uint64_t, std::vector<PodEntry>>>& hits,
-      const std::unordered_map<uint32_t, int32_t>* window_hints = nullptr) const {
+      const std::unordered_map<uint32_t, GroupWindows>* window_hints =
+          nullptr) const {
     std::unordered_map<uint32_t, double> scores;
     if (keys.empty() || hits.empty()) return scores;
 
@@ -81,37 +87,55 @@ class LongestPrefixScorer {
     }
 
     if (window_hints != nullptr && !window_hints->empty()) {
-      for (const auto& [pod, wblocks] : *window_hints) {
-        if (wblocks <= 0) continue;
-        int64_t run = 0;
-        double run_w = 0.0;  // best tier weight inside the current run
-        double best = 0.0;
-        for (size_t i = 0; i < keys.size(); ++i) {
-          double w = -1.0;
-          auto it = by_key.find(keys[i]);
-          if (it != by_key.end()) {
-            for (const auto& e : *it->second) {
-              if (e.pod != pod) continue;
-              double ww = weight(e.tier);
-              if (ww > w) w = ww;
+      for (const auto& [pod, groups] : *window_hints) {
+        if (groups.empty()) continue;
+        int64_t p_pod = static_cast<int64_t>(keys.size());
+        double w_best = 0.0;
+        for (const auto& [gid, wblocks] : groups) {
+          int64_t p_g = 0;
+          int64_t run = 0;
+          double run_w = 0.0;
+          for (size_t i = 0; i < keys.size(); ++i) {
+            double w = -1.0;
+            auto it = by_key.find(keys[i]);
+            if (it != by_key.end()) {
+              for (const auto& e : *it->second) {
+                if (e.pod != pod) continue;
+                if (e.has_group() && e.group != gid) continue;
+                double ww = weight(e.tier);
+                if (ww > w) w = ww;
+              }
+            }
+            if (w < 0.0) {
+              if (wblocks <= 0) break;  // full attention: strict prefix ends
+              run = 0;
+              run_w = 0.0;
+              continue;
+            }
+            ++run;
+            if (w > run_w) run_w = w;
+            const int64_t p = static_cast<int64_t>(i) + 1;
+            if (wblocks <= 0) {
+              if (run == p) {  // still a strict prefix
+                p_g = p;
+                if (run_w > w_best) w_best = run_w;
+              }
+            } else if (run >= std::min<int64_t>(wblocks, p)) {
+              if (p > p_g) {
+                p_g = p;
+                if (run_w > w_best) w_best = run_w;
+              }
             }
           }
-          if (w < 0.0) {
-            run = 0;
-            run_w = 0.0;
-            continue;
-          }
-          ++run;
-          if (w > run_w) run_w = w;
-          const int64_t p = static_cast<int64_t>(i) + 1;
-          if (run >= std::min<int64_t>(wblocks, p)) {
-            double s = static_cast<double>(p) * run_w;
-            if (s > best) best = s;
-          }
+          if (p_g < p_pod) p_pod = p_g;
+          if (p_pod == 0) break;
         }
-        if (best > 0.0) {
-          auto [si, inserted] = scores.emplace(pod, best);
-          if (!inserted && best > si->second) si->second = best;
+        // Replace the vanilla (any-entry) score: the hybrid walk is the
+        // authoritative model of what the engine can actually reuse.
+        if (p_pod > 0 && w_best > 0.0) {
+          scores[pod] = static_cast<double>(p_pod) * w_best;
+        } else {
+          scores.erase(pod);
         }
       }
     }

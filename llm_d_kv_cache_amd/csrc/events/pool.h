@@ -50,11 +50,8 @@ class GroupCatalog {
     return jt->second;
   }
 
-  // Sliding-window hint for window-aware scoring: the pod's window size in
-  // TOKENS if every learned KV-cache group on the pod uses sliding-window
-  // attention (engines then evict out-of-window leading blocks), else 0
-  // (full/hybrid attention needs the whole prefix; with several sliding
-  // groups the MAX window is the binding requirement).
+  // Sliding-window hint: the pod's window size in TOKENS if every learned
+  // KV-cache group on the pod uses sliding-window attention, else 0.
   int32_t sliding_window_tokens(const std::string& pod) const {
     std::lock_guard<std::mutex> g(mu_);
     auto it = catalog_.find(pod);
@@ -65,6 +62,19 @@ class GroupCatalog {
       w = std::max(w, *md.sliding_window);
     }
     return w;
+  }
+
+  // Full group structure for hybrid-aware scoring: group id -> sliding
+  // window in TOKENS (0 = full attention). Empty map = unknown pod.
+  std::unordered_map<int32_t, int32_t> group_windows(
+      const std::string& pod) const {
+    std::lock_guard<std::mutex> g(mu_);
+    std::unordered_map<int32_t, int32_t> out;
+    auto it = catalog_.find(pod);
+    if (it == catalog_.end()) return out;
+    for (const auto& [gid, md] : it->second)
+      out[gid] = md.sliding_window.value_or(0) > 0 ? *md.sliding_window : 0;
+    return out;
   }
 
  private:

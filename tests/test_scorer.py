@@ -216,3 +216,50 @@ def test_window_hints_only_for_pure_sliding_pods():
     ]))
     assert pool.native.sliding_window_tokens("pod-sw2") == 64
     assert ix.score_tokens(tokens, MODEL, ["pod-sw2"]) == {}
+
+
+def test_hybrid_per_group_scoring():
+    """Per-group hybrid walk (reference WIP): the pod's reusable prefix is
+    the MIN over its KV-cache groups — losing sliding-window leading
+    blocks is fine (engines evict those by design), losing full-attention
+    leading blocks kills reuse even while sliding entries remain."""
+    from llm_d_kv_cache_amd.events import EventPoolConfig, KVEventsPool
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_removed_payload,
+        block_stored_payload,
+        encode_batch,
+    )
+
+    ix = make_indexer()
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    tokens = list(range(128))  # 8 blocks
+    swa = list(range(1, 9))       # engine hashes, sliding group 0
+    full = list(range(11, 19))    # engine hashes, full-attn group 1
+    pool.process("kv@pod-h2@m", 0, encode_batch([
+        block_stored_payload(swa, None, tokens, 16, group_idx=0,
+                             spec_kind="sliding_window", sliding_window=32),
+        block_stored_payload(full, None, tokens, 16, group_idx=1,
+                             spec_kind="full_attention"),
+    ]))
+    assert ix.score_tokens(tokens, MODEL) == {"pod-h2": 8.0}
+
+    # sliding group evicts its out-of-window leading blocks: full reuse
+    pool.process("kv@pod-h2@m", 1, encode_batch([
+        block_removed_payload(swa[:6], group_idx=0),
+    ]))
+    assert ix.score_tokens(tokens, MODEL, ["pod-h2"]) == {"pod-h2": 8.0}
+
+    # a second hybrid pod loses its full-attention FIRST block while every
+    # sliding entry remains: nothing is reusable — but the vanilla
+    # any-entry walk would report 8 (every key still has an entry)
+    pool.process("kv@pod-h3@m", 0, encode_batch([
+        block_stored_payload(list(range(21, 29)), None, tokens, 16,
+                             group_idx=0, spec_kind="sliding_window",
+                             sliding_window=32),
+        block_stored_payload(list(range(31, 39)), None, tokens, 16,
+                             group_idx=1, spec_kind="full_attention"),
+        block_removed_payload([31], group_idx=1),
+    ]))
+    assert ix.score_tokens(tokens, MODEL, ["pod-h3"]) == {}
+    ix.config.window_aware_scoring = False
+    assert ix.score_tokens(tokens, MODEL, ["pod-h3"]) == {"pod-h3": 8.0}

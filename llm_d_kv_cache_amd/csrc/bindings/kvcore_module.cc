@@ -344,8 +344,9 @@ PYBIND11_MODULE(_kvcore, m) {
            py::arg("pod"));
 
   py::class_<ZmtpPublisher>(m, "Publisher")
-      .def(py::init<const std::string&, bool>(), py::arg("endpoint"),
-           py::arg("bind") = true)
+      .def(py::init<const std::string&, bool, std::string, std::string>(),
+           py::arg("endpoint"), py::arg("bind") = true,
+           py::arg("username") = "", py::arg("password") = "")
       .def_property_readonly("port", &ZmtpPublisher::port)
       .def_property_readonly("peer_count", &ZmtpPublisher::peer_count)
       .def(
@@ -362,7 +363,9 @@ PYBIND11_MODULE(_kvcore, m) {
   py::class_<ZmtpSubscriber>(m, "Subscriber")
       .def(py::init([](const std::string& endpoint, const std::string& topic_filter,
                        std::shared_ptr<EventPool> pool, py::object callback,
-                       bool bind, int reconnect_ms) {
+                       bool bind, int reconnect_ms,
+                       const std::string& username,
+                       const std::string& password) {
              ZmtpSubscriber::Handler h;
              if (pool) {
                // Native fast path: deliver straight into the pool, no GIL.
@@ -381,11 +384,13 @@ PYBIND11_MODULE(_kvcore, m) {
                };
              }
              return std::make_unique<ZmtpSubscriber>(endpoint, topic_filter, std::move(h),
-                                                     bind, reconnect_ms);
+                                                     bind, reconnect_ms,
+                                                     username, password);
            }),
            py::arg("endpoint"), py::arg("topic_filter") = "",
            py::arg("pool") = nullptr, py::arg("callback") = py::none(),
-           py::arg("bind") = false, py::arg("reconnect_ms") = 5000)
+           py::arg("bind") = false, py::arg("reconnect_ms") = 5000,
+           py::arg("username") = "", py::arg("password") = "")
       .def_property_readonly("port", &ZmtpSubscriber::port)
       .def("close", &ZmtpSubscriber::close, py::call_guard<py::gil_scoped_release>());
 }

@@ -24,6 +24,7 @@
 
 #include <atomic>
 #include <chrono>
+#include <algorithm>
 #include <cstring>
 #include <functional>
 #include <memory>
@@ -94,22 +95,30 @@ inline bool read_all(int fd, void* buf, size_t n) {
 }
 
 // ---- greeting + handshake ---------------------------------------------------
-inline bool send_greeting(int fd) {
+// Mechanisms: NULL (default) and PLAIN (RFC 24 username/password — cleartext,
+// for in-cluster deployments that want more than nothing; not CURVE).
+inline bool send_greeting(int fd, const char* mech = "NULL",
+                          bool as_server = false) {
   uint8_t g[64] = {0};
   g[0] = 0xff;
   g[9] = 0x7f;
   g[10] = 3;  // version major
   g[11] = 0;  // version minor: 3.0 (peers downgrade; subscriptions as messages)
-  std::memcpy(g + 12, "NULL", 4);
+  std::memcpy(g + 12, mech, std::min<size_t>(20, std::strlen(mech)));
+  g[32] = as_server ? 1 : 0;
   return write_all(fd, g, sizeof(g));
 }
 
-inline bool recv_greeting(int fd) {
+inline bool recv_greeting(int fd, std::string* mech_out = nullptr) {
   uint8_t g[64];
   if (!read_all(fd, g, 64)) return false;
   if (g[0] != 0xff || g[9] != 0x7f) return false;
   if (g[10] < 3) return false;  // ZMTP < 3 unsupported
-  return std::memcmp(g + 12, "NULL", 4) == 0 || g[12] == 0;
+  std::string mech(reinterpret_cast<char*>(g + 12), 20);
+  mech.erase(mech.find_last_not_of('\0') + 1);
+  if (mech.empty()) mech = "NULL";
+  if (mech_out != nullptr) *mech_out = mech;
+  return true;
 }
 
 struct Frame {
@@ -155,25 +164,102 @@ inline bool recv_frame(int fd, Frame& f, size_t max_size = 256ull << 20) {
   return n == 0 || read_all(fd, f.body.data(), n);
 }
 
-// READY command with a Socket-Type property.
-inline bool send_ready(int fd, const std::string& socket_type) {
+inline bool send_command(int fd, const std::string& name,
+                         const std::string& payload = "") {
   std::string body;
-  body.push_back(5);
-  body += "READY";
-  body.push_back(11);
-  body += "Socket-Type";
-  uint32_t vlen = static_cast<uint32_t>(socket_type.size());
-  for (int i = 3; i >= 0; --i) body.push_back(static_cast<char>((vlen >> (8 * i)) & 0xff));
-  body += socket_type;
-  return send_frame(fd, body.data(), body.size(), /*more=*/false, /*command=*/true);
+  body.push_back(static_cast<char>(name.size()));
+  body += name;
+  body += payload;
+  return send_frame(fd, body.data(), body.size(), /*more=*/false,
+                    /*command=*/true);
 }
 
-inline bool recv_ready(int fd) {
+inline bool recv_command(int fd, std::string& name, std::string& payload) {
   Frame f;
   if (!recv_frame(fd, f)) return false;
   if (!f.command || f.body.empty()) return false;
   uint8_t nlen = static_cast<uint8_t>(f.body[0]);
-  return f.body.size() >= 1u + nlen && f.body.compare(1, nlen, "READY") == 0;
+  if (f.body.size() < 1u + nlen) return false;
+  name = f.body.substr(1, nlen);
+  payload = f.body.substr(1 + nlen);
+  return true;
+}
+
+// Metadata body: a Socket-Type property (READY / INITIATE payloads).
+inline std::string metadata_body(const std::string& socket_type) {
+  std::string body;
+  body.push_back(11);
+  body += "Socket-Type";
+  uint32_t vlen = static_cast<uint32_t>(socket_type.size());
+  for (int i = 3; i >= 0; --i)
+    body.push_back(static_cast<char>((vlen >> (8 * i)) & 0xff));
+  body += socket_type;
+  return body;
+}
+
+// READY command with a Socket-Type property.
+inline bool send_ready(int fd, const std::string& socket_type) {
+  return send_command(fd, "READY", metadata_body(socket_type));
+}
+
+inline bool recv_ready(int fd) {
+  std::string name, payload;
+  return recv_command(fd, name, payload) && name == "READY";
+}
+
+// PLAIN handshake (RFC 24): client HELLO(user, pass) -> server WELCOME ->
+// client INITIATE(metadata) -> server READY(metadata). The binding side
+// is the PLAIN server.
+inline bool plain_client(int fd, const std::string& user,
+                         const std::string& pass,
+                         const std::string& socket_type) {
+  std::string hello;
+  hello.push_back(static_cast<char>(user.size()));
+  hello += user;
+  hello.push_back(static_cast<char>(pass.size()));
+  hello += pass;
+  if (!send_command(fd, "HELLO", hello)) return false;
+  std::string name, payload;
+  if (!recv_command(fd, name, payload) || name != "WELCOME") return false;
+  if (!send_command(fd, "INITIATE", metadata_body(socket_type))) return false;
+  return recv_command(fd, name, payload) && name == "READY";
+}
+
+inline bool plain_server(int fd, const std::string& user,
+                         const std::string& pass,
+                         const std::string& socket_type) {
+  std::string name, payload;
+  if (!recv_command(fd, name, payload) || name != "HELLO") return false;
+  if (payload.empty()) return false;
+  size_t ul = static_cast<uint8_t>(payload[0]);
+  if (payload.size() < 2 + ul) return false;
+  size_t pl = static_cast<uint8_t>(payload[1 + ul]);
+  if (payload.size() < 2 + ul + pl) return false;
+  if (payload.substr(1, ul) != user || payload.substr(2 + ul, pl) != pass) {
+    std::string reason = "Invalid username or password";
+    send_command(fd, "ERROR",
+                 std::string(1, static_cast<char>(reason.size())) + reason);
+    return false;
+  }
+  if (!send_command(fd, "WELCOME")) return false;
+  if (!recv_command(fd, name, payload) || name != "INITIATE") return false;
+  return send_command(fd, "READY", metadata_body(socket_type));
+}
+
+// Full greeting + security handshake for either mechanism. `is_server`
+// is the TCP role (accepted vs dialed); empty username selects NULL.
+inline bool handshake_auth(int fd, bool is_server, const std::string& user,
+                           const std::string& pass,
+                           const std::string& socket_type) {
+  const char* mech = user.empty() ? "NULL" : "PLAIN";
+  if (!send_greeting(fd, mech, is_server && !user.empty())) return false;
+  std::string peer_mech;
+  if (!recv_greeting(fd, &peer_mech)) return false;
+  if (peer_mech != mech) return false;  // mechanism mismatch: refuse
+  if (user.empty())
+    return send_ready(fd, socket_type) && recv_ready(fd);
+  return is_server ? plain_server(fd, user, pass, socket_type)
+                   : plain_client(fd, user, pass, socket_type);
 }
 
 inline int connect_to(const Endpoint& ep, int timeout_ms = 5000) {
@@ -235,8 +321,12 @@ class ZmtpPublisher {
  public:
   // bind=true: listen for SUB peers (centralized fan-in topology).
   // bind=false: dial a bound SUB (engine-pod publishing to a central host).
-  explicit ZmtpPublisher(const std::string& endpoint, bool bind = true)
-      : ep_(zmtp::parse_endpoint(endpoint)), bind_(bind) {
+  // username non-empty selects the PLAIN mechanism (password checked on
+  // the binding side, presented by the dialing side).
+  explicit ZmtpPublisher(const std::string& endpoint, bool bind = true,
+                         std::string username = "", std::string password = "")
+      : ep_(zmtp::parse_endpoint(endpoint)), bind_(bind),
+        user_(std::move(username)), pass_(std::move(password)) {
     if (bind_) {
       listen_fd_ = zmtp::listen_on(ep_);
       ep_.port = zmtp::bound_port(listen_fd_);  // resolve port 0
@@ -367,8 +457,8 @@ class ZmtpPublisher {
   }
 
   bool handshake(Peer* peer) {
-    if (!zmtp::send_greeting(peer->fd) || !zmtp::recv_greeting(peer->fd)) return false;
-    if (!zmtp::send_ready(peer->fd, "PUB") || !zmtp::recv_ready(peer->fd)) return false;
+    if (!zmtp::handshake_auth(peer->fd, bind_, user_, pass_, "PUB"))
+      return false;
     std::lock_guard<std::mutex> pg(peer->mu);
     peer->ready = true;
     return true;
@@ -413,6 +503,7 @@ class ZmtpPublisher {
 
   zmtp::Endpoint ep_;
   bool bind_;
+  std::string user_, pass_;
   int listen_fd_ = -1;
   std::atomic<bool> closing_{false};
   std::thread accept_thread_;
@@ -432,9 +523,11 @@ class ZmtpSubscriber {
   // reconnect_ms on failure (pod-discovery topology). bind=true: accept
   // publisher connections (centralized topology).
   ZmtpSubscriber(const std::string& endpoint, std::string topic_filter,
-                 Handler handler, bool bind = false, int reconnect_ms = 5000)
+                 Handler handler, bool bind = false, int reconnect_ms = 5000,
+                 std::string username = "", std::string password = "")
       : ep_(zmtp::parse_endpoint(endpoint)), topic_(std::move(topic_filter)),
-        handler_(std::move(handler)), bind_(bind), reconnect_ms_(reconnect_ms) {
+        handler_(std::move(handler)), bind_(bind), reconnect_ms_(reconnect_ms),
+        user_(std::move(username)), pass_(std::move(password)) {
     if (bind_) {
       listen_fd_ = zmtp::listen_on(ep_);
       ep_.port = zmtp::bound_port(listen_fd_);
@@ -498,8 +591,7 @@ class ZmtpSubscriber {
   bool handshake_sub(int fd) {
     int one = 1;
     ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
-    if (!zmtp::send_greeting(fd) || !zmtp::recv_greeting(fd)) return false;
-    if (!zmtp::send_ready(fd, "SUB") || !zmtp::recv_ready(fd)) return false;
+    if (!zmtp::handshake_auth(fd, bind_, user_, pass_, "SUB")) return false;
     // 3.0-style subscription message: 0x01 + prefix ("" = everything).
     std::string sub;
     sub.push_back(0x01);
@@ -551,6 +643,7 @@ class ZmtpSubscriber {
   Handler handler_;
   bool bind_;
   int reconnect_ms_;
+  std::string user_, pass_;
   int listen_fd_ = -1;
   std::atomic<bool> closing_{false};
   std::thread main_thread_;

@@ -25,6 +25,9 @@ class EventPoolConfig:
     dp_rank_routing: bool = False
     # per-shard backlog bound (0 = unbounded): floods drop oldest messages
     max_queue_depth: int = 0
+    # ZMTP PLAIN credentials (empty username = NULL mechanism)
+    zmq_username: str = ""
+    zmq_password: str = ""
 
 
 class KVEventsPool:
@@ -55,6 +58,8 @@ class KVEventsPool:
                 self.config.topic_filter,
                 pool=self._pool,
                 bind=True,
+                username=self.config.zmq_username,
+                password=self.config.zmq_password,
             )
 
     @property
@@ -92,10 +97,13 @@ class SubscriberManager:
     """
 
     def __init__(self, pool: KVEventsPool, topic_filter: str = "kv@",
-                 reconnect_ms: int = 5000) -> None:
+                 reconnect_ms: int = 5000, username: str = "",
+                 password: str = "") -> None:
         self._pool = pool
         self._topic = topic_filter
         self._reconnect_ms = reconnect_ms
+        self._user = username or getattr(pool.config, "zmq_username", "")
+        self._pass = password or getattr(pool.config, "zmq_password", "")
         self._k = ensure_native()
         self._subs: Dict[str, tuple] = {}  # pod -> (endpoint, subscriber)
         self._mu = threading.Lock()
@@ -114,6 +122,8 @@ class SubscriberManager:
                 pool=self._pool.native,
                 bind=False,
                 reconnect_ms=self._reconnect_ms,
+                username=self._user,
+                password=self._pass,
             )
             self._subs[pod] = (endpoint, sub)
 

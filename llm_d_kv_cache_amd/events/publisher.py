@@ -77,9 +77,11 @@ def encode_batch(events: Iterable[list], ts: Optional[float] = None,
 class EventPublisher:
     """ZMTP PUB endpoint emitting KVEvents batches for one pod."""
 
-    def __init__(self, endpoint: str, pod_id: str, model: str, bind: bool = True):
+    def __init__(self, endpoint: str, pod_id: str, model: str,
+                 bind: bool = True, username: str = "", password: str = ""):
         self._k = ensure_native()
-        self._pub = self._k.Publisher(endpoint, bind=bind)
+        self._pub = self._k.Publisher(endpoint, bind=bind, username=username,
+                                      password=password)
         self.pod_id = pod_id
         self.model = model
         self._seq = 0

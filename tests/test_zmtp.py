@@ -181,3 +181,45 @@ def test_large_payload_long_frames():
     finally:
         sub.close()
         pub.close()
+
+
+def test_plain_auth_roundtrip():
+    """PLAIN mechanism (RFC 24): correct credentials deliver; a wrong
+    password is refused server-side (ERROR + close) and a mechanism
+    mismatch never completes the handshake."""
+    import time
+
+    k = ensure_native()
+    got = []
+    sub = k.Subscriber("tcp://127.0.0.1:0", "", callback=lambda t, s, p:
+                       got.append((t, s, p)), bind=True,
+                       username="svc", password="hunter2")
+    pub = k.Publisher(f"tcp://127.0.0.1:{sub.port}", bind=False,
+                      username="svc", password="hunter2")
+    deadline = time.time() + 5
+    while pub.peer_count == 0 and time.time() < deadline:
+        time.sleep(0.02)
+    pub.publish("kv@p@m", 7, b"auth-ok")
+    deadline = time.time() + 5
+    while not got and time.time() < deadline:
+        time.sleep(0.02)
+    assert got and got[0] == ("kv@p@m", 7, b"auth-ok")
+    pub.close()
+
+    # wrong password: handshake refused, nothing delivered
+    bad = k.Publisher(f"tcp://127.0.0.1:{sub.port}", bind=False,
+                      username="svc", password="wrong", )
+    time.sleep(0.4)
+    bad.publish("kv@p@m", 8, b"should-not-arrive")
+    time.sleep(0.3)
+    assert len(got) == 1
+    bad.close()
+
+    # mechanism mismatch (NULL publisher vs PLAIN subscriber): refused
+    anon = k.Publisher(f"tcp://127.0.0.1:{sub.port}", bind=False)
+    time.sleep(0.4)
+    anon.publish("kv@p@m", 9, b"anon")
+    time.sleep(0.3)
+    assert len(got) == 1
+    anon.close()
+    sub.close()

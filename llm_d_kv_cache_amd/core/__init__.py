@@ -172,3 +172,21 @@ class KVCacheIndexer:
 
     def stats(self):
         return self.index.stats()
+
+    def save_index(self, path: str) -> None:
+        """Snapshot the in-memory index for warm restarts (the event
+        stream still re-converges state; this just skips the cold-start
+        window). Network-backed indexes are already durable."""
+        save = getattr(self.index, "save", None)
+        if save is None:
+            raise RuntimeError(
+                "index backend has no snapshot support (network-backed "
+                "indexes are durable on the server side)")
+        save(path)
+
+    def load_index(self, path: str) -> None:
+        """Merge a snapshot back in (see save_index)."""
+        load = getattr(self.index, "load", None)
+        if load is None:
+            raise RuntimeError("index backend has no snapshot support")
+        load(path)

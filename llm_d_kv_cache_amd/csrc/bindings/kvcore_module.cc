@@ -229,7 +229,11 @@ PYBIND11_MODULE(_kvcore, m) {
              return std::make_shared<InMemoryIndex>(cfg);
            }),
            py::arg("size") = 100000000, py::arg("pods_per_key") = 10,
-           py::arg("shards") = 64, py::arg("max_bytes") = 0);
+           py::arg("shards") = 64, py::arg("max_bytes") = 0)
+      .def("save", &InMemoryIndex::save, py::arg("path"),
+           py::call_guard<py::gil_scoped_release>())
+      .def("load", &InMemoryIndex::load, py::arg("path"),
+           py::call_guard<py::gil_scoped_release>());
 
   py::class_<RedisIndex, IndexBackend, std::shared_ptr<RedisIndex>>(
       m, "RedisIndex")

@@ -13,6 +13,8 @@
 
 #include <algorithm>
 #include <atomic>
+#include <cstdio>
+#include <cstring>
 #include <cstdint>
 #include <list>
 #include <mutex>
@@ -301,7 +303,139 @@ class InMemoryIndex : public IndexBackend {
     return s;
   }
 
+  // ---- snapshot / restore ---------------------------------------------------
+  // Warm-restart aid the reference lacks (its index is ephemeral-only and
+  // re-converges from the event stream — which still works here; a
+  // snapshot just skips the cold-start window). Point-in-time per shard:
+  // concurrent writers during save() land in either the file or the live
+  // index, both fine for a cache. load() MERGES into the current index.
+
+  void save(const std::string& path) {
+    FILE* f = std::fopen((path + ".tmp").c_str(), "wb");
+    if (f == nullptr) throw std::runtime_error("snapshot open failed: " + path);
+    auto w64 = [&](uint64_t v) { std::fwrite(&v, 8, 1, f); };
+    auto w32 = [&](uint32_t v) { std::fwrite(&v, 4, 1, f); };
+    std::fwrite("KVIXSNP1", 8, 1, f);
+    // string table (ids are dense 0..n-1)
+    std::vector<std::string> strs;
+    const uint32_t n_ids = static_cast<uint32_t>(strings_.size());
+    strs.reserve(n_ids);
+    for (uint32_t i = 0; i < n_ids; ++i) strs.push_back(strings_.get(i));
+    w32(static_cast<uint32_t>(strs.size()));
+    for (const auto& v : strs) {
+      w32(static_cast<uint32_t>(v.size()));
+      std::fwrite(v.data(), 1, v.size(), f);
+    }
+    uint64_t n_keys = 0;
+    long n_keys_pos = std::ftell(f);
+    w64(0);  // patched below
+    for (auto& sh : shards_) {
+      std::lock_guard<std::mutex> g(sh.mu);
+      // oldest -> newest so a load that push_fronts restores LRU order
+      for (auto it = sh.lru.rbegin(); it != sh.lru.rend(); ++it) {
+        auto ke = sh.map.find(*it);
+        if (ke == sh.map.end()) continue;
+        w64(*it);
+        w32(static_cast<uint32_t>(ke->second.pods.size()));
+        for (const auto& e : ke->second.pods) {
+          w32(e.pod);
+          w32(e.tier);
+          uint8_t fl = e.flags;
+          std::fwrite(&fl, 1, 1, f);
+          int32_t gr = e.group;
+          std::fwrite(&gr, 4, 1, f);
+        }
+        ++n_keys;
+      }
+    }
+    uint64_t n_eng = 0;
+    long n_eng_pos = std::ftell(f);
+    w64(0);
+    for (auto& es : eng_shards_) {
+      std::lock_guard<std::mutex> g(es.mu);
+      for (auto it = es.lru.rbegin(); it != es.lru.rend(); ++it) {
+        auto ee = es.map.find(*it);
+        if (ee == es.map.end()) continue;
+        w64(*it);
+        w32(static_cast<uint32_t>(ee->second.rks.size()));
+        for (uint64_t rk : ee->second.rks) w64(rk);
+        ++n_eng;
+      }
+    }
+    std::fseek(f, n_keys_pos, SEEK_SET);
+    w64(n_keys);
+    std::fseek(f, n_eng_pos, SEEK_SET);
+    w64(n_eng);
+    std::fclose(f);
+    if (std::rename((path + ".tmp").c_str(), path.c_str()) != 0)
+      throw std::runtime_error("snapshot rename failed: " + path);
+  }
+
+  void load(const std::string& path) {
+    FILE* f = std::fopen(path.c_str(), "rb");
+    if (f == nullptr) throw std::runtime_error("snapshot open failed: " + path);
+    struct Closer { FILE* f; ~Closer() { std::fclose(f); } } closer{f};
+    auto fail = [&]() -> std::runtime_error {
+      return std::runtime_error("corrupt snapshot: " + path);
+    };
+    auto r64 = [&]() {
+      uint64_t v;
+      if (std::fread(&v, 8, 1, f) != 1) throw fail();
+      return v;
+    };
+    auto r32 = [&]() {
+      uint32_t v;
+      if (std::fread(&v, 4, 1, f) != 1) throw fail();
+      return v;
+    };
+    char magic[8];
+    if (std::fread(magic, 8, 1, f) != 1 || std::memcmp(magic, "KVIXSNP1", 8))
+      throw fail();
+    uint32_t n_strs = r32();
+    if (n_strs > (1u << 24)) throw fail();
+    std::vector<uint32_t> remap(n_strs);
+    for (uint32_t i = 0; i < n_strs; ++i) {
+      uint32_t len = r32();
+      if (len > (1u << 20)) throw fail();
+      std::string v(len, '\0');
+      if (len && std::fread(v.data(), 1, len, f) != len) throw fail();
+      remap[i] = strings_.intern(v);
+    }
+    uint64_t n_keys = r64();
+    for (uint64_t i = 0; i < n_keys; ++i) {
+      uint64_t key = r64();
+      uint32_t n = r32();
+      if (n > 4096) throw fail();
+      std::vector<PodEntry> entries(n);
+      for (uint32_t j = 0; j < n; ++j) {
+        uint32_t pod = r32(), tier = r32();
+        if (pod >= n_strs || tier >= n_strs) throw fail();
+        entries[j].pod = remap[pod];
+        entries[j].tier = remap[tier];
+        uint8_t fl;
+        int32_t gr;
+        if (std::fread(&fl, 1, 1, f) != 1 || std::fread(&gr, 4, 1, f) != 1)
+          throw fail();
+        entries[j].flags = fl;
+        entries[j].group = gr;
+      }
+      // add() dedupes + maintains LRU/caps/byte budget
+      std::reverse(entries.begin(), entries.end());  // preserve front order
+      add({}, {key}, entries);
+    }
+    uint64_t n_eng = r64();
+    for (uint64_t i = 0; i < n_eng; ++i) {
+      uint64_t ek = r64();
+      uint32_t n = r32();
+      if (n > (1u << 20)) throw fail();
+      std::vector<uint64_t> rks(n);
+      for (uint32_t j = 0; j < n; ++j) rks[j] = r64();
+      put_engine_mapping(ek, rks);
+    }
+  }
+
  private:
+
   struct KeyEntry {
     std::vector<PodEntry> pods;  // front = most recently added
     std::list<uint64_t>::iterator lru_it;

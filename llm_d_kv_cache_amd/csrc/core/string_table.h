@@ -39,6 +39,11 @@ class StringTable {
     return it == ids_.end() ? kInvalid : it->second;
   }
 
+  size_t size() const {
+    std::shared_lock rl(mu_);
+    return strings_.size();
+  }
+
   std::string get(uint32_t id) const {
     std::shared_lock rl(mu_);
     return id < strings_.size() ? strings_[id] : std::string();

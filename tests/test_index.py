@@ -221,3 +221,41 @@ def test_concurrent_add_evict_lookup():
     for t in threads:
         t.join()
     assert not errors
+
+
+def test_snapshot_save_load(tmp_path):
+    """Warm-restart snapshot: a fresh index restored from the file serves
+    the same lookups (entries, tiers, flags, groups, engine bridge)."""
+    idx = k.InMemoryIndex(shards=4)
+    idx.add([10, 11], [1, 2], [k.PodEntry("pod-a", "gpu"),
+                               k.PodEntry("pod-b", "cpu", speculative=True)])
+    idx.add([], [3], [k.PodEntry("pod-c", "gpu", group=2)])
+    path = str(tmp_path / "idx.snap")
+    idx.save(path)
+
+    idx2 = k.InMemoryIndex(shards=8)  # different shard count is fine
+    idx2.load(path)
+    got = idx2.lookup([1, 2, 3])
+    assert set(got.keys()) == {1, 2, 3}
+    assert {(e.pod, e.tier) for e in got[1]} == {("pod-a", "gpu"),
+                                                 ("pod-b", "cpu")}
+    assert [e for e in got[1] if e.pod == "pod-b"][0].speculative
+    e3 = got[3][0]
+    assert e3.pod == "pod-c" and e3.group == 2
+    assert idx2.get_request_key(10) == 1
+    assert idx2.get_request_key(11) == 2
+    assert idx2.stats().keys == 3
+
+
+def test_snapshot_corrupt_file_raises(tmp_path):
+    bad = tmp_path / "bad.snap"
+    bad.write_bytes(b"KVIXSNP1" + b"\xff" * 40)
+    idx = k.InMemoryIndex(shards=4)
+    with pytest.raises(RuntimeError):
+        idx.load(str(bad))
+    bad.write_bytes(b"NOTASNAP")
+    with pytest.raises(RuntimeError):
+        idx.load(str(bad))
+    # index stays usable after a failed load
+    idx.add([], [5], [k.PodEntry("p", "gpu")])
+    assert 5 in idx.lookup([5])

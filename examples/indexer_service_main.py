@@ -23,9 +23,16 @@ def main():
     ap.add_argument("--grpc-port", type=int, default=50051)
     ap.add_argument("--zmq-endpoint", default="tcp://0.0.0.0:5557")
     ap.add_argument("--tokenizer-uds", default=None)
+    ap.add_argument("--snapshot-path", default=None,
+                    help="warm-restart snapshot: loaded at startup if "
+                         "present, saved every --snapshot-interval seconds")
+    ap.add_argument("--snapshot-interval", type=float, default=300.0)
     args = ap.parse_args()
 
     indexer = KVCacheIndexer(IndexerConfig())
+    if args.snapshot_path and os.path.exists(args.snapshot_path):
+        indexer.load_index(args.snapshot_path)
+        print(f"restored index snapshot: {indexer.stats().keys} keys")
     pool = KVEventsPool(EventPoolConfig(zmq_endpoint=args.zmq_endpoint), indexer)
     pool.start()
 
@@ -44,8 +51,12 @@ def main():
     print(f"indexer service on :{port}, events on {args.zmq_endpoint}")
     try:
         while True:
-            time.sleep(3600)
+            time.sleep(args.snapshot_interval if args.snapshot_path else 3600)
+            if args.snapshot_path:
+                indexer.save_index(args.snapshot_path)
     except KeyboardInterrupt:
+        if args.snapshot_path:
+            indexer.save_index(args.snapshot_path)
         server.stop(1.0)
         pool.shutdown()
 

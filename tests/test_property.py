@@ -84,3 +84,40 @@ def test_index_invariants_under_arbitrary_ops(ops):
         assert entries
         for ent in entries:
             assert ent.pod.startswith("p")
+
+
+@settings(max_examples=80, deadline=None)
+@given(blob=st.binary(max_size=300), flip=st.integers(0, 10**6))
+def test_snapshot_loader_rejects_garbage(tmp_path_factory, blob, flip):
+    """Arbitrary bytes and bit-flipped valid snapshots must never crash the
+    loader — they either raise or merge a harmless subset; the index stays
+    usable either way."""
+    import os
+    import tempfile
+
+    d = tempfile.mkdtemp(prefix="snapfuzz")
+    garbage = os.path.join(d, "g.snap")
+    with open(garbage, "wb") as f:
+        f.write(blob)
+    idx = k.InMemoryIndex(shards=4)
+    try:
+        idx.load(garbage)
+    except Exception:
+        pass
+    # valid snapshot with one flipped byte
+    src = k.InMemoryIndex(shards=4)
+    src.add([7], [1, 2], [k.PodEntry("pod-a", "gpu"),
+                          k.PodEntry("pod-b", "cpu")])
+    valid = os.path.join(d, "v.snap")
+    src.save(valid)
+    data = bytearray(open(valid, "rb").read())
+    pos = flip % len(data)
+    data[pos] ^= 0xFF
+    with open(valid, "wb") as f:
+        f.write(bytes(data))
+    try:
+        idx.load(valid)
+    except Exception:
+        pass
+    idx.add([], [9], [k.PodEntry("p", "gpu")])
+    assert 9 in idx.lookup([9])

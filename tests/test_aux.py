@@ -268,3 +268,57 @@ def test_routing_sim_precise_beats_baselines():
         1.5 * res["random"]["cached_prefix_frac"]
     assert res["precise"]["ttft_mean_s"] < res["load"]["ttft_mean_s"] / 1.5
     assert res["precise"]["ttft_mean_s"] < res["random"]["ttft_mean_s"] / 3
+
+
+def test_cli_serve_and_score(tmp_path):
+    """python -m llm_d_kv_cache_amd: serve + score end to end over a real
+    gRPC port, with a snapshot restored at startup."""
+    import json
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    from llm_d_kv_cache_amd import ensure_native
+    from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+
+    k = ensure_native()
+    ix = KVCacheIndexer(IndexerConfig())
+    keys = ix.compute_block_keys(list(range(32)), "cli-m")
+    ix.index.add([], keys, [k.PodEntry("pod-cli", "gpu")])
+    snap = str(tmp_path / "cli.snap")
+    ix.save_index(snap)
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    srv = subprocess.Popen(
+        [sys.executable, "-m", "llm_d_kv_cache_amd", "serve",
+         "--grpc-port", str(port), "--zmq-endpoint", "tcp://127.0.0.1:0",
+         "--snapshot-path", snap],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        deadline = time.time() + 30
+        out = None
+        while time.time() < deadline:
+            try:
+                out = subprocess.run(
+                    [sys.executable, "-m", "llm_d_kv_cache_amd", "score",
+                     "--target", f"127.0.0.1:{port}", "--model", "cli-m",
+                     "--tokens", ",".join(str(t) for t in range(32)),
+                     "--pods", "pod-cli"],
+                    capture_output=True, text=True, timeout=15)
+                if out.returncode == 0 and "pod-cli" in out.stdout:
+                    break
+            except subprocess.TimeoutExpired:
+                pass
+            time.sleep(0.5)
+        assert out is not None and "pod-cli\t2.0" in out.stdout, \
+            (out.stdout if out else "no output")
+    finally:
+        srv.send_signal(signal.SIGINT)
+        try:
+            srv.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            srv.kill()

@@ -8,6 +8,7 @@ COPY api ./api
 COPY examples ./examples
 COPY tools ./tools
 ENV PYTORCH_ROCM_ARCH=gfx950
-RUN python -c "from llm_d_kv_cache_amd._build import build_all; build_all()"
+RUN python -m llm_d_kv_cache_amd build
 # control-plane entrypoint by default; override for engine-side use
-CMD ["python", "examples/indexer_service_main.py"]
+# (python -m llm_d_kv_cache_amd {serve,score,evict,build})
+CMD ["python", "-m", "llm_d_kv_cache_amd", "serve"]

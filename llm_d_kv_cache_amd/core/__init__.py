@@ -55,6 +55,9 @@ class RedisIndexConfig:
     port: int = 6379
     pool_size: int = 4
     key_prefix: str = "kv"
+    # Placeholder mirroring the reference's experimental Valkey-over-RDMA
+    # flag (redis.go:39-40): accepted, currently a no-op on this transport.
+    enable_rdma: bool = False
 
 
 @dataclass

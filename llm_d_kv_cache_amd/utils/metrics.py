@@ -26,10 +26,12 @@ except ImportError:  # pragma: no cover
 class KVCacheMetricsCollector:
     """Scrape-time collector over native index/pool/engine stats."""
 
-    def __init__(self, indexer=None, events_pool=None, offload_engine=None):
+    def __init__(self, indexer=None, events_pool=None, offload_engine=None,
+                 peer_service=None):
         self.indexer = indexer
         self.events_pool = events_pool
         self.offload_engine = offload_engine
+        self.peer_service = peer_service
 
     def collect(self):
         if self.indexer is not None:
@@ -83,13 +85,35 @@ class KVCacheMetricsCollector:
             yield CounterMetricFamily(
                 "kv_offload_bytes_loaded_total", "bytes loaded back",
                 value=e.bytes_loaded)
+        if self.peer_service is not None:
+            ps = self.peer_service.stats()
+            yield CounterMetricFamily(
+                "kv_peer_pulls_requested_total", "peer chunk pulls requested",
+                value=ps.pulls_requested)
+            yield CounterMetricFamily(
+                "kv_peer_pulls_served_total", "peer chunk pulls served",
+                value=ps.pulls_served)
+            yield CounterMetricFamily(
+                "kv_peer_pulls_served_dram_total",
+                "peer pulls served from the host-DRAM cache",
+                value=ps.pulls_served_dram)
+            yield CounterMetricFamily(
+                "kv_peer_pulls_failed_total", "peer pulls failed/denied",
+                value=ps.pulls_failed)
+            yield CounterMetricFamily(
+                "kv_peer_bytes_sent_total", "bytes served to peers",
+                value=ps.bytes_sent)
+            yield CounterMetricFamily(
+                "kv_peer_bytes_received_total", "bytes pulled from peers",
+                value=ps.bytes_received)
 
 
 def register(indexer=None, events_pool=None, offload_engine=None,
-             registry=None):
+             peer_service=None, registry=None):
     if not HAVE_PROMETHEUS:  # pragma: no cover
         raise RuntimeError("prometheus_client is not installed")
-    collector = KVCacheMetricsCollector(indexer, events_pool, offload_engine)
+    collector = KVCacheMetricsCollector(indexer, events_pool,
+                                        offload_engine, peer_service)
     (registry or REGISTRY).register(collector)
     return collector
 

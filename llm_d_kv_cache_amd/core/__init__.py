@@ -121,6 +121,8 @@ class KVCacheIndexer:
 
     def _attach_pool(self, pool) -> None:
         self._events_pool = pool
+        self._hint_cache: Dict[str, list] = {}
+        self._hint_version = -1
 
     def _window_hints(self, pods: Sequence[str]):
         """Per-pod KV-cache group structure for hybrid-aware scoring:
@@ -129,13 +131,25 @@ class KVCacheIndexer:
         has learned get hints (others keep the vanilla prefix walk)."""
         if not self.config.window_aware_scoring or self._events_pool is None:
             return {}
+        native = self._events_pool.native
+        if native.catalog_pods() == 0:
+            return {}  # no HMA fields seen: zero per-request overhead
+        ver = native.catalog_version()
+        if ver != self._hint_version:
+            self._hint_cache = {}
+            self._hint_version = ver
         bs = self.token_processor.block_size
         hints = {}
+        cache = self._hint_cache
         for p in pods:
-            gw = self._events_pool.native.group_windows(p)
-            if gw:
-                hints[p] = [(g, -(-w // bs) if w > 0 else 0)
-                            for g, w in sorted(gw.items())]
+            h = cache.get(p)
+            if h is None:
+                gw = native.group_windows(p)
+                h = [(g, -(-w // bs) if w > 0 else 0)
+                     for g, w in sorted(gw.items())]
+                cache[p] = h
+            if h:
+                hints[p] = h
         return hints
 
     @property

@@ -345,7 +345,11 @@ PYBIND11_MODULE(_kvcore, m) {
            [](EventPool& p, const std::string& pod) {
              return p.group_catalog().group_windows(pod);
            },
-           py::arg("pod"));
+           py::arg("pod"))
+      .def("catalog_version",
+           [](EventPool& p) { return p.group_catalog().version(); })
+      .def("catalog_pods",
+           [](EventPool& p) { return p.group_catalog().pod_count(); });
 
   py::class_<ZmtpPublisher>(m, "Publisher")
       .def(py::init<const std::string&, bool, std::string, std::string>(),

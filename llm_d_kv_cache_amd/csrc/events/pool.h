@@ -40,6 +40,17 @@ class GroupCatalog {
   void learn(const std::string& pod, int32_t group, GroupMetadata md) {
     std::lock_guard<std::mutex> g(mu_);
     catalog_[pod][group] = std::move(md);
+    version_.fetch_add(1, std::memory_order_relaxed);
+  }
+
+  // Monotonic change counter + pod count: lets scorers cache window hints
+  // and skip the per-pod queries entirely while no HMA fields were seen.
+  uint64_t version() const {
+    return version_.load(std::memory_order_relaxed);
+  }
+  size_t pod_count() const {
+    std::lock_guard<std::mutex> g(mu_);
+    return catalog_.size();
   }
   std::optional<GroupMetadata> get(const std::string& pod, int32_t group) const {
     std::lock_guard<std::mutex> g(mu_);
@@ -79,6 +90,7 @@ class GroupCatalog {
 
  private:
   mutable std::mutex mu_;
+  std::atomic<uint64_t> version_{0};
   std::unordered_map<std::string, std::unordered_map<int32_t, GroupMetadata>> catalog_;
 };
 

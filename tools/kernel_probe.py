@@ -67,6 +67,35 @@ def main():
     print(f"scatter: {moved / dt / 1e9:.1f} GB/s payload "
           f"({2 * moved / dt / 1e9:.1f} GB/s r+w), {dt / ITERS * 1e6:.0f} us/launch")
 
+    # fp8 split gather (amax + quantize) and dequantizing scatter, isolated
+    nb8 = copier.packed_bytes_fp8(0, BPF)
+    slab8 = torch.empty(nb8, dtype=torch.uint8, device="cuda")
+    scratch = torch.empty(copier.fp8_scratch_bytes(0, BPF),
+                          dtype=torch.uint8, device="cuda")
+    copier.gather_fp8(0, ids, slab8.data_ptr(), scratch.data_ptr(), stream)
+    copier.scatter_fp8(0, ids, slab8.data_ptr(), stream)
+    torch.cuda.synchronize()
+    src_bytes = copier.packed_bytes(0, BPF)
+    t0 = time.perf_counter()
+    for _ in range(ITERS):
+        copier.gather_fp8(0, ids, slab8.data_ptr(), scratch.data_ptr(),
+                          stream)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    rd = 2 * src_bytes * ITERS  # amax pass + quantize pass both read source
+    wr = nb8 * ITERS
+    print(f"fp8 gather:  {src_bytes * ITERS / dt / 1e9:.1f} GB/s logical "
+          f"({(rd + wr) / dt / 1e9:.1f} GB/s r+w), "
+          f"{dt / ITERS * 1e6:.0f} us/launch")
+    t0 = time.perf_counter()
+    for _ in range(ITERS):
+        copier.scatter_fp8(0, ids, slab8.data_ptr(), stream)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"fp8 scatter: {src_bytes * ITERS / dt / 1e9:.1f} GB/s logical "
+          f"({(nb8 + src_bytes) * ITERS / dt / 1e9:.1f} GB/s r+w), "
+          f"{dt / ITERS * 1e6:.0f} us/launch")
+
 
 
 

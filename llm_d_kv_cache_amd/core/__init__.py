@@ -136,21 +136,17 @@ class KVCacheIndexer:
             return {}  # no HMA fields seen: zero per-request overhead
         ver = native.catalog_version()
         if ver != self._hint_version:
-            self._hint_cache = {}
+            bs = self.token_processor.block_size
+            self._hint_cache = {
+                pod: [(g, -(-w // bs) if w > 0 else 0)
+                      for g, w in sorted(gw.items())]
+                for pod, gw in native.catalog_snapshot().items()
+            }
             self._hint_version = ver
-        bs = self.token_processor.block_size
-        hints = {}
         cache = self._hint_cache
-        for p in pods:
-            h = cache.get(p)
-            if h is None:
-                gw = native.group_windows(p)
-                h = [(g, -(-w // bs) if w > 0 else 0)
-                     for g, w in sorted(gw.items())]
-                cache[p] = h
-            if h:
-                hints[p] = h
-        return hints
+        if not pods:  # unfiltered scoring: hint every cataloged pod
+            return dict(cache)
+        return {p: cache[p] for p in pods if p in cache}
 
     @property
     def block_size(self) -> int:

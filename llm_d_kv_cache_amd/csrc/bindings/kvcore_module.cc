@@ -348,6 +348,8 @@ PYBIND11_MODULE(_kvcore, m) {
            py::arg("pod"))
       .def("catalog_version",
            [](EventPool& p) { return p.group_catalog().version(); })
+      .def("catalog_snapshot",
+           [](EventPool& p) { return p.group_catalog().snapshot(); })
       .def("catalog_pods",
            [](EventPool& p) { return p.group_catalog().pod_count(); });
 

@@ -88,6 +88,20 @@ class GroupCatalog {
     return out;
   }
 
+  // Whole-catalog snapshot (pod -> group -> window tokens): unfiltered
+  // scoring needs hints for every cataloged pod.
+  std::unordered_map<std::string, std::unordered_map<int32_t, int32_t>>
+  snapshot() const {
+    std::lock_guard<std::mutex> g(mu_);
+    std::unordered_map<std::string, std::unordered_map<int32_t, int32_t>> out;
+    for (const auto& [pod, groups] : catalog_) {
+      auto& m = out[pod];
+      for (const auto& [gid, md] : groups)
+        m[gid] = md.sliding_window.value_or(0) > 0 ? *md.sliding_window : 0;
+    }
+    return out;
+  }
+
  private:
   mutable std::mutex mu_;
   std::atomic<uint64_t> version_{0};

@@ -171,10 +171,12 @@ class PeerMigrationService:
         blocks to fp8 e4m3 on the fly — half the xGMI bytes, e4m3 rounding
         on the payload (DRAM-cached chunks always travel in their stored
         codec; the ACK says which arrived)."""
+        ids = [int(b) for b in dst_block_ids]
+        if not ids or len(ids) > 64:
+            raise ValueError("dst_block_ids must have 1..64 blocks")
         fut: concurrent.futures.Future = concurrent.futures.Future()
         with self._q_mu:
-            self._cmd_q.append(("pull", chunk_hash, group,
-                                [int(b) for b in dst_block_ids], src_rank,
+            self._cmd_q.append(("pull", chunk_hash, group, ids, src_rank,
                                 time.time() + timeout, fut, bool(fp8)))
         return fut
 
@@ -191,6 +193,9 @@ class PeerMigrationService:
         fut: concurrent.futures.Future = concurrent.futures.Future()
         norm = [(int(h), int(g), [int(b) for b in ids])
                 for (h, g, ids) in chunks]
+        for _, _, ids in norm:
+            if not ids or len(ids) > 64:
+                raise ValueError("each chunk must have 1..64 blocks")
         with self._q_mu:
             self._cmd_q.append(("pull_many", norm, src_rank,
                                 time.time() + timeout, fut, bool(fp8)))
@@ -311,40 +316,11 @@ class PeerMigrationService:
                     st = pending_pulls[req_id]
                     if st["tracker"] is not None and st["tracker"].done():
                         made_progress = True
-                        stream = (self._comm_stream.cuda_stream
-                                  if self.gpu_mode else 0)
-                        if "chunks" in st:  # batched pull
-                            base = st["buf"].data_ptr()
-                            off = 0
-                            for ci, (h, g, ids) in enumerate(st["chunks"]):
-                                if not st["granted"][ci]:
-                                    continue
-                                if st["fp8_chunks"][ci]:
-                                    self._copier.scatter_fp8(g, ids,
-                                                             base + off,
-                                                             stream)
-                                    off += self._copier.packed_bytes_fp8(
-                                        g, len(ids))
-                                else:
-                                    self._copier.scatter(g, ids, base + off,
-                                                         stream)
-                                    off += self._copier.packed_bytes(
-                                        g, len(ids))
-                            result = st["granted"]
-                        else:
-                            if st.get("fp8"):
-                                self._copier.scatter_fp8(
-                                    st["group"], st["dst_ids"],
-                                    st["buf"].data_ptr(), stream)
-                            else:
-                                self._copier.scatter(
-                                    st["group"], st["dst_ids"],
-                                    st["buf"].data_ptr(), stream)
-                            result = True
-                        if self.gpu_mode:
-                            self._comm_stream.synchronize()
-                        self._stats.bytes_received += st["buf"].numel()
-                        st["fut"].set_result(result)
+                        try:
+                            self._finish_pull(st)
+                        except Exception as e:
+                            st["fut"].set_exception(e)
+                            self._stats.pulls_failed += 1
                         del pending_pulls[req_id]
                     elif time.time() > st["deadline"]:
                         st["fut"].set_exception(
@@ -364,6 +340,37 @@ class PeerMigrationService:
         for st in pending_pulls.values():
             if not st["fut"].done():
                 st["fut"].set_exception(RuntimeError("peer service closed"))
+
+    def _finish_pull(self, st) -> None:
+        """Scatter a completed transfer into the destination pages and
+        resolve the future (exceptions surface on the future, never kill
+        the service loop)."""
+        stream = self._comm_stream.cuda_stream if self.gpu_mode else 0
+        if "chunks" in st:  # batched pull
+            base = st["buf"].data_ptr()
+            off = 0
+            for ci, (h, g, ids) in enumerate(st["chunks"]):
+                if not st["granted"][ci]:
+                    continue
+                if st["fp8_chunks"][ci]:
+                    self._copier.scatter_fp8(g, ids, base + off, stream)
+                    off += self._copier.packed_bytes_fp8(g, len(ids))
+                else:
+                    self._copier.scatter(g, ids, base + off, stream)
+                    off += self._copier.packed_bytes(g, len(ids))
+            result = st["granted"]
+        else:
+            if st.get("fp8"):
+                self._copier.scatter_fp8(st["group"], st["dst_ids"],
+                                         st["buf"].data_ptr(), stream)
+            else:
+                self._copier.scatter(st["group"], st["dst_ids"],
+                                     st["buf"].data_ptr(), stream)
+            result = True
+        if self.gpu_mode:
+            self._comm_stream.synchronize()
+        self._stats.bytes_received += st["buf"].numel()
+        st["fut"].set_result(result)
 
     def _local_serve(self, chunk_hash, group, dst_ids):
         """Self-pull short circuit (RCCL cannot send to self): resolve from

@@ -201,6 +201,19 @@ def scenario_self_pull(rank, svc, group):
     dist.barrier()
 
 
+def scenario_invalid_pull_sizes(rank, svc, group):
+    import pytest as _pytest
+    import torch.distributed as dist
+
+    with _pytest.raises(ValueError):
+        svc.pull(0x1, 0, [], src_rank=0)
+    with _pytest.raises(ValueError):
+        svc.pull(0x1, 0, list(range(65)), src_rank=0)
+    with _pytest.raises(ValueError):
+        svc.pull_many([(0x1, 0, list(range(65)))], src_rank=0)
+    dist.barrier()
+
+
 def scenario_multi_group(rank, svc, group):
     import torch.distributed as dist
 
@@ -308,6 +321,7 @@ def scenario_dram_tier(rank, svc, group):
     "scenario_dram_tier",
     "scenario_self_pull",
     "scenario_fp8_wire",
+    "scenario_invalid_pull_sizes",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):

@@ -52,6 +52,14 @@ class GroupCatalog {
     std::lock_guard<std::mutex> g(mu_);
     return catalog_.size();
   }
+
+  // A cleared (restarted) pod may come back with a different model
+  // config: forget its learned structure until new events re-teach it.
+  void forget(const std::string& pod) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (catalog_.erase(pod) > 0)
+      version_.fetch_add(1, std::memory_order_relaxed);
+  }
   std::optional<GroupMetadata> get(const std::string& pod, int32_t group) const {
     std::lock_guard<std::mutex> g(mu_);
     auto it = catalog_.find(pod);
@@ -259,6 +267,7 @@ class EventPool {
           // replica may be asked to clear a pod whose stores another
           // process wrote.
           index_->clear(index_->strings().intern(pod));
+          catalog_.forget(pod);
           break;
         }
       }

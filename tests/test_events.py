@@ -308,3 +308,25 @@ def test_hybrid_swa_eviction_keeps_full_attention_chain(setup):
         block_removed_payload([2], group_idx=0),
     ]))
     assert ix.score_tokens(tokens, MODEL) == {POD: 1.0}
+
+
+def test_clear_forgets_group_catalog():
+    """AllBlocksCleared also drops the pod's learned group structure: a
+    restarted pod may run a different model config, so stale window
+    metadata must not shape scoring until new events re-teach it."""
+    from llm_d_kv_cache_amd.events.publisher import (
+        all_blocks_cleared_payload,
+        block_stored_payload,
+        encode_batch,
+    )
+
+    ix = KVCacheIndexer(IndexerConfig())
+    pool = KVEventsPool(EventPoolConfig(), ix)
+    pool.process("kv@pod-r@m", 0, encode_batch([
+        block_stored_payload([1, 2], None, list(range(32)), 16, group_idx=0,
+                             spec_kind="sliding_window", sliding_window=64)]))
+    assert pool.native.sliding_window_tokens("pod-r") == 64
+    assert pool.native.catalog_pods() == 1
+    pool.process("kv@pod-r@m", 1, encode_batch([all_blocks_cleared_payload()]))
+    assert pool.native.sliding_window_tokens("pod-r") == 0
+    assert pool.native.catalog_pods() == 0

@@ -497,3 +497,44 @@ def test_mla_geometry_roundtrip(tmp_path):
     assert wait_finished(load)[0].success
     for t, o in zip(group, orig):
         assert torch.equal(t[:8], o)
+
+
+def test_fp8_with_dram_cache_and_writeback(tmp_path):
+    """fp8 serialize composed with the pinned cache + write-back policy:
+    the cache slot holds the fp8 payload, a hit load dequantizes from DRAM
+    (file deleted to prove it), and the async flush writes the fp8 file."""
+    import time
+
+    torch.manual_seed(11)
+    nl = 2
+    group = [
+        (torch.randn(NUM_BLOCKS, BLOCK_BYTES // 2) * 2).to(torch.bfloat16)
+        for _ in range(nl)
+    ]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="host", serialize="fp8_e4m3",
+                            host_cache_bytes=64 << 20, write_policy="back"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="fp8wb"))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    ids = list(range(8))
+    store.transfer_async([0xFB], {0: ids})
+    assert wait_finished(store)[0].success
+    deadline = time.time() + 10
+    while eng.stats().writeback_flushes < 1 and time.time() < deadline:
+        time.sleep(0.01)
+    path = mapper.file_name(0xFB, 0)
+    assert os.path.getsize(path) == BLOCKS_PER_FILE * nl * (BLOCK_BYTES // 2 + 4)
+    os.unlink(path)  # DRAM hit must serve the load
+    orig = [t[:8].float().clone() for t in group]
+    for t in group:
+        t.zero_()
+    load.transfer_async([0xFB], {0: ids})
+    assert wait_finished(load)[0].success
+    assert eng.stats().host_cache_hits >= 1
+    for t, o in zip(group, orig):
+        amax = o.abs().amax()
+        assert (t[:8].float() - o).abs().max() <= 0.07 * amax

@@ -311,3 +311,44 @@ def test_gpu_obj_fp8_roundtrip():
             assert (got - want).abs().max() <= 0.07 * amax
     finally:
         srv.shutdown()
+
+
+def test_gpu_fp8_dram_writeback_compose(tmp_path):
+    """fp8 + pinned DRAM cache + write-back on the staged GPU path: flush
+    writes the fp8 file, a DRAM hit (file deleted) dequantizes on device."""
+    torch.manual_seed(12)
+    nl = 4
+    group = [
+        (torch.randn(NUM_BLOCKS, BLOCK_BYTES // 2, device="cuda") * 2)
+        .to(torch.bfloat16)
+        for _ in range(nl)
+    ]
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=BPF,
+                            copy_path="staged", serialize="fp8_e4m3",
+                            host_cache_bytes=256 << 20, write_policy="back"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gfp8wb"))
+    store = GPUToStorageHandler(eng, mapper, [BPF])
+    load = StorageToGPUHandler(eng, mapper, [BPF])
+    ids = list(range(BPF))
+    store.transfer_async([0xFC], {0: ids})
+    assert wait_finished(store)[0].success
+    deadline = time.time() + 15
+    while eng.stats().writeback_flushes < 1 and time.time() < deadline:
+        time.sleep(0.01)
+    path = mapper.file_name(0xFC, 0)
+    assert os.path.getsize(path) == BPF * nl * (BLOCK_BYTES // 2 + 4)
+    os.remove(path)
+    orig = [t[:BPF].float().cpu() for t in group]
+    for t in group:
+        t.zero_()
+    torch.cuda.synchronize()
+    load.transfer_async([0xFC], {0: ids})
+    assert wait_finished(load)[0].success
+    torch.cuda.synchronize()
+    assert eng.stats().host_cache_hits >= 1
+    for t, o in zip(group, orig):
+        amax = o.abs().amax()
+        assert (t[:BPF].float().cpu() - o).abs().max() <= 0.07 * amax

@@ -149,6 +149,56 @@ def main():
     print(f"Score() during storm: p50 {lat[len(lat)//2]*1e6:.0f} us, "
           f"p99 {lat[int(len(lat)*0.99)]*1e6:.0f} us")
 
+    peer_isolation()
+
+
+def peer_isolation():
+    """GEMM retention under a block-migration storm: continuous gather →
+    scatter of 32 MiB chunks through the CDNA4 kernels (what a peer pull
+    does on each side of the xGMI hop, minus the link itself)."""
+    import threading
+
+    from llm_d_kv_cache_amd import ensure_offload_native
+
+    ko = ensure_offload_native()
+    group = [torch.randint(0, 255, (2048, 64 * 1024), dtype=torch.uint8,
+                           device="cuda") for _ in range(4)]
+    copier = ko.BlockCopier(
+        [([t.data_ptr() for t in group], [t.stride(0) for t in group],
+          64 * 1024, 2048)], gpu_mode=True)
+    nb = copier.packed_bytes(0, 16)
+    stream = torch.cuda.Stream()
+    slab = torch.empty(nb, dtype=torch.uint8, device="cuda")
+    stop = threading.Event()
+    moved = [0]
+
+    def migration_storm():
+        i = 0
+        with torch.cuda.stream(stream):
+            while not stop.is_set():
+                src = list(range((i * 16) % 1024, (i * 16) % 1024 + 16))
+                dst = list(range(1024 + (i * 16) % 1024,
+                                 1040 + (i * 16) % 1024))
+                copier.gather(0, src, slab.data_ptr(), stream.cuda_stream)
+                copier.scatter(0, dst, slab.data_ptr(), stream.cuda_stream)
+                if i % 8 == 0:
+                    stream.synchronize()
+                moved[0] += 2 * nb
+                i += 1
+        stream.synchronize()
+
+    t = threading.Thread(target=migration_storm, daemon=True)
+    t0 = time.perf_counter()
+    t.start()
+    time.sleep(0.5)
+    print("GEMM under migration storm:     ", end="", flush=True)
+    storm = gemm_tflops()
+    dt = time.perf_counter() - t0
+    stop.set()
+    t.join(timeout=30)
+    print(f"{storm:.0f} TF/s  (migrating concurrently: "
+          f"~{moved[0] / dt / 1e9:.0f} GB/s block copies)")
+
 
 if __name__ == "__main__":
     main()

@@ -5,8 +5,10 @@ built from, for embedding in other MI355X services.
   (BlockCopier; see csrc/offload/kernels.hip).
 - prefix_hash: batched chained block hashing on device token buffers
   (one lane per sequence).
-- fp8 serialize: fused quantize/dequantize variants live in the offload
-  engine (serialize="fp8_e4m3").
+- fp8 serialize: quantizing gather / dequantizing scatter are methods on
+  the copier (gather_fp8 / scatter_fp8 with packed_bytes_fp8 /
+  fp8_scratch_bytes sizing); the offload engine's serialize="fp8_e4m3"
+  rides the same kernels.
 """
 from __future__ import annotations
 

@@ -31,6 +31,7 @@ from llm_d_kv_cache_amd.offload import (
 SECONDS = float(sys.argv[1]) if len(sys.argv) > 1 else 60.0
 SERIALIZE = sys.argv[2] if len(sys.argv) > 2 else "raw"  # raw | fp8_e4m3
 COPY_PATH = sys.argv[3] if len(sys.argv) > 3 else ""  # staged | zero_copy
+WRITE_POLICY = sys.argv[4] if len(sys.argv) > 4 else "through"
 DISK_BUDGET = 4 * 1024**3  # hard cap on bytes resident in the root
 NUM_LAYERS = 16
 BLOCK_BYTES = 64 * 1024
@@ -61,6 +62,7 @@ def main():
             copy_path=(COPY_PATH or "staged") if gpu else "host",
             host_cache_bytes=1 * 1024**3,
             serialize=SERIALIZE,
+            write_policy=WRITE_POLICY,
             max_write_queued_seconds=2.0,  # provoke drops under storms
         ),
     )

@@ -310,7 +310,10 @@ class PeerMigrationService:
                         logging.getLogger(__name__).error(
                             "peer service error handling %s: %s", kind, e)
                         if kind == "cmd" and payload[0] != "bye":
-                            payload[-1].set_exception(e)
+                            # pull payloads end with (..., fut, want_fp8)
+                            fut = payload[-2]
+                            if not fut.done():
+                                fut.set_exception(e)
 
                 for req_id in list(pending_pulls):
                     st = pending_pulls[req_id]

@@ -322,6 +322,7 @@ def scenario_dram_tier(rank, svc, group):
     "scenario_self_pull",
     "scenario_fp8_wire",
     "scenario_invalid_pull_sizes",
+    "scenario_cmd_error_resolves_future",
     "scenario_tiered_loader",
 ])
 def test_peer_migration(scenario, tmp_path):
@@ -408,4 +409,34 @@ def scenario_tiered_loader(rank, svc, group):
             [CH_STORAGE, CH_PEER, CH_STORAGE, CH_MISS],
             list(range(8, 24)), 4)
         assert filled == 3
+    dist.barrier()
+
+
+def scenario_cmd_error_resolves_future(rank, svc, group):
+    """A command that blows up inside the service thread must resolve its
+    future with the exception (regression: the future moved to payload[-2]
+    when the fp8 flag was added)."""
+    import torch.distributed as dist
+
+    import concurrent.futures
+    import time as _time
+
+    if rank == 0:
+        # an invalid destination rank makes dist.send raise inside the
+        # service thread; the loop's error isolation must resolve the
+        # future instead of dying
+        fut = concurrent.futures.Future()
+        with svc._q_mu:
+            svc._cmd_q.append(("pull", 0x1, 0, [0], 99,
+                               _time.time() + 30, fut, False))
+        try:
+            fut.result(timeout=30)
+            raise AssertionError("expected an exception")
+        except AssertionError:
+            raise
+        except Exception:
+            pass  # surfaced correctly
+        # the service loop survived: a normal miss still resolves
+        assert svc.pull(0x1, 0, [0], src_rank=1 - rank).result(
+            timeout=30) is False
     dist.barrier()

@@ -223,3 +223,27 @@ def test_plain_auth_roundtrip():
     assert len(got) == 1
     anon.close()
     sub.close()
+
+
+def test_plain_auth_publisher_binds():
+    """PLAIN in the pod-discovery topology: publisher binds (PLAIN server),
+    subscriber dials with credentials."""
+    import time
+
+    k = ensure_native()
+    pub = k.Publisher("tcp://127.0.0.1:0", bind=True,
+                      username="eng", password="s3cret")
+    got = []
+    sub = k.Subscriber(f"tcp://127.0.0.1:{pub.port}", "",
+                       callback=lambda t, s, p: got.append((t, s, p)),
+                       bind=False, username="eng", password="s3cret")
+    deadline = time.time() + 5
+    while pub.peer_count == 0 and time.time() < deadline:
+        time.sleep(0.02)
+    deadline = time.time() + 5
+    while not got and time.time() < deadline:
+        pub.publish("kv@x@m", 1, b"hello-plain")
+        time.sleep(0.05)
+    assert got and got[-1][2] == b"hello-plain"
+    sub.close()
+    pub.close()

@@ -136,3 +136,41 @@ def start_metrics_logging(indexer, interval_s: float = 60.0,
     t = threading.Thread(target=beat, daemon=True, name="kvcache-metrics-beat")
     t.start()
     return stop_event
+
+
+class MeteredIndexer:
+    """Decorator over KVCacheIndexer exporting the reference's scoring
+    metrics (collector.go:28-75): `kvcache_lookup_latency_seconds`
+    histogram and `kvcache_max_pod_hit_count` counter of blocks on the
+    best pod per request. Composes with TracedIndexer (wrap either way)."""
+
+    def __init__(self, indexer, registry=None):
+        if not HAVE_PROMETHEUS:  # pragma: no cover
+            raise RuntimeError("prometheus_client is not installed")
+        import time as _time
+
+        from prometheus_client import Counter, Histogram
+
+        self._time = _time
+        self._ix = indexer
+        kw = {"registry": registry} if registry is not None else {}
+        self._lat = Histogram(
+            "kvcache_lookup_latency_seconds",
+            "score_tokens end-to-end latency",
+            buckets=(1e-5, 3e-5, 1e-4, 3e-4, 1e-3, 3e-3, 1e-2, 0.1), **kw)
+        self._max_hits = Counter(
+            "kvcache_max_pod_hit_count",
+            "sum over requests of the best pod's matched block count", **kw)
+
+    def __getattr__(self, name):
+        return getattr(self._ix, name)
+
+    def score_tokens(self, tokens, model_name, pod_identifiers=(),
+                     extra_features=None):
+        t0 = self._time.perf_counter()
+        scores = self._ix.score_tokens(tokens, model_name, pod_identifiers,
+                                       extra_features)
+        self._lat.observe(self._time.perf_counter() - t0)
+        if scores:
+            self._max_hits.inc(int(max(scores.values())))
+        return scores

@@ -322,3 +322,22 @@ def test_cli_serve_and_score(tmp_path):
             srv.wait(timeout=10)
         except subprocess.TimeoutExpired:
             srv.kill()
+
+
+def test_metered_indexer_metrics():
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from llm_d_kv_cache_amd import ensure_native
+    from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+    from llm_d_kv_cache_amd.utils.metrics import MeteredIndexer
+
+    k = ensure_native()
+    reg = CollectorRegistry()
+    ix = MeteredIndexer(KVCacheIndexer(IndexerConfig()), registry=reg)
+    keys = ix.compute_block_keys(list(range(48)), "m")
+    ix.index.add([], keys, [k.PodEntry("p", "gpu")])
+    assert ix.score_tokens(list(range(48)), "m") == {"p": 3.0}
+    ix.score_tokens(list(range(48)), "m")
+    text = generate_latest(reg).decode()
+    assert "kvcache_lookup_latency_seconds_count 2.0" in text
+    assert "kvcache_max_pod_hit_count_total 6.0" in text

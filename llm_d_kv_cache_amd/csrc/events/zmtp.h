@@ -248,9 +248,23 @@ inline bool plain_server(int fd, const std::string& user,
 
 // Full greeting + security handshake for either mechanism. `is_server`
 // is the TCP role (accepted vs dialed); empty username selects NULL.
+// The whole exchange runs under a receive timeout so a peer that
+// connects and stalls cannot pin a handshake thread forever; the
+// timeout is cleared before application traffic (long-idle SUBs are
+// normal).
 inline bool handshake_auth(int fd, bool is_server, const std::string& user,
                            const std::string& pass,
-                           const std::string& socket_type) {
+                           const std::string& socket_type,
+                           int timeout_ms = 10000) {
+  struct timeval tv = {timeout_ms / 1000, (timeout_ms % 1000) * 1000};
+  ::setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  struct Clear {
+    int fd;
+    ~Clear() {
+      struct timeval off = {0, 0};
+      ::setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &off, sizeof(off));
+    }
+  } clear{fd};
   const char* mech = user.empty() ? "NULL" : "PLAIN";
   if (!send_greeting(fd, mech, is_server && !user.empty())) return false;
   std::string peer_mech;

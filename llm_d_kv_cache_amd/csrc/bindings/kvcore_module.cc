@@ -291,6 +291,7 @@ PYBIND11_MODULE(_kvcore, m) {
       .def_readonly("enqueued", &PoolStats::enqueued)
       .def_readonly("processed", &PoolStats::processed)
       .def_readonly("parse_failures", &PoolStats::parse_failures)
+      .def_readonly("handler_failures", &PoolStats::handler_failures)
       .def_readonly("dropped_parent_misses", &PoolStats::dropped_parent_misses)
       .def_readonly("dropped_backpressure", &PoolStats::dropped_backpressure);
 

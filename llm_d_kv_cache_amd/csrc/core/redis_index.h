@@ -12,6 +12,7 @@
 // once every spanned request key is empty.
 #pragma once
 
+#include <charconv>
 #include <cstdio>
 #include <sstream>
 
@@ -148,23 +149,20 @@ class RedisIndex : public IndexBackend {
       }
       if (cmd.size() > 2) cmds.push_back(std::move(cmd));
     }
-    for (uint64_t rk : rks) cmds.push_back({"HLEN", rkey(rk)});
-    const size_t n_hdel = cmds.size() - rks.size();
-    auto replies = pool_.with(
-        [&](RespConnection& c) { return c.pipeline(cmds); });
+    if (!cmds.empty())
+      pool_.with([&](RespConnection& c) { return c.pipeline(cmds); });
     evictions_.fetch_add(1, std::memory_order_relaxed);
     if (type == KeyType::kEngine) {
-      bool all_empty = true;
-      std::vector<std::vector<std::string>> cleanup;
-      for (size_t i = 0; i < rks.size(); ++i) {
-        const auto& r = replies[n_hdel + i];
-        int64_t len = (!r.is_error && !r.is_nil()) ? r.integer() : 0;
-        if (len > 0) all_empty = false;
-        if (len == 0) cleanup.push_back({"DEL", rkey(rks[i])});
-      }
-      if (all_empty) cleanup.push_back({"DEL", ekey(key)});
-      if (!cleanup.empty())
-        pool_.with([&](RespConnection& c) { return c.pipeline(cleanup); });
+      // Atomic prune-if-empty: the empty-check and the deletes run as ONE
+      // server-side Lua script, so a concurrent add() interleaving between
+      // "hash is empty" and "DEL engine key" cannot lose entries (parity
+      // with the reference's Lua prune, redis.go:160-169; the old separate
+      // HLEN -> DEL round trips had exactly that race).
+      std::vector<std::string> ev{"EVAL", prune_script(),
+                                  std::to_string(rks.size() + 1)};
+      for (uint64_t rk : rks) ev.push_back(rkey(rk));
+      ev.push_back(ekey(key));
+      pool_.with([&](RespConnection& c) { return c.command(ev); });
     }
   }
 
@@ -228,6 +226,24 @@ class RedisIndex : public IndexBackend {
   }
 
  private:
+  // KEYS[1..n-1] = request-key hashes, KEYS[n] = the engine key.
+  // Deletes each empty request hash and, only if ALL are empty, the engine
+  // mapping — atomically (Redis runs scripts with no interleaved commands).
+  static const std::string& prune_script() {
+    static const std::string s =
+        "local all_empty = 1\n"
+        "for i = 1, #KEYS - 1 do\n"
+        "  if redis.call('HLEN', KEYS[i]) == 0 then\n"
+        "    redis.call('DEL', KEYS[i])\n"
+        "  else\n"
+        "    all_empty = 0\n"
+        "  end\n"
+        "end\n"
+        "if all_empty == 1 then redis.call('DEL', KEYS[#KEYS]) end\n"
+        "return all_empty\n";
+    return s;
+  }
+
   static std::string hex(uint64_t v) {
     char buf[17];
     snprintf(buf, sizeof(buf), "%016llx", static_cast<unsigned long long>(v));
@@ -247,6 +263,17 @@ class RedisIndex : public IndexBackend {
     return out;
   }
 
+  // Shared-backend fields can be written by other versions or edited by
+  // operators: every numeric parse validates and reports failure instead of
+  // throwing out of lookup()/evict() (which would feed the event pool's
+  // process-terminating path). Undecodable entries are skipped.
+  static bool parse_i64(const char* first, const char* last, int64_t* out,
+                        int base = 10) {
+    if (first == last) return false;
+    auto res = std::from_chars(first, last, *out, base);
+    return res.ec == std::errc() && res.ptr == last;
+  }
+
   bool decode_entry(const std::string& s, PodEntry* e) {
     size_t a = s.find('\x1f');
     if (a == std::string::npos) return false;
@@ -254,10 +281,17 @@ class RedisIndex : public IndexBackend {
     if (b == std::string::npos) return false;
     size_t c = s.find('\x1f', b + 1);
     if (c == std::string::npos) return false;
+    int64_t flags = 0, group = 0;
+    const char* p = s.data();
+    if (!parse_i64(p + b + 1, p + c, &flags) || flags < 0 || flags > 255)
+      return false;
+    if (!parse_i64(p + c + 1, p + s.size(), &group) || group < INT32_MIN ||
+        group > INT32_MAX)
+      return false;
     e->pod = strings_.intern(s.substr(0, a));
     e->tier = strings_.intern(s.substr(a + 1, b - a - 1));
-    e->flags = static_cast<uint8_t>(std::stoi(s.substr(b + 1, c - b - 1)));
-    e->group = std::stoi(s.substr(c + 1));
+    e->flags = static_cast<uint8_t>(flags);
+    e->group = static_cast<int32_t>(group);
     return true;
   }
 
@@ -265,8 +299,13 @@ class RedisIndex : public IndexBackend {
     std::vector<uint64_t> out;
     std::stringstream ss(s);
     std::string item;
-    while (std::getline(ss, item, ','))
-      if (!item.empty()) out.push_back(std::stoull(item, nullptr, 16));
+    while (std::getline(ss, item, ',')) {
+      if (item.empty()) continue;
+      uint64_t u = 0;  // request keys are full 64-bit values: unsigned hex
+      auto res = std::from_chars(item.data(), item.data() + item.size(), u, 16);
+      if (res.ec == std::errc() && res.ptr == item.data() + item.size())
+        out.push_back(u);
+    }
     return out;
   }
 

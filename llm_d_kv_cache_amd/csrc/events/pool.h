@@ -126,6 +126,7 @@ struct PoolStats {
   uint64_t enqueued = 0;
   uint64_t processed = 0;
   uint64_t parse_failures = 0;
+  uint64_t handler_failures = 0;
   uint64_t dropped_parent_misses = 0;
   uint64_t dropped_backpressure = 0;
 };
@@ -206,6 +207,7 @@ class EventPool {
     s.enqueued = enqueued_.load(std::memory_order_relaxed);
     s.processed = processed_.load(std::memory_order_relaxed);
     s.parse_failures = parse_failures_.load(std::memory_order_relaxed);
+    s.handler_failures = handler_failures_.load(std::memory_order_relaxed);
     s.dropped_parent_misses = dropped_parent_misses_.load(std::memory_order_relaxed);
     s.dropped_backpressure = dropped_backpressure_.load(std::memory_order_relaxed);
     return s;
@@ -255,21 +257,30 @@ class EventPool {
     if (dp_rank_routing_ && batch.dp_rank.has_value())
       pod += "-dp" + std::to_string(*batch.dp_rank);
     for (const auto& ev : batch.events) {
-      switch (ev.type) {
-        case EventType::kBlockStored:
-          handle_stored(pod, model, ev.stored);
-          break;
-        case EventType::kBlockRemoved:
-          handle_removed(pod, ev.removed);
-          break;
-        case EventType::kAllBlocksCleared: {
-          // Intern (not find): with a shared backend (Redis/Valkey) this
-          // replica may be asked to clear a pod whose stores another
-          // process wrote.
-          index_->clear(index_->strings().intern(pod));
-          catalog_.forget(pod);
-          break;
+      // Handlers can throw past the parser: a shared backend (RespError on
+      // a transient Redis/Valkey outage, malformed numeric fields) must not
+      // escape the worker thread — that would std::terminate the whole
+      // indexer process on a routine network blip. Count and drop the event;
+      // the index converges from later events / the other replicas.
+      try {
+        switch (ev.type) {
+          case EventType::kBlockStored:
+            handle_stored(pod, model, ev.stored);
+            break;
+          case EventType::kBlockRemoved:
+            handle_removed(pod, ev.removed);
+            break;
+          case EventType::kAllBlocksCleared: {
+            // Intern (not find): with a shared backend (Redis/Valkey) this
+            // replica may be asked to clear a pod whose stores another
+            // process wrote.
+            index_->clear(index_->strings().intern(pod));
+            catalog_.forget(pod);
+            break;
+          }
         }
+      } catch (const std::exception&) {
+        handler_failures_.fetch_add(1, std::memory_order_relaxed);
       }
     }
     processed_.fetch_add(1, std::memory_order_relaxed);
@@ -382,7 +393,7 @@ class EventPool {
   size_t max_queue_depth_ = 0;
   std::atomic<bool> running_{false};
   std::atomic<uint64_t> enqueued_{0}, processed_{0}, parse_failures_{0},
-      dropped_parent_misses_{0}, dropped_backpressure_{0};
+      handler_failures_{0}, dropped_parent_misses_{0}, dropped_backpressure_{0};
 };
 
 }  // namespace kvc

@@ -33,6 +33,7 @@
 #include <stdexcept>
 #include <string>
 #include <thread>
+#include <unordered_map>
 #include <vector>
 
 #include "events.h"  // RawMessage lives in pool.h; forward-declare instead
@@ -44,6 +45,54 @@ struct ZmtpError : std::runtime_error {
 };
 
 namespace zmtp {
+
+// Self-reaping thread set for per-connection reader threads: a finished
+// thread marks itself done and the next spawn() (or join_all()) joins and
+// erases it, so a flapping peer reconnecting every few seconds does not
+// grow an unjoined-handle list for the process lifetime.
+class ThreadSet {
+ public:
+  template <class F>
+  void spawn(F&& fn) {
+    std::lock_guard<std::mutex> g(mu_);
+    reap_finished_locked();
+    const uint64_t id = next_id_++;
+    threads_.emplace(id,
+                     std::thread([this, id, f = std::forward<F>(fn)]() mutable {
+                       f();
+                       std::lock_guard<std::mutex> g2(mu_);
+                       finished_.push_back(id);
+                     }));
+  }
+
+  void join_all() {
+    std::unordered_map<uint64_t, std::thread> taken;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      taken.swap(threads_);
+      finished_.clear();
+    }
+    for (auto& [id, t] : taken)
+      if (t.joinable()) t.join();
+  }
+
+ private:
+  void reap_finished_locked() {
+    for (uint64_t id : finished_) {
+      auto it = threads_.find(id);
+      if (it != threads_.end()) {
+        it->second.join();  // thread is past its body; join returns at once
+        threads_.erase(it);
+      }
+    }
+    finished_.clear();
+  }
+
+  std::mutex mu_;
+  uint64_t next_id_ = 0;
+  std::unordered_map<uint64_t, std::thread> threads_;
+  std::vector<uint64_t> finished_;
+};
 
 // ---- endpoint parsing: "tcp://host:port" -----------------------------------
 struct Endpoint {
@@ -394,8 +443,7 @@ class ZmtpPublisher {
     }
     if (accept_thread_.joinable()) accept_thread_.join();
     if (dial_thread_.joinable()) dial_thread_.join();
-    for (auto& t : reader_threads_)
-      if (t.joinable()) t.join();
+    reader_threads_.join_all();
     std::lock_guard<std::mutex> g(peers_mu_);
     for (auto& p : peers_) ::close(p->fd);
     peers_.clear();
@@ -464,7 +512,7 @@ class ZmtpPublisher {
       std::lock_guard<std::mutex> g(peers_mu_);
       peers_.push_back(std::move(peer));
     }
-    reader_threads_.emplace_back([this, raw] {
+    reader_threads_.spawn([this, raw] {
       if (handshake(raw)) reader(raw);
       reap(raw);
     });
@@ -522,7 +570,7 @@ class ZmtpPublisher {
   std::atomic<bool> closing_{false};
   std::thread accept_thread_;
   std::thread dial_thread_;
-  std::vector<std::thread> reader_threads_;
+  zmtp::ThreadSet reader_threads_;
   std::mutex peers_mu_;
   std::vector<std::unique_ptr<Peer>> peers_;
 };
@@ -564,8 +612,7 @@ class ZmtpSubscriber {
       for (int fd : live_fds_) ::shutdown(fd, SHUT_RDWR);
     }
     if (main_thread_.joinable()) main_thread_.join();
-    for (auto& t : reader_threads_)
-      if (t.joinable()) t.join();
+    reader_threads_.join_all();
   }
 
  private:
@@ -594,7 +641,7 @@ class ZmtpSubscriber {
         continue;
       }
       track(fd, true);
-      reader_threads_.emplace_back([this, fd] {
+      reader_threads_.spawn([this, fd] {
         if (handshake_sub(fd)) read_messages(fd);
         track(fd, false);
         ::close(fd);
@@ -661,7 +708,7 @@ class ZmtpSubscriber {
   int listen_fd_ = -1;
   std::atomic<bool> closing_{false};
   std::thread main_thread_;
-  std::vector<std::thread> reader_threads_;
+  zmtp::ThreadSet reader_threads_;
   std::mutex fds_mu_;
   std::set<int> live_fds_;
 };

@@ -19,8 +19,10 @@
 #pragma once
 
 #include <algorithm>
+#include <array>
 #include <cmath>
 #include <chrono>
+#include <functional>
 #include <thread>
 #include <list>
 #include <map>
@@ -235,6 +237,10 @@ class StorageOffloadEngine {
   // ---- store ----------------------------------------------------------------
 
   int64_t async_store(std::vector<FileTransfer> files, uintptr_t caller_stream) {
+    // Validate EVERYTHING before any task is enqueued: a throw mid-list
+    // after partial enqueue would leave job->remaining never reaching 0
+    // (wait_job deadlock, pending_writes_ leak).
+    for (const auto& ft : files) validate(ft);
     auto job = new_job(files.size(), /*is_store=*/true);
     if (files.empty()) {
       complete_empty(job);
@@ -245,13 +251,13 @@ class StorageOffloadEngine {
     // exceeds threads * max_queued_seconds / avg_write_seconds, the whole
     // store is dropped — offload is a cache, dropping is always safe.
     {
-      std::lock_guard<std::mutex> g(jobs_mu_);
-      double avg = stats_.avg_write_seconds;
+      double avg = avg_write_seconds();
       if (avg > 0) {
         size_t limit = static_cast<size_t>(
             pool_->size() * cfg_.max_write_queued_seconds / avg);
-        if (pending_writes_ > limit) {
-          stats_.writes_dropped += files.size();
+        if (pending_writes_.load(std::memory_order_relaxed) > limit) {
+          stats_inc([&](EngineStats& s) { s.writes_dropped += files.size(); });
+          std::lock_guard<std::mutex> g(jobs_mu_);
           job->remaining = 0;
           finished_.push_back({job->id, true, /*dropped=*/true});
           jobs_.erase(job->id);
@@ -259,7 +265,7 @@ class StorageOffloadEngine {
           return job->id;
         }
       }
-      pending_writes_ += files.size();
+      pending_writes_.fetch_add(files.size(), std::memory_order_relaxed);
     }
 
     if (cfg_.gpu_mode) {
@@ -270,7 +276,6 @@ class StorageOffloadEngine {
     stats_inc([&](EngineStats& s) { s.stores_submitted += files.size(); });
 
     for (auto& ft : files) {
-      validate(ft);
       pool_->enqueue(Priority::kNormal, [this, job, ft](WorkerCtx& ctx) {
         bool ok = true;
         double t0 = now_s();
@@ -292,18 +297,8 @@ class StorageOffloadEngine {
           stats_inc([](EngineStats& s) { s.errors++; });
           ok = false;
         }
-        if (wrote) {
-          double dt = now_s() - t0;
-          std::lock_guard<std::mutex> g(jobs_mu_);
-          // EMA alpha=0.05 of write duration drives the queue limit.
-          stats_.avg_write_seconds = stats_.avg_write_seconds == 0
-                                         ? dt
-                                         : 0.95 * stats_.avg_write_seconds + 0.05 * dt;
-        }
-        {
-          std::lock_guard<std::mutex> g(jobs_mu_);
-          pending_writes_--;
-        }
+        if (wrote) update_write_ema(now_s() - t0);
+        pending_writes_.fetch_sub(1, std::memory_order_relaxed);
         task_done(job, ok);
       });
     }
@@ -313,6 +308,7 @@ class StorageOffloadEngine {
   // ---- load -----------------------------------------------------------------
 
   int64_t async_load(std::vector<FileTransfer> files) {
+    for (const auto& ft : files) validate(ft);  // before ANY enqueue (see store)
     auto job = new_job(files.size(), /*is_store=*/false);
     if (files.empty()) {
       complete_empty(job);
@@ -320,7 +316,6 @@ class StorageOffloadEngine {
     }
     stats_inc([&](EngineStats& s) { s.loads_submitted += files.size(); });
     for (auto& ft : files) {
-      validate(ft);
       pool_->enqueue(Priority::kHigh, [this, job, ft](WorkerCtx& ctx) {
         bool ok = true;
         try {
@@ -366,13 +361,36 @@ class StorageOffloadEngine {
   }
 
   EngineStats stats() {
-    std::lock_guard<std::mutex> g(jobs_mu_);
-    return stats_;
+    EngineStats out;
+    for (auto& sh : stats_shards_) {
+      std::lock_guard<std::mutex> g(sh.mu);
+      const EngineStats& s = sh.s;
+      out.stores_submitted += s.stores_submitted;
+      out.loads_submitted += s.loads_submitted;
+      out.files_written += s.files_written;
+      out.files_deduped += s.files_deduped;
+      out.files_read += s.files_read;
+      out.writes_dropped += s.writes_dropped;
+      out.tasks_cancelled += s.tasks_cancelled;
+      out.host_cache_hits += s.host_cache_hits;
+      out.host_cache_stores += s.host_cache_stores;
+      out.writeback_flushes += s.writeback_flushes;
+      out.errors += s.errors;
+      out.bytes_stored += s.bytes_stored;
+      out.bytes_loaded += s.bytes_loaded;
+      out.t_gather_ms += s.t_gather_ms;
+      out.t_d2h_ms += s.t_d2h_ms;
+      out.t_write_ms += s.t_write_ms;
+      out.t_read_ms += s.t_read_ms;
+      out.t_h2d_ms += s.t_h2d_ms;
+      out.t_scatter_ms += s.t_scatter_ms;
+    }
+    out.avg_write_seconds = avg_write_seconds();
+    return out;
   }
 
   size_t pending_writes() {
-    std::lock_guard<std::mutex> g(jobs_mu_);
-    return pending_writes_;
+    return pending_writes_.load(std::memory_order_relaxed);
   }
 
  private:
@@ -430,10 +448,46 @@ class StorageOffloadEngine {
     }
   }
 
+  // Per-thread-sharded stats: phase-timing updates run ~6x per task from
+  // every worker, so they must not contend on the job-completion mutex
+  // (or each other). Each worker thread maps to one shard; stats()
+  // aggregates.
+  static constexpr size_t kStatsShards = 32;
+  struct alignas(64) StatsShard {
+    std::mutex mu;
+    EngineStats s;
+  };
+
   template <typename F>
   void stats_inc(F f) {
-    std::lock_guard<std::mutex> g(jobs_mu_);
-    f(stats_);
+    thread_local size_t idx =
+        std::hash<std::thread::id>{}(std::this_thread::get_id()) % kStatsShards;
+    auto& sh = stats_shards_[idx];
+    std::lock_guard<std::mutex> g(sh.mu);
+    f(sh.s);
+  }
+
+  double avg_write_seconds() const {
+    uint64_t bits = write_ema_bits_.load(std::memory_order_relaxed);
+    double v;
+    std::memcpy(&v, &bits, 8);
+    return v;
+  }
+
+  void update_write_ema(double dt) {
+    // EMA alpha=0.05 of write duration drives the queue limit; lock-free
+    // CAS keeps it off the completion mutex.
+    uint64_t old_bits = write_ema_bits_.load(std::memory_order_relaxed);
+    for (;;) {
+      double old_v;
+      std::memcpy(&old_v, &old_bits, 8);
+      double new_v = old_v == 0 ? dt : 0.95 * old_v + 0.05 * dt;
+      uint64_t new_bits;
+      std::memcpy(&new_bits, &new_v, 8);
+      if (write_ema_bits_.compare_exchange_weak(old_bits, new_bits,
+                                                std::memory_order_relaxed))
+        return;
+    }
   }
 
   size_t tile_record_bytes(const GroupDesc& g) const {
@@ -941,13 +995,14 @@ class StorageOffloadEngine {
   std::unique_ptr<PcieMover> mover_;
   std::unique_ptr<HostPinnedCache> cache_;
 
-  std::mutex jobs_mu_;
+  std::mutex jobs_mu_;  // guards jobs_/finished_/next_job_id_ only
   std::condition_variable done_cv_;
   std::unordered_map<int64_t, std::shared_ptr<Job>> jobs_;
   std::vector<FinishedJob> finished_;
   int64_t next_job_id_ = 1;
-  size_t pending_writes_ = 0;
-  EngineStats stats_;
+  std::atomic<size_t> pending_writes_{0};
+  std::array<StatsShard, kStatsShards> stats_shards_;
+  std::atomic<uint64_t> write_ema_bits_{0};  // double bit-pattern (0.0 = unset)
 };
 
 }  // namespace kvo

@@ -34,12 +34,18 @@ class _JobInfo:
 
 class _BaseHandler:
     def __init__(self, engine: TorchOffloadEngine, mapper: FileMapper,
-                 blocks_per_file: Sequence[int]):
+                 blocks_per_file: Sequence[int],
+                 group_block_tokens: Optional[Sequence[int]] = None):
         """blocks_per_file: per group, engine blocks per offloaded chunk
-        (= offloaded_block_tokens // group_block_size_tokens)."""
+        (= offloaded_block_tokens // group_block_size_tokens).
+        group_block_tokens: per group, engine block size in tokens — needed
+        to convert a token-level load skip into each group's block units in
+        the hybrid multi-group case."""
         self.engine = engine
         self.mapper = mapper
         self.blocks_per_file = list(blocks_per_file)
+        self.group_block_tokens = (
+            list(group_block_tokens) if group_block_tokens is not None else None)
         self._jobs: Dict[int, _JobInfo] = {}
 
     def _file_bytes(self, group: int, n_blocks: int) -> int:
@@ -103,7 +109,11 @@ class StorageToGPUHandler(_BaseHandler):
 
     ``skip_leading_blocks`` skips engine blocks already on the GPU: whole
     leading files are dropped and the first remaining file is tail-seeked
-    via slot_offset.
+    via slot_offset. It is expressed in the group's own block units, which
+    is only well-defined when every group uses the same block size; for
+    hybrid multi-group configs pass ``skip_leading_tokens`` instead and the
+    skip is converted per group (skip_g = tokens // group_block_tokens[g],
+    the reference worker's per-group logical start index).
     """
 
     def transfer_async(
@@ -111,7 +121,17 @@ class StorageToGPUHandler(_BaseHandler):
         chunk_hashes: Sequence[int],
         block_ids_per_group: Dict[int, Sequence[int]],
         skip_leading_blocks: int = 0,
+        skip_leading_tokens: Optional[int] = None,
     ) -> int:
+        heterogeneous = len(set(self.blocks_per_file)) > 1
+        if (skip_leading_blocks and heterogeneous
+                and skip_leading_tokens is None):
+            raise ValueError(
+                "skip_leading_blocks is ambiguous with heterogeneous group "
+                "block sizes; pass skip_leading_tokens instead")
+        if skip_leading_tokens is not None and self.group_block_tokens is None:
+            raise ValueError(
+                "skip_leading_tokens needs group_block_tokens at construction")
         files: List[Tuple[int, str, List[int], int]] = []
         total_bytes = 0
         for group, block_ids in block_ids_per_group.items():
@@ -119,8 +139,10 @@ class StorageToGPUHandler(_BaseHandler):
             block_ids = list(block_ids)
             # block_ids correspond to blocks AFTER the skip: the caller
             # passes only the ids it wants filled.
-            skip = skip_leading_blocks
-            # translate token-level skip (expressed in this group's blocks)
+            if skip_leading_tokens is not None:
+                skip = skip_leading_tokens // self.group_block_tokens[group]
+            else:
+                skip = skip_leading_blocks
             first_file = skip // bpf
             slot = skip % bpf
             cursor = 0

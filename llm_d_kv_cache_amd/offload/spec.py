@@ -80,9 +80,11 @@ class OffloadConnector:
             publisher=self.publisher,
         )
         self.store_handler = GPUToStorageHandler(
-            self.engine, self.mapper, self.blocks_per_file)
+            self.engine, self.mapper, self.blocks_per_file,
+            group_block_tokens=config.group_block_tokens)
         self.load_handler = StorageToGPUHandler(
-            self.engine, self.mapper, self.blocks_per_file)
+            self.engine, self.mapper, self.blocks_per_file,
+            group_block_tokens=config.group_block_tokens)
 
     def get_manager(self) -> SharedStorageOffloadManager:
         return self.manager

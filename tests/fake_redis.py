@@ -12,6 +12,10 @@ class FakeRedis:
     def __init__(self):
         self.hashes: Dict[bytes, Dict[bytes, bytes]] = {}
         self.strings: Dict[bytes, bytes] = {}
+        # test hook: called (under the server lock) just before an EVAL
+        # executes — lets tests inject a "concurrent" mutation at the exact
+        # point the prune script's atomicity matters.
+        self.on_eval = None
         self.lock = threading.Lock()
         self._srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         self._srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
@@ -20,6 +24,7 @@ class FakeRedis:
         self.port = self._srv.getsockname()[1]
         self._stop = False
         self._threads = []
+        self._conns = []
         t = threading.Thread(target=self._accept_loop, daemon=True)
         t.start()
         self._threads.append(t)
@@ -27,9 +32,23 @@ class FakeRedis:
     def close(self):
         self._stop = True
         try:
+            # shutdown() wakes the blocked accept(); close() alone leaves the
+            # listener alive inside the in-flight accept syscall
+            self._srv.shutdown(socket.SHUT_RDWR)
+        except OSError:
+            pass
+        try:
             self._srv.close()
         except OSError:
             pass
+        # drop accepted connections too: a real outage severs live clients,
+        # not just the listener
+        for c in list(self._conns):
+            try:
+                c.shutdown(socket.SHUT_RDWR)
+                c.close()
+            except OSError:
+                pass
 
     # ---- RESP plumbing ------------------------------------------------------
 
@@ -39,6 +58,7 @@ class FakeRedis:
                 conn, _ = self._srv.accept()
             except OSError:
                 return
+            self._conns.append(conn)
             t = threading.Thread(target=self._serve, args=(conn,), daemon=True)
             t.start()
             self._threads.append(t)
@@ -151,6 +171,30 @@ class FakeRedis:
                 return b"+OK\r\n"
             if cmd == b"GET":
                 return self._bulk(self.strings.get(args[1]))
+            if cmd == b"EVAL" and self.on_eval is not None:
+                hook, self.on_eval = self.on_eval, None
+                hook(self)
+            if cmd == b"EVAL":
+                # Atomic script execution (miniredis role): the only script
+                # the client sends is the prune-if-empty one; interpret its
+                # semantics under the single server lock so it is atomic
+                # w.r.t. every other command, exactly like real Redis Lua.
+                script = args[1]
+                if b"HLEN" not in script or b"all_empty" not in script:
+                    return b"-ERR unsupported script\r\n"
+                numkeys = int(args[2])
+                keys = args[3:3 + numkeys]
+                all_empty = 1
+                for k in keys[:-1]:
+                    if len(self.hashes.get(k, {})) == 0:
+                        self.hashes.pop(k, None)
+                    else:
+                        all_empty = 0
+                if all_empty and keys:
+                    ek = keys[-1]
+                    self.hashes.pop(ek, None)
+                    self.strings.pop(ek, None)
+                return self._int(all_empty)
             if cmd == b"SCAN":
                 # single-pass cursor: return everything matching on cursor 0
                 pattern = b"*"

@@ -150,3 +150,69 @@ def test_redis_cross_process_pod_filter(redis_pair):
     reader = KVCacheIndexer(cfg())
     assert reader.score_tokens(list(range(64)), "m",
                                ["pod-x", "pod-absent"]) == {"pod-x": 4.0}
+
+
+def test_redis_atomic_prune_no_lost_entries(redis_pair):
+    """A concurrent add landing at the prune's empty-check must not be
+    lost: the prune-if-empty runs as ONE atomic server-side script
+    (reference redis.go:160-169), so an entry written just before it
+    executes survives, and the engine bridge stays intact."""
+    srv, idx = redis_pair
+    idx.add([10], [1], [entry("pod-a")])
+
+    def concurrent_add(s):
+        # simulates another replica's add() racing the evict: the request
+        # hash regains a field right before the prune script runs
+        rkey = b"kv:r:" + b"%016x" % 1
+        s.hashes.setdefault(rkey, {})[b"pod-b\x1fgpu\x1f0\x1f0"] = b"1"
+
+    srv.on_eval = concurrent_add
+    idx.evict(10, "engine", [entry("pod-a")])
+    got = idx.lookup([1])
+    assert [e.pod for e in got[1]] == ["pod-b"], "concurrent add was lost"
+    assert idx.get_request_key(10) == 1, "engine bridge pruned under live key"
+
+
+def test_redis_malformed_fields_skipped(redis_pair):
+    """Corrupt shared-state fields (other versions, operator edits) are
+    skipped, never thrown out of lookup()/evict()."""
+    srv, idx = redis_pair
+    idx.add([10], [1], [entry("pod-a")])
+    rkey = b"kv:r:" + b"%016x" % 1
+    with srv.lock:
+        h = srv.hashes[rkey]
+        h[b"pod-x\x1fgpu\x1fnot-a-number\x1f0"] = b"1"   # bad flags
+        h[b"pod-y\x1fgpu\x1f0\x1f99999999999999999999"] = b"1"  # group overflow
+        h[b"no-separators-at-all"] = b"1"
+        srv.strings[b"kv:e:" + b"%016x" % 11] = b"zzzz,0001"  # bad rk list
+    got = idx.lookup([1])
+    assert [e.pod for e in got[1]] == ["pod-a"]
+    assert idx.get_request_key(11) == 1  # undecodable item skipped, good one used
+    idx.evict(10, "engine", [entry("pod-a")])  # must not raise
+
+
+def test_pool_survives_backend_outage(redis_pair):
+    """A transient Redis outage mid-stream must not kill the worker thread
+    (std::terminate would take down the whole indexer process): events
+    during the outage are counted as handler_failures and dropped."""
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_stored_payload,
+        encode_batch,
+    )
+
+    srv, idx = redis_pair
+    tp = k.TokenProcessor(16, "")
+    pool = k.EventPool(tp, idx, 1)
+    pool.start()
+    payload = encode_batch([block_stored_payload([1], None, list(range(16)), 16)])
+    pool.add_task("kv@pod-a@m", 0, payload)
+    pool.drain()
+    assert pool.stats().handler_failures == 0
+    srv.close()  # outage
+    for i in range(3):
+        pool.add_task("kv@pod-a@m", i + 1, payload)
+    pool.drain()  # workers must still be alive to drain
+    st = pool.stats()
+    assert st.handler_failures == 3
+    assert st.processed == 4
+    pool.shutdown()

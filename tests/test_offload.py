@@ -161,6 +161,42 @@ def test_multi_group_hma(tmp_path):
         assert torch.equal(t[:4], o[:4])
 
 
+def test_multi_group_skip_token_units(tmp_path):
+    # Heterogeneous group block sizes: a load skip must be translated into
+    # each group's own block units (skip_g = tokens // group_block_tokens[g])
+    # — applying one block count to every group loads the wrong file spans.
+    g0 = make_group(seed=3, num_layers=2, block_bytes=1024)  # 16-token blocks
+    g1 = make_group(seed=4, num_layers=2, block_bytes=2048)  # 32-token blocks
+    eng = TorchOffloadEngine(
+        [g0, g1],
+        OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=8, copy_path="host"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="het"))
+    store = GPUToStorageHandler(eng, mapper, [8, 4], group_block_tokens=[16, 32])
+    load = StorageToGPUHandler(eng, mapper, [8, 4], group_block_tokens=[16, 32])
+    hashes = [71, 72]  # 2 chunks x 128 tokens
+    store.transfer_async(hashes, {0: list(range(16)), 1: list(range(8))})
+    assert wait_finished(store)[0].success
+    orig0 = [t.clone() for t in g0]
+    orig1 = [t.clone() for t in g1]
+    for t in g0 + g1:
+        t.zero_()
+    # skip the first 192 tokens: g0 skips 12 blocks (file 1 + slot 4),
+    # g1 skips 6 blocks (file 1 + slot 2)
+    load.transfer_async(hashes, {0: list(range(12, 16)), 1: list(range(6, 8))},
+                        skip_leading_tokens=192)
+    assert wait_finished(load)[0].success
+    for t, o in zip(g0, orig0):
+        assert torch.equal(t[12:16], o[12:16])
+        assert (t[:12] == 0).all()
+    for t, o in zip(g1, orig1):
+        assert torch.equal(t[6:8], o[6:8])
+        assert (t[:6] == 0).all()
+    # a bare block-unit skip is ambiguous across heterogeneous groups
+    with pytest.raises(ValueError):
+        load.transfer_async(hashes, {0: [12], 1: [6]}, skip_leading_blocks=6)
+
+
 def test_wait_job_cancels_queued(setup):
     group, eng, mapper, store, load = setup
     # flood the 4-thread pool, then cancel the last job: its queued tasks bail

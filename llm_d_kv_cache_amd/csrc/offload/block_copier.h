@@ -8,10 +8,6 @@
 #include "common.h"
 #include "engine.h"  // GroupDesc, kernel launcher decls
 
-extern "C" hipError_t kvc_launch_gather_fp8_split(
-    const void* const*, const uint64_t*, int, uint64_t, const int32_t*, int,
-    uint8_t*, float*, hipStream_t);
-
 namespace kvo {
 
 class BlockCopier {
@@ -74,7 +70,7 @@ class BlockCopier {
     hipError_t err = kvc_launch_gather_fp8_split(
         const_cast<const void* const*>(dev_layer_ptrs_[group]),
         dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
-        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()), nullptr,
         static_cast<uint8_t*>(dst), static_cast<float*>(scratch),
         reinterpret_cast<hipStream_t>(stream));
     if (err != hipSuccess) throw HipError(hipGetErrorString(err));
@@ -94,7 +90,7 @@ class BlockCopier {
     hipError_t err = kvc_launch_scatter_fp8(
         const_cast<const void* const*>(dev_layer_ptrs_[group]),
         dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
-        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()), nullptr,
         static_cast<const uint8_t*>(src), reinterpret_cast<hipStream_t>(stream));
     if (err != hipSuccess) throw HipError(hipGetErrorString(err));
   }
@@ -112,7 +108,7 @@ class BlockCopier {
     hipError_t err = kvc_launch_gather(
         const_cast<const void* const*>(dev_layer_ptrs_[group]),
         dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
-        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()), nullptr,
         static_cast<uint8_t*>(dst), reinterpret_cast<hipStream_t>(stream));
     if (err != hipSuccess) throw HipError(hipGetErrorString(err));
   }
@@ -129,7 +125,7 @@ class BlockCopier {
     hipError_t err = kvc_launch_scatter(
         const_cast<const void* const*>(dev_layer_ptrs_[group]),
         dev_layer_strides_[group], static_cast<int>(g.layer_ptrs.size()),
-        g.block_bytes, ids.data(), static_cast<int>(ids.size()),
+        g.block_bytes, ids.data(), static_cast<int>(ids.size()), nullptr,
         static_cast<const uint8_t*>(src), reinterpret_cast<hipStream_t>(stream));
     if (err != hipSuccess) throw HipError(hipGetErrorString(err));
   }
@@ -137,7 +133,7 @@ class BlockCopier {
  private:
   static void check_ids(const std::vector<int32_t>& ids, const GroupDesc& g) {
     if (ids.empty() || ids.size() > kMaxBlocksPerFileHost)
-      throw std::invalid_argument("block count must be in [1, 64]");
+      throw std::invalid_argument("block count must be in [1, 128]");
     if (g.num_blocks > 0) {
       for (int32_t id : ids)
         if (id < 0 || id >= g.num_blocks)

@@ -34,25 +34,32 @@
 #include "pcie_mover.h"
 #include "thread_pool.h"
 
+// Launchers take host block ids (by-value kernarg, <=128) OR a device
+// pointer (ids_dev != nullptr) for bigger transfers — no kernel-side
+// blocks-per-file ceiling.
 extern "C" hipError_t kvc_launch_gather(const void* const*, const uint64_t*, int,
-                                        uint64_t, const int32_t*, int, uint8_t*,
-                                        hipStream_t);
+                                        uint64_t, const int32_t*, int,
+                                        const int32_t*, uint8_t*, hipStream_t);
 extern "C" hipError_t kvc_launch_scatter(const void* const*, const uint64_t*, int,
                                          uint64_t, const int32_t*, int,
-                                         const uint8_t*, hipStream_t);
+                                         const int32_t*, const uint8_t*,
+                                         hipStream_t);
 extern "C" hipError_t kvc_launch_gather_fp8(const void* const*, const uint64_t*,
                                             int, uint64_t, const int32_t*, int,
-                                            uint8_t*, hipStream_t);
+                                            const int32_t*, uint8_t*,
+                                            hipStream_t);
 extern "C" hipError_t kvc_launch_gather_fp8_split(
     const void* const*, const uint64_t*, int, uint64_t, const int32_t*, int,
-    uint8_t*, float*, hipStream_t);
+    const int32_t*, uint8_t*, float*, hipStream_t);
 extern "C" hipError_t kvc_launch_scatter_fp8(const void* const*, const uint64_t*,
                                              int, uint64_t, const int32_t*, int,
-                                             const uint8_t*, hipStream_t);
+                                             const int32_t*, const uint8_t*,
+                                             hipStream_t);
 
 namespace kvo {
 
-constexpr int kMaxBlocksPerFileHost = 64;  // mirrors kernels.hip kMaxBlocksPerFile
+constexpr int kMaxBlocksPerFileHost = 128;  // mirrors kernels.hip kMaxBlocksPerFile
+constexpr int kMaxBlocksPerFileDev = 4096;  // via per-worker device id buffer
 
 // Intra-transfer pipeline granularity: the PCIe hop and the file I/O of one
 // transfer overlap at this chunk size (copy chunk i+1 rides the SDMA stream
@@ -147,8 +154,9 @@ class StorageOffloadEngine {
  public:
   StorageOffloadEngine(EngineConfig cfg, std::vector<GroupDesc> groups)
       : cfg_(cfg), groups_(std::move(groups)) {
-    if (cfg_.gpu_blocks_per_file > kMaxBlocksPerFileHost)
-      throw std::invalid_argument("gpu_blocks_per_file exceeds kernel limit (64)");
+    if (cfg_.gpu_blocks_per_file > kMaxBlocksPerFileDev)
+      throw std::invalid_argument(
+          "gpu_blocks_per_file exceeds engine limit (4096)");
     if (cfg_.gpu_mode) {
       // Transient init failures have been observed right after another
       // process released the device: retry briefly before giving up.
@@ -499,6 +507,16 @@ class StorageOffloadEngine {
     return n_blocks * g.layer_ptrs.size() * tile_record_bytes(g);
   }
 
+  // Block ids above the kernarg limit ride the worker's device buffer.
+  const int32_t* stage_ids(WorkerCtx& ctx, const FileTransfer& ft) {
+    if (ft.block_ids.size() <= static_cast<size_t>(kMaxBlocksPerFileHost))
+      return nullptr;
+    KVO_HIP_CHECK(hipMemcpyAsync(ctx.dev_ids, ft.block_ids.data(),
+                                 ft.block_ids.size() * sizeof(int32_t),
+                                 hipMemcpyHostToDevice, ctx.stream));
+    return ctx.dev_ids;
+  }
+
   void store_one(WorkerCtx& ctx, Job& job, const FileTransfer& ft) {
     const GroupDesc& g = groups_[ft.group];
     const int nb = static_cast<int>(ft.block_ids.size());
@@ -559,6 +577,7 @@ class StorageOffloadEngine {
       uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
                                 ? ctx.device_staging->ptr()
                                 : ctx.host_staging->device();
+      const int32_t* ids_dev = stage_ids(ctx, ft);
       hipError_t err;
       if (cfg_.serialize == Serialize::kFp8E4M3) {
         if (cfg_.copy_path == CopyPath::kStaged) {
@@ -572,20 +591,21 @@ class StorageOffloadEngine {
           err = kvc_launch_gather_fp8_split(
               const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
               dev_layer_strides_[ft.group], nl, g.block_bytes,
-              ft.block_ids.data(), nb, kernel_dst, scratch, ctx.stream);
+              ft.block_ids.data(), nb, ids_dev, kernel_dst, scratch,
+              ctx.stream);
         } else {
           // zero-copy writes pinned host directly: atomics over PCIe are
           // pathological, keep the fused single-workgroup variant
           err = kvc_launch_gather_fp8(
               const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
               dev_layer_strides_[ft.group], nl, g.block_bytes,
-              ft.block_ids.data(), nb, kernel_dst, ctx.stream);
+              ft.block_ids.data(), nb, ids_dev, kernel_dst, ctx.stream);
         }
       } else {
         err = kvc_launch_gather(
             const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
             dev_layer_strides_[ft.group], nl, g.block_bytes,
-            ft.block_ids.data(), nb, kernel_dst, ctx.stream);
+            ft.block_ids.data(), nb, ids_dev, kernel_dst, ctx.stream);
       }
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       double t1 = now_s();
@@ -773,18 +793,19 @@ class StorageOffloadEngine {
       for (auto& f : futs) f.get();
       t_h2d = now_s() - h0;
       touch_atime(ft.path);
+      const int32_t* ids_dev = stage_ids(ctx, ft);
       hipError_t err =
           cfg_.serialize == Serialize::kFp8E4M3
               ? kvc_launch_scatter_fp8(
                     const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
                     dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, ctx.device_staging->ptr(),
-                    ctx.stream)
+                    ft.block_ids.data(), nb, ids_dev,
+                    ctx.device_staging->ptr(), ctx.stream)
               : kvc_launch_scatter(
                     const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
                     dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, ctx.device_staging->ptr(),
-                    ctx.stream);
+                    ft.block_ids.data(), nb, ids_dev,
+                    ctx.device_staging->ptr(), ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       double s0 = now_s();
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
@@ -829,16 +850,17 @@ class StorageOffloadEngine {
       const uint8_t* kernel_src = ctx.host_staging->device();
       double t2 = now_s();
       stats_inc([&](EngineStats& s) { s.t_h2d_ms += (t2 - t1) * 1e3; });
+      const int32_t* ids_dev = stage_ids(ctx, ft);
       hipError_t err =
           cfg_.serialize == Serialize::kFp8E4M3
               ? kvc_launch_scatter_fp8(
                     const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
                     dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, kernel_src, ctx.stream)
+                    ft.block_ids.data(), nb, ids_dev, kernel_src, ctx.stream)
               : kvc_launch_scatter(
                     const_cast<const void* const*>(dev_layer_ptrs_[ft.group]),
                     dev_layer_strides_[ft.group], nl, g.block_bytes,
-                    ft.block_ids.data(), nb, kernel_src, ctx.stream);
+                    ft.block_ids.data(), nb, ids_dev, kernel_src, ctx.stream);
       if (err != hipSuccess) throw HipError(hipGetErrorString(err));
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       double t3 = now_s();

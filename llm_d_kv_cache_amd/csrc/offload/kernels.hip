@@ -4,12 +4,22 @@
 // loop (csrc/storage/tensor_copier_kernels.cu:54-151) with an MI355X-first
 // design:
 //   - one launch covers ALL (block, layer) tiles of a file transfer;
-//   - 16 B/lane vectorized copies (dwordx4), wave64-shaped, grid-stride
-//     within each tile so the chip is filled regardless of tile count;
+//   - 16 B/lane vectorized copies (dwordx4), wave64-shaped;
+//   - LARGE tiles: 2D grid, tile on blockIdx.x (the unbounded dimension —
+//     consecutive tiles land on consecutive XCDs via the b%8 placement),
+//     slice on blockIdx.y, grid-stride inside the tile;
+//   - SMALL tiles (<=32 KiB — the 70B-TP8 / MLA geometry): a FLAT grid over
+//     the packed buffer with an LDS per-tile base-address table and
+//     magic-number division, 4 independent 16 B loads in flight per lane —
+//     the per-tile-workgroup mapping is launch/occupancy-bound there;
 //   - gather destination is contiguous staging, so the PCIe hop can run on
 //     the SDMA engines (hipMemcpyAsync) with zero CU occupancy, or the
 //     kernel can write device-mapped pinned host memory directly
 //     (zero-copy mode) — the engine chooses per config.
+//
+// Block ids travel by value (kernarg) up to kMaxBlocksPerFile; bigger
+// transfers pass a device pointer (ids_dev) instead, so there is no hard
+// ceiling on blocks-per-file from the kernel side.
 //
 // A batched prefix-hash kernel (FNV-64a over canonical CBOR, one lane per
 // sequence) accelerates bulk block-key computation for event floods.
@@ -19,34 +29,42 @@
 
 namespace kvo {
 
-constexpr int kMaxBlocksPerFile = 64;
+constexpr int kMaxBlocksPerFile = 128;   // by-value kernarg limit
+constexpr int kMaxBlocksPerFileDev = 4096;  // via ids_dev device pointer
 
 struct BlockList {
   int32_t ids[kMaxBlocksPerFile];
 };
 
-// ---- gather / scatter -------------------------------------------------------
+__device__ __forceinline__ int32_t block_id(const BlockList& bl,
+                                            const int32_t* __restrict__ ids_dev,
+                                            int bi) {
+  return ids_dev != nullptr ? ids_dev[bi] : bl.ids[bi];
+}
+
+// ---- gather / scatter (large tiles: 2D grid) --------------------------------
 // Tile (bi, l): block_bytes contiguous bytes.
 //   device side: layer_ptrs[l] + ids[bi] * layer_strides[l]
 //   staging side: dst + (bi * num_layers + l) * block_bytes
-// Grid: x = workgroups per tile (grid-stride inside the tile),
-//       y = tile index. 16-byte vectors; block_bytes % 16 == 0 (host-checked).
+// Grid: x = tile index, y = slices per tile (grid-stride inside the tile).
+// 16-byte vectors; block_bytes % 16 == 0 (host-checked).
 
 __global__ __launch_bounds__(256) void kvc_gather_blocks(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    uint64_t block_bytes, BlockList blocks, uint8_t* __restrict__ dst) {
-  const uint32_t tile = blockIdx.y;
+    uint64_t block_bytes, BlockList blocks,
+    const int32_t* __restrict__ ids_dev, uint8_t* __restrict__ dst) {
+  const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
   const uint4* __restrict__ src = reinterpret_cast<const uint4*>(
       static_cast<const uint8_t*>(layer_ptrs[l]) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   uint4* __restrict__ out =
       reinterpret_cast<uint4*>(dst + static_cast<uint64_t>(tile) * block_bytes);
   const uint64_t nvec = block_bytes / 16;
-  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
-  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.y) * blockDim.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.y) * blockDim.x + threadIdx.x;
        v < nvec; v += stride)
     out[v] = src[v];
 }
@@ -54,20 +72,119 @@ __global__ __launch_bounds__(256) void kvc_gather_blocks(
 __global__ __launch_bounds__(256) void kvc_scatter_blocks(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    uint64_t block_bytes, BlockList blocks, const uint8_t* __restrict__ src) {
-  const uint32_t tile = blockIdx.y;
+    uint64_t block_bytes, BlockList blocks,
+    const int32_t* __restrict__ ids_dev, const uint8_t* __restrict__ src) {
+  const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
   uint4* __restrict__ out = reinterpret_cast<uint4*>(
       static_cast<uint8_t*>(const_cast<void*>(layer_ptrs[l])) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   const uint4* __restrict__ in = reinterpret_cast<const uint4*>(
       src + static_cast<uint64_t>(tile) * block_bytes);
   const uint64_t nvec = block_bytes / 16;
-  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
-  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.y) * blockDim.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.y) * blockDim.x + threadIdx.x;
        v < nvec; v += stride)
     out[v] = in[v];
+}
+
+// ---- gather / scatter (small tiles: flat grid) ------------------------------
+// One flat index space over the packed buffer. Each workgroup first builds
+// the per-tile device base-address table in LDS (one 8-byte entry per
+// tile), then every lane copies kFlatUnroll independent 16-byte vectors —
+// enough loads in flight to hide HBM latency where the per-tile 2D grid
+// degenerates to one-load-per-lane workgroups. tile = v / vec_per_tile via
+// magic-number division (exact for v < 2^24; total vecs host-checked).
+
+constexpr int kFlatUnroll = 4;
+constexpr uint32_t kFlatMaxTiles = 4096;      // 32 KiB LDS table
+constexpr uint64_t kFlatMaxVecs = 1u << 24;   // magic-division validity bound
+
+struct MagicDiv {
+  uint64_t mul;
+  uint32_t shift;
+};
+
+// Granlund-Montgomery round-up division for 24-bit numerators.
+inline MagicDiv make_magic(uint32_t d) {
+  uint32_t log2d = 0;
+  while ((1u << log2d) < d) ++log2d;
+  const uint32_t s = 24 + log2d;
+  return {(((uint64_t)1 << s) + d - 1) / d, s};
+}
+
+__device__ __forceinline__ uint32_t magic_div(uint32_t v, uint64_t mul,
+                                              uint32_t shift) {
+  return static_cast<uint32_t>((static_cast<uint64_t>(v) * mul) >> shift);
+}
+
+__global__ __launch_bounds__(256) void kvc_gather_blocks_flat(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    BlockList blocks, const int32_t* __restrict__ ids_dev, uint32_t tiles,
+    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t div_mul,
+    uint32_t div_shift, uint8_t* __restrict__ dst) {
+  extern __shared__ uint64_t lds_src_base[];
+  for (uint32_t t = threadIdx.x; t < tiles; t += blockDim.x) {
+    const int l = t % num_layers;
+    const int bi = t / num_layers;
+    lds_src_base[t] =
+        reinterpret_cast<uint64_t>(layer_ptrs[l]) +
+        static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l];
+  }
+  __syncthreads();
+  uint4* __restrict__ out = reinterpret_cast<uint4*>(dst);
+  const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
+  uint4 val[kFlatUnroll];
+#pragma unroll
+  for (int j = 0; j < kFlatUnroll; ++j) {
+    const uint32_t v = base + j * blockDim.x;
+    if (v < total_vecs) {
+      const uint32_t tile = magic_div(v, div_mul, div_shift);
+      const uint32_t within = v - tile * vec_per_tile;
+      val[j] = reinterpret_cast<const uint4*>(lds_src_base[tile])[within];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < kFlatUnroll; ++j) {
+    const uint32_t v = base + j * blockDim.x;
+    if (v < total_vecs) out[v] = val[j];
+  }
+}
+
+__global__ __launch_bounds__(256) void kvc_scatter_blocks_flat(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    BlockList blocks, const int32_t* __restrict__ ids_dev, uint32_t tiles,
+    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t div_mul,
+    uint32_t div_shift, const uint8_t* __restrict__ src) {
+  extern __shared__ uint64_t lds_dst_base[];
+  for (uint32_t t = threadIdx.x; t < tiles; t += blockDim.x) {
+    const int l = t % num_layers;
+    const int bi = t / num_layers;
+    lds_dst_base[t] =
+        reinterpret_cast<uint64_t>(layer_ptrs[l]) +
+        static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l];
+  }
+  __syncthreads();
+  const uint4* __restrict__ in = reinterpret_cast<const uint4*>(src);
+  const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
+  uint4 val[kFlatUnroll];
+#pragma unroll
+  for (int j = 0; j < kFlatUnroll; ++j) {
+    const uint32_t v = base + j * blockDim.x;
+    if (v < total_vecs) val[j] = in[v];
+  }
+#pragma unroll
+  for (int j = 0; j < kFlatUnroll; ++j) {
+    const uint32_t v = base + j * blockDim.x;
+    if (v < total_vecs) {
+      const uint32_t tile = magic_div(v, div_mul, div_shift);
+      const uint32_t within = v - tile * vec_per_tile;
+      reinterpret_cast<uint4*>(lds_dst_base[tile])[within] = val[j];
+    }
+  }
 }
 
 // ---- batched prefix hashing -------------------------------------------------
@@ -136,32 +253,84 @@ inline dim3 copy_grid(uint32_t tiles, uint64_t block_bytes) {
   uint32_t want = tiles >= 2048 ? 1 : (2048 + tiles - 1) / tiles;
   uint32_t wg_per_tile = want < max_wg_per_tile ? want : max_wg_per_tile;
   if (wg_per_tile == 0) wg_per_tile = 1;
-  return dim3(wg_per_tile, tiles);
+  if (wg_per_tile > 65535) wg_per_tile = 65535;  // y-dimension limit
+  return dim3(tiles, wg_per_tile);
+}
+
+// Small-tile flat path applies when the LDS table fits and magic division
+// stays exact.
+inline bool use_flat_path(uint32_t tiles, uint64_t block_bytes) {
+  const uint64_t total_vecs = tiles * (block_bytes / 16);
+  return block_bytes <= 32768 && tiles <= kFlatMaxTiles &&
+         total_vecs < kFlatMaxVecs && total_vecs > 0;
+}
+
+struct FlatLaunch {
+  dim3 grid;
+  uint32_t vec_per_tile, total_vecs;
+  MagicDiv div;
+  size_t lds_bytes;
+};
+
+inline FlatLaunch flat_launch(uint32_t tiles, uint64_t block_bytes) {
+  FlatLaunch f;
+  f.vec_per_tile = static_cast<uint32_t>(block_bytes / 16);
+  f.total_vecs = tiles * f.vec_per_tile;
+  const uint32_t per_wg = 256 * kFlatUnroll;
+  f.grid = dim3((f.total_vecs + per_wg - 1) / per_wg);
+  f.div = make_magic(f.vec_per_tile);
+  f.lds_bytes = static_cast<size_t>(tiles) * sizeof(uint64_t);
+  return f;
 }
 
 extern "C" hipError_t kvc_launch_gather(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
-    int num_blocks, uint8_t* dst, hipStream_t stream) {
+    int num_blocks, const int32_t* ids_dev, uint8_t* dst, hipStream_t stream) {
   BlockList bl;
-  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
-  dim3 grid = copy_grid(static_cast<uint32_t>(num_blocks) * num_layers, block_bytes);
+  if (ids_dev == nullptr) {
+    if (num_blocks > kMaxBlocksPerFile) return hipErrorInvalidValue;
+    for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  }
+  const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
+  if (use_flat_path(tiles, block_bytes)) {
+    FlatLaunch f = flat_launch(tiles, block_bytes);
+    hipLaunchKernelGGL(kvc_gather_blocks_flat, f.grid, dim3(256), f.lds_bytes,
+                       stream, layer_ptrs_dev, layer_strides_dev, num_layers,
+                       bl, ids_dev, tiles, f.vec_per_tile, f.total_vecs,
+                       f.div.mul, f.div.shift, dst);
+    return hipGetLastError();
+  }
+  dim3 grid = copy_grid(tiles, block_bytes);
   hipLaunchKernelGGL(kvc_gather_blocks, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, dst);
+                     bl, ids_dev, dst);
   return hipGetLastError();
 }
 
 extern "C" hipError_t kvc_launch_scatter(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
-    int num_blocks, const uint8_t* src, hipStream_t stream) {
+    int num_blocks, const int32_t* ids_dev, const uint8_t* src,
+    hipStream_t stream) {
   BlockList bl;
-  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
-  dim3 grid = copy_grid(static_cast<uint32_t>(num_blocks) * num_layers, block_bytes);
+  if (ids_dev == nullptr) {
+    if (num_blocks > kMaxBlocksPerFile) return hipErrorInvalidValue;
+    for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  }
+  const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
+  if (use_flat_path(tiles, block_bytes)) {
+    FlatLaunch f = flat_launch(tiles, block_bytes);
+    hipLaunchKernelGGL(kvc_scatter_blocks_flat, f.grid, dim3(256), f.lds_bytes,
+                       stream, layer_ptrs_dev, layer_strides_dev, num_layers,
+                       bl, ids_dev, tiles, f.vec_per_tile, f.total_vecs,
+                       f.div.mul, f.div.shift, src);
+    return hipGetLastError();
+  }
+  dim3 grid = copy_grid(tiles, block_bytes);
   hipLaunchKernelGGL(kvc_scatter_blocks, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, src);
+                     bl, ids_dev, src);
   return hipGetLastError();
 }
 
@@ -185,9 +354,9 @@ namespace kvo {
 // halving PCIe and storage bytes. Packed slab layout is self-contained per
 // tile record (so partial-span tail-seek loads work):
 //   [ tile fp8 payload (block_bytes/2) | f32 scale ] x tiles
-// One workgroup per tile (amax needs a tile-wide reduction): pass 1
-// computes amax over the tile's bf16 elements (wave + LDS reduce), pass 2
-// converts with the derived scale. gfx950 is OCP e4m3fn (not fnuz).
+// gfx950 is OCP e4m3fn (not fnuz). The store side runs split amax+quantize
+// passes, the load side a chip-filling sliced dequant — all three grids
+// put the tile on blockIdx.x and slices on blockIdx.y like the raw path.
 
 namespace {
 
@@ -205,13 +374,6 @@ __device__ __forceinline__ uint16_t f32_to_bf16(float f) {
 
 constexpr float kFp8Max = 448.0f;  // e4m3fn max normal
 
-__device__ __forceinline__ uint8_t f32_to_fp8_e4m3(float x, float inv_scale) {
-  float v = x * inv_scale;
-  // clamp to representable range; builtin handles rounding + saturation
-  uint32_t packed = __builtin_amdgcn_cvt_pk_fp8_f32(v, 0.0f, 0, false);
-  return static_cast<uint8_t>(packed & 0xff);
-}
-
 __device__ __forceinline__ float fp8_e4m3_to_f32(uint8_t b) {
   return __builtin_amdgcn_cvt_f32_fp8(static_cast<uint32_t>(b), 0);
 }
@@ -219,27 +381,28 @@ __device__ __forceinline__ float fp8_e4m3_to_f32(uint8_t b) {
 }  // namespace
 
 // Pass 1 of the split fp8 gather: per-(tile, wg-slice) partial amax,
-// written densely to scratch[tile * gridDim.x + slice] (plain stores — no
+// written densely to scratch[tile * gridDim.y + slice] (plain stores — no
 // atomics, no zero-init memset, whose fixed ~32 us blit cost ate the
 // split's win). Pass 2 reduces the <=slices partials per tile at its
-// head. Grid: x = slices per tile, y = tile; both passes use the SAME
+// head. Grid: x = tile, y = slices per tile; both passes use the SAME
 // grid.
 __global__ __launch_bounds__(256) void kvc_fp8_amax(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    uint64_t block_bytes, BlockList blocks, float* __restrict__ scales) {
-  const uint32_t tile = blockIdx.y;
+    uint64_t block_bytes, BlockList blocks,
+    const int32_t* __restrict__ ids_dev, float* __restrict__ scales) {
+  const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
   const uint64_t n_elems = block_bytes / 2;
   const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
       static_cast<const uint8_t*>(layer_ptrs[l]) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   const uint4* __restrict__ vsrc = reinterpret_cast<const uint4*>(src);
   const uint64_t nvec = n_elems / 8;
-  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.y) * blockDim.x;
   float amax = 0.0f;
-  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.y) * blockDim.x + threadIdx.x;
        v < nvec; v += stride) {
     uint4 w = vsrc[v];
     const uint32_t* dw = reinterpret_cast<const uint32_t*>(&w);
@@ -256,7 +419,7 @@ __global__ __launch_bounds__(256) void kvc_fp8_amax(
   __syncthreads();
   if (threadIdx.x == 0) {
     float m = fmaxf(fmaxf(lds_max[0], lds_max[1]), fmaxf(lds_max[2], lds_max[3]));
-    scales[static_cast<uint64_t>(tile) * gridDim.x + blockIdx.x] = m;
+    scales[static_cast<uint64_t>(tile) * gridDim.y + blockIdx.y] = m;
   }
 }
 
@@ -267,28 +430,29 @@ __global__ __launch_bounds__(256) void kvc_fp8_quant(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
     uint64_t block_bytes, BlockList blocks,
-    const float* __restrict__ scales, uint8_t* __restrict__ dst) {
-  const uint32_t tile = blockIdx.y;
+    const int32_t* __restrict__ ids_dev, const float* __restrict__ scales,
+    uint8_t* __restrict__ dst) {
+  const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
   const uint64_t n_elems = block_bytes / 2;
   const uint64_t record = n_elems + 4;
   const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
       static_cast<const uint8_t*>(layer_ptrs[l]) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   const uint4* __restrict__ vsrc = reinterpret_cast<const uint4*>(src);
   uint8_t* __restrict__ payload = dst + static_cast<uint64_t>(tile) * record;
   float amax = 0.0f;
-  for (uint32_t g = 0; g < gridDim.x; ++g)
-    amax = fmaxf(amax, scales[static_cast<uint64_t>(tile) * gridDim.x + g]);
+  for (uint32_t g = 0; g < gridDim.y; ++g)
+    amax = fmaxf(amax, scales[static_cast<uint64_t>(tile) * gridDim.y + g]);
   if (amax <= 0.0f) amax = 1.0f;
   const float inv_scale = kFp8Max / amax;
-  if (blockIdx.x == 0 && threadIdx.x == 0)
+  if (blockIdx.y == 0 && threadIdx.x == 0)
     *reinterpret_cast<float*>(payload + n_elems) = amax / kFp8Max;
   uint2* __restrict__ vout = reinterpret_cast<uint2*>(payload);
   const uint64_t nvec = n_elems / 8;
-  const uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
-  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const uint64_t stride = static_cast<uint64_t>(gridDim.y) * blockDim.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.y) * blockDim.x + threadIdx.x;
        v < nvec; v += stride) {
     uint4 w = vsrc[v];
     const uint32_t* dw = reinterpret_cast<const uint32_t*>(&w);
@@ -312,14 +476,14 @@ __global__ __launch_bounds__(256) void kvc_fp8_quant(
   }
 }
 
-// Legacy single-workgroup-per-tile fused variant (kept for reference and
-// as the fallback when no scratch scale buffer is available).
+// Legacy single-workgroup-per-tile fused variant: the zero-copy path keeps
+// it (its destination is device-mapped pinned host memory, where the
+// split's scratch round trip would cross PCIe).
 __global__ __launch_bounds__(256) void kvc_gather_fp8(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    uint64_t block_bytes, BlockList blocks, int num_tiles,
-    uint8_t* __restrict__ dst) {
-  (void)num_tiles;
+    uint64_t block_bytes, BlockList blocks,
+    const int32_t* __restrict__ ids_dev, uint8_t* __restrict__ dst) {
   const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
@@ -327,7 +491,7 @@ __global__ __launch_bounds__(256) void kvc_gather_fp8(
   const uint64_t record = n_elems + 4;       // payload + f32 scale
   const uint16_t* __restrict__ src = reinterpret_cast<const uint16_t*>(
       static_cast<const uint8_t*>(layer_ptrs[l]) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   uint8_t* __restrict__ payload = dst + static_cast<uint64_t>(tile) * record;
   float* __restrict__ scale_out =
       reinterpret_cast<float*>(payload + n_elems);
@@ -384,12 +548,15 @@ __global__ __launch_bounds__(256) void kvc_gather_fp8(
   }
 }
 
+// Chip-filling dequantize+scatter: grid (tile, slices) like the raw path —
+// dequant needs no cross-workgroup reduction, the per-tile scale is read
+// from the record tail by every slice (uniform L2-hit load). Replaces the
+// one-workgroup-per-tile legacy that left the chip 25+% idle on loads.
 __global__ __launch_bounds__(256) void kvc_scatter_fp8(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    uint64_t block_bytes, BlockList blocks, int num_tiles,
-    const uint8_t* __restrict__ src) {
-  (void)num_tiles;
+    uint64_t block_bytes, BlockList blocks,
+    const int32_t* __restrict__ ids_dev, const uint8_t* __restrict__ src) {
   const uint32_t tile = blockIdx.x;
   const int l = tile % num_layers;
   const int bi = tile / num_layers;
@@ -397,7 +564,7 @@ __global__ __launch_bounds__(256) void kvc_scatter_fp8(
   const uint64_t record = n_elems + 4;
   uint16_t* __restrict__ out = reinterpret_cast<uint16_t*>(
       static_cast<uint8_t*>(const_cast<void*>(layer_ptrs[l])) +
-      static_cast<uint64_t>(blocks.ids[bi]) * layer_strides[l]);
+      static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l]);
   const uint8_t* __restrict__ payload =
       src + static_cast<uint64_t>(tile) * record;
   const float scale =
@@ -406,7 +573,9 @@ __global__ __launch_bounds__(256) void kvc_scatter_fp8(
   const uint2* __restrict__ vin = reinterpret_cast<const uint2*>(payload);
   uint4* __restrict__ vout = reinterpret_cast<uint4*>(out);
   const uint64_t nvec = n_elems / 8;
-  for (uint64_t v = threadIdx.x; v < nvec; v += blockDim.x) {
+  const uint64_t stride = static_cast<uint64_t>(gridDim.y) * blockDim.x;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.y) * blockDim.x + threadIdx.x;
+       v < nvec; v += stride) {
     uint2 w = vin[v];
     uint4 o;
     uint32_t* od = reinterpret_cast<uint32_t*>(&o);
@@ -426,45 +595,57 @@ __global__ __launch_bounds__(256) void kvc_scatter_fp8(
 extern "C" hipError_t kvc_launch_gather_fp8(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
-    int num_blocks, uint8_t* dst, hipStream_t stream) {
+    int num_blocks, const int32_t* ids_dev, uint8_t* dst, hipStream_t stream) {
   BlockList bl;
-  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  if (ids_dev == nullptr) {
+    if (num_blocks > kMaxBlocksPerFile) return hipErrorInvalidValue;
+    for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  }
   int tiles = num_blocks * num_layers;
   hipLaunchKernelGGL(kvc_gather_fp8, dim3(tiles), dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, tiles, dst);
+                     bl, ids_dev, dst);
   return hipGetLastError();
 }
 
 // Split fp8 gather: amax pass + quantize pass, both chip-filling. scales
-// is caller-provided device scratch of `num_blocks * num_layers` floats.
+// is caller-provided device scratch of `tiles * slices` floats.
 extern "C" hipError_t kvc_launch_gather_fp8_split(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
-    int num_blocks, uint8_t* dst, float* scales_scratch, hipStream_t stream) {
+    int num_blocks, const int32_t* ids_dev, uint8_t* dst,
+    float* scales_scratch, hipStream_t stream) {
   BlockList bl;
-  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  if (ids_dev == nullptr) {
+    if (num_blocks > kMaxBlocksPerFile) return hipErrorInvalidValue;
+    for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  }
   uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
   dim3 grid = copy_grid(tiles, block_bytes);
   hipLaunchKernelGGL(kvc_fp8_amax, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, scales_scratch);
+                     bl, ids_dev, scales_scratch);
   hipLaunchKernelGGL(kvc_fp8_quant, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, scales_scratch, dst);
+                     bl, ids_dev, scales_scratch, dst);
   return hipGetLastError();
 }
 
 extern "C" hipError_t kvc_launch_scatter_fp8(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
-    int num_blocks, const uint8_t* src, hipStream_t stream) {
+    int num_blocks, const int32_t* ids_dev, const uint8_t* src,
+    hipStream_t stream) {
   BlockList bl;
-  for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
-  int tiles = num_blocks * num_layers;
-  hipLaunchKernelGGL(kvc_scatter_fp8, dim3(tiles), dim3(256), 0, stream,
+  if (ids_dev == nullptr) {
+    if (num_blocks > kMaxBlocksPerFile) return hipErrorInvalidValue;
+    for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
+  }
+  uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
+  dim3 grid = copy_grid(tiles, block_bytes);
+  hipLaunchKernelGGL(kvc_scatter_fp8, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
-                     bl, tiles, src);
+                     bl, ids_dev, src);
   return hipGetLastError();
 }
 

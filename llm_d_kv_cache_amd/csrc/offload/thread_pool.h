@@ -30,6 +30,7 @@ struct WorkerCtx {
   hipStream_t stream = nullptr;           // null in host mode
   HostStaging* host_staging = nullptr;    // pinned (GPU mode) or aligned malloc
   DeviceStaging* device_staging = nullptr;  // HBM bounce; empty in host mode
+  int32_t* dev_ids = nullptr;  // device block-id buffer (> kernarg-limit paths)
 };
 
 enum class Priority { kHigh = 0, kNormal = 1 };
@@ -126,6 +127,8 @@ class IoThreadPool {
         KVO_HIP_CHECK(hipSetDevice(device_));
         pin_to_numa(i);
         KVO_HIP_CHECK(hipStreamCreateWithFlags(&ctx.stream, hipStreamNonBlocking));
+        // 16 KiB: block-id staging for transfers above the kernarg limit
+        KVO_HIP_CHECK(hipMalloc(&ctx.dev_ids, 4096 * sizeof(int32_t)));
       }
       host_staging = std::make_unique<HostStaging>(host_staging_bytes_, gpu_mode_,
                                                    mapped_host_staging_);
@@ -163,6 +166,7 @@ class IoThreadPool {
         KVO_LOG_ERROR("io task failed: %s", e.what());
       }
     }
+    if (ctx.dev_ids) (void)hipFree(ctx.dev_ids);
     if (ctx.stream) (void)hipStreamDestroy(ctx.stream);
   }
 

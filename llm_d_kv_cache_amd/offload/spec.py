@@ -53,8 +53,9 @@ class OffloadConnector:
             config.offloaded_block_tokens // int(b)
             for b in config.group_block_tokens
         ]
-        if max(blocks_per_file) > 64:
-            raise ValueError("offloaded chunk spans more than 64 engine blocks")
+        if max(blocks_per_file) > 4096:
+            raise ValueError(
+                "offloaded chunk spans more than 4096 engine blocks")
         eng_cfg = config.engine
         eng_cfg.gpu_blocks_per_file = max(blocks_per_file)
         self.engine = TorchOffloadEngine(groups, eng_cfg)

@@ -234,6 +234,25 @@ inline bool recv_command(int fd, std::string& name, std::string& payload) {
   return true;
 }
 
+// ZMTP 3.1 heartbeat (RFC 37): a peer configured with heartbeats sends
+// PING [2-byte TTL][context] and CLOSES the connection unless a PONG
+// echoing the context comes back — reply or the whole event stream dies
+// silently. Returns true when the frame was a PING it answered.
+inline bool maybe_pong(int fd, const Frame& f) {
+  if (!f.command || f.body.empty()) return false;
+  const uint8_t nlen = static_cast<uint8_t>(f.body[0]);
+  if (f.body.size() < 1u + nlen || f.body.compare(1, nlen, "PING") != 0)
+    return false;
+  std::string context;
+  if (f.body.size() > 1u + nlen + 2) context = f.body.substr(1 + nlen + 2);
+  std::string body;
+  body.push_back(4);
+  body += "PONG";
+  body += context;
+  send_frame(fd, body.data(), body.size(), /*more=*/false, /*command=*/true);
+  return true;
+}
+
 // Metadata body: a Socket-Type property (READY / INITIATE payloads).
 inline std::string metadata_body(const std::string& socket_type) {
   std::string body;
@@ -533,6 +552,7 @@ class ZmtpPublisher {
     while (!closing_ && zmtp::recv_frame(peer->fd, f)) {
       std::lock_guard<std::mutex> pg(peer->mu);
       if (f.command) {
+        if (zmtp::maybe_pong(peer->fd, f)) continue;
         if (f.body.size() >= 1) {
           uint8_t nlen = static_cast<uint8_t>(f.body[0]);
           if (f.body.size() >= 1u + nlen) {
@@ -664,7 +684,10 @@ class ZmtpSubscriber {
     std::vector<zmtp::Frame> parts;
     zmtp::Frame f;
     while (!closing_ && zmtp::recv_frame(fd, f)) {
-      if (f.command) continue;  // PING etc: ignored
+      if (f.command) {
+        zmtp::maybe_pong(fd, f);  // heartbeat peers close without a PONG
+        continue;
+      }
       parts.push_back(f);
       if (f.more) continue;
       deliver(parts);

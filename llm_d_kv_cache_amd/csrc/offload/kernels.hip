@@ -119,21 +119,45 @@ __device__ __forceinline__ uint32_t magic_div(uint32_t v, uint64_t mul,
   return static_cast<uint32_t>((static_cast<uint64_t>(v) * mul) >> shift);
 }
 
+// Per-workgroup LDS tables hold the INPUTS (per-layer {ptr, stride} and
+// the block-id list — <=2 KB, one strided build pass), never the per-tile
+// product: rebuilding a tiles-sized table in every workgroup costs as much
+// as the payload at this size, and per-lane dynamic indexing of a
+// by-value kernarg array lowers to scratch. The per-vector tile mapping
+// is two magic divisions + two LDS reads, hidden under the HBM traffic.
+__device__ __forceinline__ void build_flat_tables(
+    const void* const* __restrict__ layer_ptrs,
+    const uint64_t* __restrict__ layer_strides, int num_layers,
+    int num_blocks, const BlockList& blocks,
+    const int32_t* __restrict__ ids_dev, uint64_t* lptr, uint64_t* lstride,
+    int32_t* lids) {
+  for (int t = threadIdx.x; t < num_layers; t += blockDim.x) {
+    lptr[t] = reinterpret_cast<uint64_t>(layer_ptrs[t]);
+    lstride[t] = layer_strides[t];
+  }
+  if (ids_dev != nullptr) {
+    for (int t = threadIdx.x; t < num_blocks; t += blockDim.x)
+      lids[t] = ids_dev[t];
+  } else {
+    for (int t = threadIdx.x; t < num_blocks; t += blockDim.x)
+      lids[t] = blocks.ids[t];
+  }
+  __syncthreads();
+}
+
 __global__ __launch_bounds__(256) void kvc_gather_blocks_flat(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    BlockList blocks, const int32_t* __restrict__ ids_dev, uint32_t tiles,
-    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t div_mul,
-    uint32_t div_shift, uint8_t* __restrict__ dst) {
-  extern __shared__ uint64_t lds_src_base[];
-  for (uint32_t t = threadIdx.x; t < tiles; t += blockDim.x) {
-    const int l = t % num_layers;
-    const int bi = t / num_layers;
-    lds_src_base[t] =
-        reinterpret_cast<uint64_t>(layer_ptrs[l]) +
-        static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l];
-  }
-  __syncthreads();
+    int num_blocks, BlockList blocks, const int32_t* __restrict__ ids_dev,
+    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t mul_vpt,
+    uint32_t sh_vpt, uint64_t mul_nl, uint32_t sh_nl,
+    uint8_t* __restrict__ dst) {
+  extern __shared__ uint64_t lds_tab[];
+  uint64_t* lptr = lds_tab;
+  uint64_t* lstride = lptr + num_layers;
+  int32_t* lids = reinterpret_cast<int32_t*>(lstride + num_layers);
+  build_flat_tables(layer_ptrs, layer_strides, num_layers, num_blocks, blocks,
+                    ids_dev, lptr, lstride, lids);
   uint4* __restrict__ out = reinterpret_cast<uint4*>(dst);
   const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
   uint4 val[kFlatUnroll];
@@ -141,9 +165,14 @@ __global__ __launch_bounds__(256) void kvc_gather_blocks_flat(
   for (int j = 0; j < kFlatUnroll; ++j) {
     const uint32_t v = base + j * blockDim.x;
     if (v < total_vecs) {
-      const uint32_t tile = magic_div(v, div_mul, div_shift);
+      const uint32_t tile = magic_div(v, mul_vpt, sh_vpt);
       const uint32_t within = v - tile * vec_per_tile;
-      val[j] = reinterpret_cast<const uint4*>(lds_src_base[tile])[within];
+      const uint32_t bi = magic_div(tile, mul_nl, sh_nl);
+      const uint32_t l = tile - bi * num_layers;
+      const uint64_t src = lptr[l] +
+                           static_cast<uint64_t>(lids[bi]) * lstride[l] +
+                           static_cast<uint64_t>(within) * 16;
+      val[j] = *reinterpret_cast<const uint4*>(src);
     }
   }
 #pragma unroll
@@ -156,18 +185,16 @@ __global__ __launch_bounds__(256) void kvc_gather_blocks_flat(
 __global__ __launch_bounds__(256) void kvc_scatter_blocks_flat(
     const void* const* __restrict__ layer_ptrs,
     const uint64_t* __restrict__ layer_strides, int num_layers,
-    BlockList blocks, const int32_t* __restrict__ ids_dev, uint32_t tiles,
-    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t div_mul,
-    uint32_t div_shift, const uint8_t* __restrict__ src) {
-  extern __shared__ uint64_t lds_dst_base[];
-  for (uint32_t t = threadIdx.x; t < tiles; t += blockDim.x) {
-    const int l = t % num_layers;
-    const int bi = t / num_layers;
-    lds_dst_base[t] =
-        reinterpret_cast<uint64_t>(layer_ptrs[l]) +
-        static_cast<uint64_t>(block_id(blocks, ids_dev, bi)) * layer_strides[l];
-  }
-  __syncthreads();
+    int num_blocks, BlockList blocks, const int32_t* __restrict__ ids_dev,
+    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t mul_vpt,
+    uint32_t sh_vpt, uint64_t mul_nl, uint32_t sh_nl,
+    const uint8_t* __restrict__ src) {
+  extern __shared__ uint64_t lds_tab[];
+  uint64_t* lptr = lds_tab;
+  uint64_t* lstride = lptr + num_layers;
+  int32_t* lids = reinterpret_cast<int32_t*>(lstride + num_layers);
+  build_flat_tables(layer_ptrs, layer_strides, num_layers, num_blocks, blocks,
+                    ids_dev, lptr, lstride, lids);
   const uint4* __restrict__ in = reinterpret_cast<const uint4*>(src);
   const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
   uint4 val[kFlatUnroll];
@@ -180,9 +207,14 @@ __global__ __launch_bounds__(256) void kvc_scatter_blocks_flat(
   for (int j = 0; j < kFlatUnroll; ++j) {
     const uint32_t v = base + j * blockDim.x;
     if (v < total_vecs) {
-      const uint32_t tile = magic_div(v, div_mul, div_shift);
+      const uint32_t tile = magic_div(v, mul_vpt, sh_vpt);
       const uint32_t within = v - tile * vec_per_tile;
-      reinterpret_cast<uint4*>(lds_dst_base[tile])[within] = val[j];
+      const uint32_t bi = magic_div(tile, mul_nl, sh_nl);
+      const uint32_t l = tile - bi * num_layers;
+      const uint64_t dst = lptr[l] +
+                           static_cast<uint64_t>(lids[bi]) * lstride[l] +
+                           static_cast<uint64_t>(within) * 16;
+      *reinterpret_cast<uint4*>(dst) = val[j];
     }
   }
 }
@@ -268,18 +300,21 @@ inline bool use_flat_path(uint32_t tiles, uint64_t block_bytes) {
 struct FlatLaunch {
   dim3 grid;
   uint32_t vec_per_tile, total_vecs;
-  MagicDiv div;
+  MagicDiv div_vpt, div_nl;
   size_t lds_bytes;
 };
 
-inline FlatLaunch flat_launch(uint32_t tiles, uint64_t block_bytes) {
+inline FlatLaunch flat_launch(uint32_t tiles, uint64_t block_bytes,
+                              int num_layers, int num_blocks) {
   FlatLaunch f;
   f.vec_per_tile = static_cast<uint32_t>(block_bytes / 16);
   f.total_vecs = tiles * f.vec_per_tile;
   const uint32_t per_wg = 256 * kFlatUnroll;
   f.grid = dim3((f.total_vecs + per_wg - 1) / per_wg);
-  f.div = make_magic(f.vec_per_tile);
-  f.lds_bytes = static_cast<size_t>(tiles) * sizeof(uint64_t);
+  f.div_vpt = make_magic(f.vec_per_tile);
+  f.div_nl = make_magic(static_cast<uint32_t>(num_layers));
+  f.lds_bytes = 2 * num_layers * sizeof(uint64_t) +
+                ((num_blocks + 1) & ~1) * sizeof(int32_t);
   return f;
 }
 
@@ -294,11 +329,12 @@ extern "C" hipError_t kvc_launch_gather(
   }
   const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
   if (use_flat_path(tiles, block_bytes)) {
-    FlatLaunch f = flat_launch(tiles, block_bytes);
+    FlatLaunch f = flat_launch(tiles, block_bytes, num_layers, num_blocks);
     hipLaunchKernelGGL(kvc_gather_blocks_flat, f.grid, dim3(256), f.lds_bytes,
                        stream, layer_ptrs_dev, layer_strides_dev, num_layers,
-                       bl, ids_dev, tiles, f.vec_per_tile, f.total_vecs,
-                       f.div.mul, f.div.shift, dst);
+                       num_blocks, bl, ids_dev, f.vec_per_tile, f.total_vecs,
+                       f.div_vpt.mul, f.div_vpt.shift, f.div_nl.mul,
+                       f.div_nl.shift, dst);
     return hipGetLastError();
   }
   dim3 grid = copy_grid(tiles, block_bytes);
@@ -320,11 +356,12 @@ extern "C" hipError_t kvc_launch_scatter(
   }
   const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
   if (use_flat_path(tiles, block_bytes)) {
-    FlatLaunch f = flat_launch(tiles, block_bytes);
+    FlatLaunch f = flat_launch(tiles, block_bytes, num_layers, num_blocks);
     hipLaunchKernelGGL(kvc_scatter_blocks_flat, f.grid, dim3(256), f.lds_bytes,
                        stream, layer_ptrs_dev, layer_strides_dev, num_layers,
-                       bl, ids_dev, tiles, f.vec_per_tile, f.total_vecs,
-                       f.div.mul, f.div.shift, src);
+                       num_blocks, bl, ids_dev, f.vec_per_tile, f.total_vecs,
+                       f.div_vpt.mul, f.div_vpt.shift, f.div_nl.mul,
+                       f.div_nl.shift, src);
     return hipGetLastError();
   }
   dim3 grid = copy_grid(tiles, block_bytes);

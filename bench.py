@@ -79,6 +79,123 @@ def pick_root():
     return os.path.join("/tmp", "kvcache_bench")
 
 
+def find_nvme_root(min_free_gb=20):
+    """A writable filesystem backed by a real block device (preferring
+    NVMe), for the storage-tier-in-the-loop aux bench. Returns (path,
+    device, is_nvme) or None — tmpfs/overlay mounts never qualify."""
+    best = None
+    try:
+        with open("/proc/mounts") as f:
+            for line in f:
+                parts = line.split()
+                if len(parts) < 3:
+                    continue
+                dev, mnt, fstype = parts[0], parts[1], parts[2]
+                if fstype not in ("ext4", "xfs", "btrfs", "f2fs"):
+                    continue
+                if not os.access(mnt, os.W_OK):
+                    continue
+                try:
+                    free = shutil.disk_usage(mnt).free
+                except OSError:
+                    continue
+                if free < min_free_gb * 1024**3:
+                    continue
+                score = ("nvme" in dev, free)
+                if best is None or score > best[0]:
+                    best = (score, mnt, dev)
+    except OSError:
+        return None
+    if best is None:
+        return None
+    return os.path.join(best[1], "kvcache_bench_nvme"), best[2], best[0][0]
+
+
+def bench_offload_variant(group, root, host_cache_gb, direct_io, steps,
+                          local_rank, io_threads, gpu):
+    """Short sequential store->load cycle used for the aux variants (cache
+    disabled / real NVMe): every timed byte crosses the storage tier the
+    variant names. Returns GB/s and the phase breakdown."""
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        StorageToGPUHandler,
+        TorchOffloadEngine,
+    )
+
+    shutil.rmtree(root, ignore_errors=True)
+    os.makedirs(root, exist_ok=True)
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=io_threads,
+                            gpu_blocks_per_file=BLOCKS_PER_FILE,
+                            copy_path="staged" if gpu else "host",
+                            host_cache_bytes=int(host_cache_gb * 1024**3),
+                            direct_io=direct_io, device=local_rank),
+    )
+    mapper = FileMapper(root, KVCacheLayoutConfig(model=MODEL))
+    store = GPUToStorageHandler(eng, mapper, [BLOCKS_PER_FILE])
+    load = StorageToGPUHandler(eng, mapper, [BLOCKS_PER_FILE])
+    blocks_per_step = FILES_PER_STEP * BLOCKS_PER_FILE
+    step_bytes = blocks_per_step * len(group) * BLOCK_BYTES
+
+    def one_step(sid, timed):
+        base = sid * FILES_PER_STEP + 1
+        hashes = list(range(base, base + FILES_PER_STEP))
+        ids = list(range(blocks_per_step))
+        n = 0
+        for i in range(0, FILES_PER_STEP, 8):
+            store.transfer_async(
+                hashes[i:i + 8],
+                {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
+            n += 1
+        done = 0
+        while done < n:
+            done += len(store.get_finished())
+            time.sleep(0.0002)
+        for i in range(0, FILES_PER_STEP, 8):
+            load.transfer_async(
+                hashes[i:i + 8],
+                {0: ids[i * BLOCKS_PER_FILE:(i + 8) * BLOCKS_PER_FILE]})
+        done = 0
+        while done < n:
+            done += len(load.get_finished())
+            time.sleep(0.0002)
+        # delete the older generation so disk usage stays bounded
+        prev = (sid - 1) * FILES_PER_STEP + 1
+        if sid > -1:
+            for h in range(prev, prev + FILES_PER_STEP):
+                try:
+                    os.unlink(mapper.file_name(h, 0))
+                except OSError:
+                    pass
+
+    try:
+        one_step(-1, timed=False)  # warmup
+        import torch
+        if gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for s in range(steps):
+            one_step(s, timed=True)
+        if gpu:
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        stats = eng.stats()
+        return {
+            "GBps": round(2 * step_bytes * steps / dt / 1e9, 2),
+            "ms_per_step": round(dt / steps * 1e3, 1),
+            "steps": steps,
+            "host_cache_hits": stats.host_cache_hits,
+            "files_written": stats.files_written,
+            "files_read": stats.files_read,
+        }
+    finally:
+        shutil.rmtree(root, ignore_errors=True)
+
+
 def bench_control_plane():
     """Score req/s + p50, ingest events/s (CPU-side, per rank)."""
     import numpy as np
@@ -521,6 +638,34 @@ def main():
     aux = bench_control_plane() if rank == 0 else None
     if aux is not None and peer_aux is not None:
         aux["peer_xgmi"] = peer_aux
+
+    # Storage-tier-in-the-loop variants: the headline's
+    # DRAM-tier cache serves most loads, so report (a) a cache-disabled run
+    # where every load reads the file and (b) a real-NVMe O_DIRECT run when
+    # the box has one — probed honestly, skipped honestly.
+    if aux is not None and world == 1 and gpu:
+        try:
+            aux["no_cache"] = bench_offload_variant(
+                group, os.path.join(root, "nocache"), host_cache_gb=0,
+                direct_io=False, steps=4, local_rank=local_rank,
+                io_threads=args.io_threads, gpu=gpu)
+        except Exception as e:
+            aux["no_cache"] = {"error": str(e)}
+        nvme = find_nvme_root()
+        if nvme is None:
+            aux["nvme"] = {"skipped": "no NVMe/block-device filesystem "
+                                      "with >=20 GB free on this box"}
+        else:
+            nvme_root, nvme_dev, is_nvme = nvme
+            try:
+                aux["nvme"] = bench_offload_variant(
+                    group, nvme_root, host_cache_gb=0, direct_io=True,
+                    steps=3, local_rank=local_rank,
+                    io_threads=args.io_threads, gpu=gpu)
+                aux["nvme"]["device"] = nvme_dev
+                aux["nvme"]["is_nvme"] = is_nvme
+            except Exception as e:
+                aux["nvme"] = {"error": str(e), "device": nvme_dev}
 
     shutil.rmtree(rank_root, ignore_errors=True)
     if rank == 0:

@@ -82,6 +82,57 @@ void bench_threads(int nthreads) {
          nthreads, N * (double)ITERS * nthreads / dt / 1e9);
 }
 
+// Full-duplex: one thread streams D2H, another H2D, own streams + own
+// buffers. Decides whether ~50 GB/s TOTAL in the offload bench is an
+// orchestration gap or the platform's duplex ceiling (both directions are
+// blit kernels here, not SDMA — see profiles/r01_offload_profile.md).
+void bench_duplex(int streams_per_dir) {
+  struct Dir {
+    double gbps = 0;
+  } res[2];
+  std::vector<std::thread> ts;
+  double t0 = now();
+  for (int dir = 0; dir < 2; ++dir) {
+    ts.emplace_back([dir, streams_per_dir, &res] {
+      std::vector<void*> d(streams_per_dir), h(streams_per_dir);
+      std::vector<hipStream_t> s(streams_per_dir);
+      for (int i = 0; i < streams_per_dir; ++i) {
+        if (hipMalloc(&d[i], N) != hipSuccess) return;
+        if (hipHostMalloc(&h[i], N, hipHostMallocPortable) != hipSuccess)
+          return;
+        hipStreamCreateWithFlags(&s[i], hipStreamNonBlocking);
+      }
+      auto kind = dir == 0 ? hipMemcpyDeviceToHost : hipMemcpyHostToDevice;
+      // warmup
+      for (int i = 0; i < streams_per_dir; ++i) {
+        hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], N,
+                       kind, s[i]);
+        hipStreamSynchronize(s[i]);
+      }
+      double t1 = now();
+      for (int it = 0; it < ITERS; ++it) {
+        for (int i = 0; i < streams_per_dir; ++i)
+          hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], N,
+                         kind, s[i]);
+        for (int i = 0; i < streams_per_dir; ++i) hipStreamSynchronize(s[i]);
+      }
+      double dt = now() - t1;
+      res[dir].gbps = N * (double)ITERS * streams_per_dir / dt / 1e9;
+      for (int i = 0; i < streams_per_dir; ++i) {
+        hipStreamDestroy(s[i]);
+        hipHostFree(h[i]);
+        hipFree(d[i]);
+      }
+    });
+  }
+  for (auto& t : ts) t.join();
+  double wall = now() - t0;
+  printf("duplex %d stream/dir: D2H %.1f + H2D %.1f = %.1f GB/s total "
+         "(wall %.2fs)\n",
+         streams_per_dir, res[0].gbps, res[1].gbps, res[0].gbps + res[1].gbps,
+         wall);
+}
+
 void run_file_bench();
 int main() {
   int count = 0;
@@ -98,6 +149,8 @@ int main() {
   bench("WithStream + default pinned", hipHostMallocDefault, true, true);
   bench("sync hipMemcpy + default pinned", hipHostMallocDefault, false, true, true);
   bench_threads(4);
+  bench_duplex(1);
+  bench_duplex(2);
   run_file_bench();
   return 0;
 }

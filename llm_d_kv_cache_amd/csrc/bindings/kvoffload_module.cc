@@ -151,6 +151,7 @@ PYBIND11_MODULE(_kvoffload, m) {
            })
       .def("wait_job", &StorageOffloadEngine::wait_job,
            py::call_guard<py::gil_scoped_release>(), py::arg("job_id"))
+      .def("cancel_job", &StorageOffloadEngine::cancel_job, py::arg("job_id"))
       .def(
           "host_cache_read",
           [](kvo::StorageOffloadEngine& e, const std::string& path,

@@ -352,6 +352,18 @@ class StorageOffloadEngine {
     return out;
   }
 
+  // Non-blocking cancel: queued tasks of the job bail when dequeued.
+  // Returns false when the job already finished. Preemption sets flags for
+  // every affected job FIRST, then waits — a combined cancel+wait per job
+  // always loses the race against the worker's next dequeue.
+  bool cancel_job(int64_t id) {
+    std::lock_guard<std::mutex> g(jobs_mu_);
+    auto it = jobs_.find(id);
+    if (it == jobs_.end()) return false;
+    it->second->cancelled.store(true, std::memory_order_release);
+    return true;
+  }
+
   // Cancel a job's queued tasks (preemption) and wait for in-flight ones.
   // Returns true when every task that DID run succeeded.
   bool wait_job(int64_t id) {

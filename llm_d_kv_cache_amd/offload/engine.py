@@ -161,6 +161,11 @@ class TorchOffloadEngine:
     def wait_job(self, job_id: int) -> bool:
         return self._engine.wait_job(job_id)
 
+    def cancel_job(self, job_id: int) -> bool:
+        """Non-blocking cancel: queued tasks bail when dequeued. Set flags
+        for every job being preempted first, then wait_job each."""
+        return self._engine.cancel_job(job_id)
+
     def stats(self):
         return self._engine.stats()
 

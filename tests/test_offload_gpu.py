@@ -415,3 +415,44 @@ def test_gpu_dev_ids_path_bit_exact(tmp_path):
     assert wait_finished(load)[0].success
     for t, o in zip(group, orig):
         assert torch.equal(t[ids], o[ids])
+
+
+def test_gpu_read_latency_percentiles_under_saturation(tmp_path):
+    """GPU twin of the host-path percentile QoS test: reads submitted into
+    a saturated write queue keep bounded p50/p99 and a <5x tail ratio
+    (reference test_priority_queue.py:257)."""
+    group = make_group()
+    eng = TorchOffloadEngine(
+        [group],
+        OffloadEngineConfig(io_threads=4, gpu_blocks_per_file=BPF,
+                            copy_path="staged"),
+    )
+    mapper = FileMapper(str(tmp_path), KVCacheLayoutConfig(model="gpu-pct"))
+    store = GPUToStorageHandler(eng, mapper, [BPF])
+    load = StorageToGPUHandler(eng, mapper, [BPF])
+
+    n_read_files = 5
+    for i in range(n_read_files):
+        store.transfer_async([i], {0: list(range(i * BPF, (i + 1) * BPF))})
+    assert len(wait_finished(store, n_read_files)) == n_read_files
+
+    for i in range(60):
+        store.transfer_async([1000 + i], {0: list(range(BPF))})
+    lats = []
+    for i in range(15):
+        f = i % n_read_files
+        t0 = time.time()
+        load.transfer_async([f], {0: list(range(f * BPF, (f + 1) * BPF))})
+        assert wait_finished(load, 1)[0].success
+        lats.append(time.time() - t0)
+        time.sleep(0.01)
+    backlog = eng.native.pending_writes
+    wait_finished(store, 60, timeout=60)
+    lats.sort()
+    p50 = lats[len(lats) // 2]
+    p99 = lats[-1]
+    tail = p99 / max(p50, 0.005)
+    assert backlog > 0, "write backlog drained before the reads"
+    assert p50 < 0.5, f"GPU read p50 {p50:.3f}s under write saturation"
+    assert p99 < 1.5, f"GPU read p99 {p99:.3f}s under write saturation"
+    assert tail < 5.0, f"GPU tail ratio {tail:.1f}x under saturation"

@@ -108,8 +108,13 @@ try:  # pragma: no cover - exercised only inside a vLLM worker
         engine. Mirrors the reference plugin's extra-config keys
         (shared_storage_path, offloaded_block_size, ...)."""
 
-        def __init__(self, vllm_config):
-            super().__init__(vllm_config)
+        def __init__(self, vllm_config, kv_cache_config=None):
+            # newer vLLM passes (vllm_config, kv_cache_config); older takes
+            # vllm_config only — support both base arities
+            try:
+                super().__init__(vllm_config, kv_cache_config)
+            except TypeError:
+                super().__init__(vllm_config)
             extra = self.extra_config
             parallel = vllm_config.parallel_config
             layout = KVCacheLayoutConfig(
@@ -138,6 +143,8 @@ try:  # pragma: no cover - exercised only inside a vLLM worker
             groups = [list(kv_caches.values())]
             cfg = self._connector_config
             cfg.group_block_tokens = (self.gpu_block_size,)
+            if groups[0] and not groups[0][0].is_cuda:
+                cfg.engine.copy_path = "host"  # CPU-stubbed test harness
             self._connector = OffloadConnector(groups, cfg)
             return self._connector.get_handlers()
 

@@ -288,8 +288,13 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
             svc.close()
         return {"ok": False, "error": err or "peer rank failed init"}
 
-    n_chunks = 32
     bpf = BLOCKS_PER_FILE
+    nb = int(group[0].shape[0])
+    # src chunks live in the head of the page pool, pull destinations in
+    # the tail — sized to the ACTUAL pool so small --device-blocks runs
+    # (CPU sims, tiny boxes) stay in range
+    n_chunks = max(1, min(32, nb // (2 * bpf)))
+    dst_base = nb - n_chunks * bpf
     src = (rank + 1) % world
     ok_local = True
     try:
@@ -303,7 +308,8 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
         return {"ok": False, "error": err or "peer rank failed register"}
 
     try:
-        svc.pull(0xE000 + src * 1000, 0, list(range(1024, 1024 + bpf)),
+        svc.pull(0xE000 + src * 1000, 0,
+                 list(range(dst_base, dst_base + bpf)),
                  src_rank=src).result(timeout=120)  # warmup
     except Exception as e:
         ok_local, err = False, f"warmup pull: {e}"
@@ -317,7 +323,7 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     try:
         futs = [
             svc.pull(0xE000 + src * 1000 + c, 0,
-                     list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
+                     list(range(dst_base + c * bpf, dst_base + (c + 1) * bpf)),
                      src_rank=src, timeout=120)
             for c in range(n_chunks)
         ]
@@ -344,7 +350,7 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     try:
         futs = [
             svc.pull(0xE000 + src * 1000 + c, 0,
-                     list(range(1024 + c * bpf, 1024 + (c + 1) * bpf)),
+                     list(range(dst_base + c * bpf, dst_base + (c + 1) * bpf)),
                      src_rank=src, timeout=120, fp8=True)
             for c in range(n8)
         ]
@@ -408,11 +414,24 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
     if world > 1:
+        # Fail FAST before rendezvous when this box cannot host the rank:
+        # a rank that dies inside init_process_group leaves the others
+        # hanging until the NCCL watchdog; exiting pre-rendezvous makes
+        # torchrun kill the job immediately.
+        if torch.cuda.is_available() and torch.cuda.device_count() <= local_rank:
+            print(f"[bench] rank {rank}: LOCAL_RANK {local_rank} >= "
+                  f"{torch.cuda.device_count()} visible GPUs — aborting "
+                  f"before rendezvous", file=sys.stderr, flush=True)
+            sys.exit(3)
+        from datetime import timedelta
+
         import torch.distributed as tdist
 
         dist = tdist
         backend = "nccl" if torch.cuda.is_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        # bounded: a wedged peer turns into a clean error, not a hang
+        dist.init_process_group(backend=backend,
+                                timeout=timedelta(seconds=300))
 
     gpu = torch.cuda.is_available()
     cpu_full = bool(os.environ.get("KVC_BENCH_CPU_FULL"))

@@ -53,3 +53,36 @@ def test_bench_single_rank_cpu_control_plane():
     d = json.loads([ln for ln in out.stdout.splitlines() if ln.startswith("{")][0])
     assert d["aux"]["score_req_s"] > 100
     assert d["aux"]["ingest_batches_s"] > 100
+
+
+@pytest.mark.timeout(600)
+def test_bench_world8_tiny_cpu(tmp_path):
+    """World-8 CPU sim of the driver's 8-GPU scaling run:
+    the full bench path — rank coordination, barriers, the peer-mesh
+    phase, max-over-ranks — at the scaling run's world size, tiny
+    geometry."""
+    env = dict(os.environ)
+    env.update({
+        "KVC_BENCH_CPU_FULL": "1",
+        "KVC_BENCH_TINY": "1",
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": "29573",
+    })
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29573", str(REPO / "bench.py"),
+         "--gpus", "8", "--steps", "1", "--warmup", "0",
+         "--root", str(tmp_path), "--device-blocks", "512",
+         "--io-threads", "1"],
+        capture_output=True, text=True, timeout=580, env=env, cwd=str(REPO),
+    )
+    assert out.returncode == 0, f"stdout:\n{out.stdout}\nstderr:\n{out.stderr}"
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "dp8"
+    assert d["value"] is not None and d["value"] > 0
+    peer = d["aux"]["peer_xgmi"]
+    assert peer["ok"] is True, peer
+    assert peer["pull_GBps_aggregate"] > 0

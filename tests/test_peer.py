@@ -310,6 +310,98 @@ def scenario_dram_tier(rank, svc, group):
         dist.barrier()
 
 
+def scenario_index_driven_peer_resolution(rank, svc, group):
+    """The full serving-path flow (SURVEY §2.5 'third StorageHandler-like
+    path'): KVEvents register a peer pod's blocks under the peer-gpu tier,
+    the scorer picks that pod, a pod->rank map drives the TieredKVLoader's
+    pull over the data plane (bit-compared), and chunks no longer
+    registered on the peer fall back to the storage tier or miss."""
+    import tempfile
+    import time
+
+    import torch.distributed as dist
+
+    from llm_d_kv_cache_amd.core import IndexerConfig, KVCacheIndexer
+    from llm_d_kv_cache_amd.events.publisher import (
+        block_stored_payload,
+        encode_batch,
+    )
+    from llm_d_kv_cache_amd.offload import (
+        FileMapper,
+        GPUToStorageHandler,
+        KVCacheLayoutConfig,
+        OffloadEngineConfig,
+        StorageToGPUHandler,
+        TorchOffloadEngine,
+    )
+    from llm_d_kv_cache_amd.peer.tiered import TieredKVLoader
+
+    MODEL = "m"
+    tokens = list(range(64))  # 4 hash blocks of 16 tokens = 1 offload chunk
+    POD_RANK = {"pod-r0": 0, "pod-r1": 1}
+
+    ix = KVCacheIndexer(IndexerConfig())
+    keys = ix.compute_block_keys(tokens, MODEL)
+    chunk_hash = keys[-1]  # chunk addressed by its last covered block key
+
+    if rank == 0:
+        # the peer holds the chunk's blocks in HBM (data plane registry)
+        svc.register_blocks(chunk_hash, 0, [0, 1, 2, 3])
+        golden = [group[0][[0, 1, 2, 3]].clone(), group[1][[0, 1, 2, 3]].clone()]
+        dist.broadcast(golden[0], src=0)
+        dist.broadcast(golden[1], src=0)
+        dist.barrier()   # rank 1 resolves
+        dist.barrier()   # done
+        assert svc.stats().pulls_served >= 1
+        return
+
+    golden = [torch.zeros(4, 4096, dtype=torch.uint8),
+              torch.zeros(4, 4096, dtype=torch.uint8)]  # two layers of group 0
+    dist.broadcast(golden[0], src=0)
+    dist.broadcast(golden[1], src=0)
+
+    # control plane: rank 0's engine announced the stores as KVEvents with
+    # the peer-gpu tier; this replica's pool ingests them
+    from llm_d_kv_cache_amd import ensure_native
+
+    k = ensure_native()
+    pool = k.EventPool(ix.token_processor, ix.index, 1)
+    pool.process(f"kv@pod-r0@{MODEL}", 0, encode_batch([
+        block_stored_payload(list(keys), None, tokens, 16, medium="peer-gpu")
+    ]))
+    scores = ix.score_tokens(tokens, MODEL)
+    assert scores == {"pod-r0": pytest.approx(4 * 0.95)}
+    best_pod = max(scores, key=scores.get)
+    assert POD_RANK[best_pod] == 0
+
+    # local storage tier for the fallback chunk
+    eng = TorchOffloadEngine(
+        [group], OffloadEngineConfig(io_threads=2, gpu_blocks_per_file=4,
+                                     copy_path="host"))
+    root = tempfile.mkdtemp(prefix="ixpeer_")
+    mapper = FileMapper(root, KVCacheLayoutConfig(model="ixpeer"))
+    store = GPUToStorageHandler(eng, mapper, [4])
+    load = StorageToGPUHandler(eng, mapper, [4])
+    CH_LOCAL, CH_GONE = 0x71, 0x72
+    store.transfer_async([CH_LOCAL], {0: [8, 9, 10, 11]})
+    deadline = time.time() + 10
+    while not store.get_finished() and time.time() < deadline:
+        time.sleep(0.01)
+
+    loader = TieredKVLoader(load_handler=load, peer_service=svc,
+                            peer_ranks=[POD_RANK[best_pod]])
+    # scored chunk: pulled from the peer's HBM over the data plane
+    assert loader.resolve(chunk_hash, [16, 17, 18, 19]) == "peer"
+    assert torch.equal(group[0][[16, 17, 18, 19]], golden[0])
+    assert torch.equal(group[1][[16, 17, 18, 19]], golden[1])
+    # chunk the peer never registered: served by the local storage tier
+    assert loader.resolve(CH_LOCAL, [20, 21, 22, 23]) == "storage"
+    # nowhere: miss -> the caller recomputes
+    assert loader.resolve(CH_GONE, [24]) == "miss"
+    dist.barrier()
+    dist.barrier()
+
+
 # ---- tests ------------------------------------------------------------------
 
 @pytest.mark.parametrize("scenario", [
@@ -324,6 +416,7 @@ def scenario_dram_tier(rank, svc, group):
     "scenario_invalid_pull_sizes",
     "scenario_cmd_error_resolves_future",
     "scenario_tiered_loader",
+    "scenario_index_driven_peer_resolution",
 ])
 def test_peer_migration(scenario, tmp_path):
     spawn2(scenario, tmp_path)

@@ -141,6 +141,7 @@ PYBIND11_MODULE(_kvcore, m) {
   py::class_<IndexStats>(m, "IndexStats")
       .def_readonly("admissions", &IndexStats::admissions)
       .def_readonly("evictions", &IndexStats::evictions)
+      .def_readonly("rejections", &IndexStats::rejections)
       .def_readonly("lookups", &IndexStats::lookups)
       .def_readonly("hits", &IndexStats::hits)
       .def_readonly("keys", &IndexStats::keys);

@@ -56,6 +56,7 @@ enum class KeyType { kEngine = 0, kRequest = 1 };
 struct IndexStats {
   uint64_t admissions = 0;
   uint64_t evictions = 0;
+  uint64_t rejections = 0;  // admission-control rejects (byte-budget mode)
   uint64_t lookups = 0;
   uint64_t hits = 0;
   uint64_t keys = 0;
@@ -127,6 +128,7 @@ class InMemoryIndex : public IndexBackend {
       if (it == sh.map.end()) continue;
       if (it->second.pods.empty()) break;  // chain break: present but empty
       sh.touch(it);
+      if (cfg_.max_bytes > 0) sh.freq.bump(key);  // lookups keep keys warm
       std::vector<PodEntry> pods;
       if (pod_filter.empty()) {
         pods = it->second.pods;
@@ -177,12 +179,27 @@ class InMemoryIndex : public IndexBackend {
       std::lock_guard<std::mutex> g(sh.mu);
       auto it = sh.map.find(key);
       if (it == sh.map.end()) {
+        if (cfg_.max_bytes > 0) {
+          sh.freq.bump(key);
+          // Admission control: a new key that would force eviction must be
+          // at least as warm as the LRU victim, or it is rejected — a scan
+          // flood of one-shot keys cannot wash out the looked-up set.
+          size_t shard_budget = cfg_.max_bytes / cfg_.shards;
+          if (sh.bytes + kKeyOverheadBytes + entries.size() * sizeof(PodEntry) >
+                  shard_budget &&
+              !sh.lru.empty() &&
+              sh.freq.estimate(key) < sh.freq.estimate(sh.lru.back())) {
+            rejections_.fetch_add(1, std::memory_order_relaxed);
+            continue;
+          }
+        }
         it = sh.map.emplace(key, KeyEntry{}).first;
         sh.lru.push_front(key);
         it->second.lru_it = sh.lru.begin();
         sh.bytes += kKeyOverheadBytes;
         if (sh.map.size() > shard_cap_) evict_lru_locked(sh);
       } else {
+        if (cfg_.max_bytes > 0) sh.freq.bump(key);
         sh.touch(it);
       }
       auto& pods = it->second.pods;
@@ -294,6 +311,7 @@ class InMemoryIndex : public IndexBackend {
     IndexStats s;
     s.admissions = admissions_.load(std::memory_order_relaxed);
     s.evictions = evictions_.load(std::memory_order_relaxed);
+    s.rejections = rejections_.load(std::memory_order_relaxed);
     s.lookups = lookups_.load(std::memory_order_relaxed);
     s.hits = hits_.load(std::memory_order_relaxed);
     for (const auto& sh : shards_) {
@@ -440,11 +458,42 @@ class InMemoryIndex : public IndexBackend {
     std::vector<PodEntry> pods;  // front = most recently added
     std::list<uint64_t>::iterator lru_it;
   };
+  // TinyLFU-style frequency sketch (byte-budget mode): two 8-bit counters
+  // per key (min estimate), halved every 8x-table-size bumps so history
+  // ages out. Plays ristretto's admission role (reference
+  // cost_aware_memory.go:41-52): a one-shot scan flood cannot evict keys
+  // that lookups keep hot.
+  struct FreqSketch {
+    std::vector<uint8_t> c;
+    uint64_t ops = 0;
+    static constexpr size_t kSlots = 8192;  // power of two
+
+    void ensure() {
+      if (c.empty()) c.assign(kSlots, 0);
+    }
+    static uint64_t h2(uint64_t k) { return k * 0x9E3779B97F4A7C15ull; }
+    void bump(uint64_t k) {
+      ensure();
+      const size_t i1 = k & (kSlots - 1), i2 = h2(k) & (kSlots - 1);
+      if (c[i1] < 255) c[i1]++;
+      if (i2 != i1 && c[i2] < 255) c[i2]++;  // colliding slots count once
+      if (++ops >= kSlots * 8) {
+        for (auto& x : c) x >>= 1;
+        ops = 0;
+      }
+    }
+    uint32_t estimate(uint64_t k) const {
+      if (c.empty()) return 0;
+      return std::min(c[k & (kSlots - 1)], c[h2(k) & (kSlots - 1)]);
+    }
+  };
+
   struct Shard {
     std::mutex mu;
     std::unordered_map<uint64_t, KeyEntry> map;
     std::list<uint64_t> lru;  // front = most recent
     size_t bytes = 0;  // approximate resident cost (byte-budget mode)
+    FreqSketch freq;   // admission sketch (byte-budget mode only)
 
     Shard() = default;
     Shard(const Shard&) {}
@@ -540,7 +589,8 @@ class InMemoryIndex : public IndexBackend {
   size_t shard_cap_;
   std::vector<Shard> shards_;
   std::vector<EngShard> eng_shards_;
-  std::atomic<uint64_t> admissions_{0}, evictions_{0}, lookups_{0}, hits_{0};
+  std::atomic<uint64_t> admissions_{0}, evictions_{0}, rejections_{0},
+      lookups_{0}, hits_{0};
 };
 
 }  // namespace kvc

@@ -216,3 +216,22 @@ def test_pool_survives_backend_outage(redis_pair):
     assert st.handler_failures == 3
     assert st.processed == 4
     pool.shutdown()
+
+
+def test_cost_aware_admission_protects_hot_keys():
+    """TinyLFU-style admission (ristretto's role, reference
+    cost_aware_memory.go:41-52): a one-shot scan flood cannot evict the
+    keys lookups keep hot — cold newcomers are rejected, not admitted at
+    the hot set's expense."""
+    idx = k.InMemoryIndex(shards=1, max_bytes=50 * (96 + 16))
+    hot = list(range(20))
+    for i in hot:
+        idx.add([], [i], [entry("a")])
+    for _ in range(30):  # repeated lookups warm the frequency sketch
+        idx.lookup(hot)
+    for i in range(10_000, 12_000):  # scan flood of one-shot keys
+        idx.add([], [i], [entry("b")])
+    found = idx.lookup(hot)
+    assert len(found) >= 15, f"hot set washed out: {len(found)}/20 survive"
+    st = idx.stats()
+    assert st.rejections > 0, "no admission rejects during the flood"

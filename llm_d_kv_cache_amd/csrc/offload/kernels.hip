@@ -8,10 +8,11 @@
 //   - LARGE tiles: 2D grid, tile on blockIdx.x (the unbounded dimension —
 //     consecutive tiles land on consecutive XCDs via the b%8 placement),
 //     slice on blockIdx.y, grid-stride inside the tile;
-//   - SMALL tiles (<=32 KiB — the 70B-TP8 / MLA geometry): a FLAT grid over
-//     the packed buffer with an LDS per-tile base-address table and
-//     magic-number division, 4 independent 16 B loads in flight per lane —
-//     the per-tile-workgroup mapping is launch/occupancy-bound there;
+//     (a flat-grid variant with magic-division tile lookup was measured
+//     SLOWER for small tiles than this 2D shape — per-vector mapping
+//     overhead at 2.3 TB/s vs 5.4-6.6 TB/s for the sliced 2D fp8 kernels
+//     at the same 8 KiB x 80-layer geometry; rocprof table in
+//     profiles/r02 — so every tile size uses the 2D grid);
 //   - gather destination is contiguous staging, so the PCIe hop can run on
 //     the SDMA engines (hipMemcpyAsync) with zero CU occupancy, or the
 //     kernel can write device-mapped pinned host memory directly
@@ -89,136 +90,6 @@ __global__ __launch_bounds__(256) void kvc_scatter_blocks(
     out[v] = in[v];
 }
 
-// ---- gather / scatter (small tiles: flat grid) ------------------------------
-// One flat index space over the packed buffer. Each workgroup first builds
-// the per-tile device base-address table in LDS (one 8-byte entry per
-// tile), then every lane copies kFlatUnroll independent 16-byte vectors —
-// enough loads in flight to hide HBM latency where the per-tile 2D grid
-// degenerates to one-load-per-lane workgroups. tile = v / vec_per_tile via
-// magic-number division (exact for v < 2^24; total vecs host-checked).
-
-constexpr int kFlatUnroll = 4;
-constexpr uint32_t kFlatMaxTiles = 4096;      // 32 KiB LDS table
-constexpr uint64_t kFlatMaxVecs = 1u << 24;   // magic-division validity bound
-
-struct MagicDiv {
-  uint64_t mul;
-  uint32_t shift;
-};
-
-// Granlund-Montgomery round-up division for 24-bit numerators.
-inline MagicDiv make_magic(uint32_t d) {
-  uint32_t log2d = 0;
-  while ((1u << log2d) < d) ++log2d;
-  const uint32_t s = 24 + log2d;
-  return {(((uint64_t)1 << s) + d - 1) / d, s};
-}
-
-__device__ __forceinline__ uint32_t magic_div(uint32_t v, uint64_t mul,
-                                              uint32_t shift) {
-  return static_cast<uint32_t>((static_cast<uint64_t>(v) * mul) >> shift);
-}
-
-// Per-workgroup LDS tables hold the INPUTS (per-layer {ptr, stride} and
-// the block-id list — <=2 KB, one strided build pass), never the per-tile
-// product: rebuilding a tiles-sized table in every workgroup costs as much
-// as the payload at this size, and per-lane dynamic indexing of a
-// by-value kernarg array lowers to scratch. The per-vector tile mapping
-// is two magic divisions + two LDS reads, hidden under the HBM traffic.
-__device__ __forceinline__ void build_flat_tables(
-    const void* const* __restrict__ layer_ptrs,
-    const uint64_t* __restrict__ layer_strides, int num_layers,
-    int num_blocks, const BlockList& blocks,
-    const int32_t* __restrict__ ids_dev, uint64_t* lptr, uint64_t* lstride,
-    int32_t* lids) {
-  for (int t = threadIdx.x; t < num_layers; t += blockDim.x) {
-    lptr[t] = reinterpret_cast<uint64_t>(layer_ptrs[t]);
-    lstride[t] = layer_strides[t];
-  }
-  if (ids_dev != nullptr) {
-    for (int t = threadIdx.x; t < num_blocks; t += blockDim.x)
-      lids[t] = ids_dev[t];
-  } else {
-    for (int t = threadIdx.x; t < num_blocks; t += blockDim.x)
-      lids[t] = blocks.ids[t];
-  }
-  __syncthreads();
-}
-
-__global__ __launch_bounds__(256) void kvc_gather_blocks_flat(
-    const void* const* __restrict__ layer_ptrs,
-    const uint64_t* __restrict__ layer_strides, int num_layers,
-    int num_blocks, BlockList blocks, const int32_t* __restrict__ ids_dev,
-    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t mul_vpt,
-    uint32_t sh_vpt, uint64_t mul_nl, uint32_t sh_nl,
-    uint8_t* __restrict__ dst) {
-  extern __shared__ uint64_t lds_tab[];
-  uint64_t* lptr = lds_tab;
-  uint64_t* lstride = lptr + num_layers;
-  int32_t* lids = reinterpret_cast<int32_t*>(lstride + num_layers);
-  build_flat_tables(layer_ptrs, layer_strides, num_layers, num_blocks, blocks,
-                    ids_dev, lptr, lstride, lids);
-  uint4* __restrict__ out = reinterpret_cast<uint4*>(dst);
-  const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
-  uint4 val[kFlatUnroll];
-#pragma unroll
-  for (int j = 0; j < kFlatUnroll; ++j) {
-    const uint32_t v = base + j * blockDim.x;
-    if (v < total_vecs) {
-      const uint32_t tile = magic_div(v, mul_vpt, sh_vpt);
-      const uint32_t within = v - tile * vec_per_tile;
-      const uint32_t bi = magic_div(tile, mul_nl, sh_nl);
-      const uint32_t l = tile - bi * num_layers;
-      const uint64_t src = lptr[l] +
-                           static_cast<uint64_t>(lids[bi]) * lstride[l] +
-                           static_cast<uint64_t>(within) * 16;
-      val[j] = *reinterpret_cast<const uint4*>(src);
-    }
-  }
-#pragma unroll
-  for (int j = 0; j < kFlatUnroll; ++j) {
-    const uint32_t v = base + j * blockDim.x;
-    if (v < total_vecs) out[v] = val[j];
-  }
-}
-
-__global__ __launch_bounds__(256) void kvc_scatter_blocks_flat(
-    const void* const* __restrict__ layer_ptrs,
-    const uint64_t* __restrict__ layer_strides, int num_layers,
-    int num_blocks, BlockList blocks, const int32_t* __restrict__ ids_dev,
-    uint32_t vec_per_tile, uint32_t total_vecs, uint64_t mul_vpt,
-    uint32_t sh_vpt, uint64_t mul_nl, uint32_t sh_nl,
-    const uint8_t* __restrict__ src) {
-  extern __shared__ uint64_t lds_tab[];
-  uint64_t* lptr = lds_tab;
-  uint64_t* lstride = lptr + num_layers;
-  int32_t* lids = reinterpret_cast<int32_t*>(lstride + num_layers);
-  build_flat_tables(layer_ptrs, layer_strides, num_layers, num_blocks, blocks,
-                    ids_dev, lptr, lstride, lids);
-  const uint4* __restrict__ in = reinterpret_cast<const uint4*>(src);
-  const uint32_t base = blockIdx.x * (blockDim.x * kFlatUnroll) + threadIdx.x;
-  uint4 val[kFlatUnroll];
-#pragma unroll
-  for (int j = 0; j < kFlatUnroll; ++j) {
-    const uint32_t v = base + j * blockDim.x;
-    if (v < total_vecs) val[j] = in[v];
-  }
-#pragma unroll
-  for (int j = 0; j < kFlatUnroll; ++j) {
-    const uint32_t v = base + j * blockDim.x;
-    if (v < total_vecs) {
-      const uint32_t tile = magic_div(v, mul_vpt, sh_vpt);
-      const uint32_t within = v - tile * vec_per_tile;
-      const uint32_t bi = magic_div(tile, mul_nl, sh_nl);
-      const uint32_t l = tile - bi * num_layers;
-      const uint64_t dst = lptr[l] +
-                           static_cast<uint64_t>(lids[bi]) * lstride[l] +
-                           static_cast<uint64_t>(within) * 16;
-      *reinterpret_cast<uint4*>(dst) = val[j];
-    }
-  }
-}
-
 // ---- batched prefix hashing -------------------------------------------------
 // One lane per sequence: walks the chain hash_{c+1} = FNV64a(CBOR([h_c,
 // chunk, null])). Sequences share a flat token buffer with offsets.
@@ -289,35 +160,6 @@ inline dim3 copy_grid(uint32_t tiles, uint64_t block_bytes) {
   return dim3(tiles, wg_per_tile);
 }
 
-// Small-tile flat path applies when the LDS table fits and magic division
-// stays exact.
-inline bool use_flat_path(uint32_t tiles, uint64_t block_bytes) {
-  const uint64_t total_vecs = tiles * (block_bytes / 16);
-  return block_bytes <= 32768 && tiles <= kFlatMaxTiles &&
-         total_vecs < kFlatMaxVecs && total_vecs > 0;
-}
-
-struct FlatLaunch {
-  dim3 grid;
-  uint32_t vec_per_tile, total_vecs;
-  MagicDiv div_vpt, div_nl;
-  size_t lds_bytes;
-};
-
-inline FlatLaunch flat_launch(uint32_t tiles, uint64_t block_bytes,
-                              int num_layers, int num_blocks) {
-  FlatLaunch f;
-  f.vec_per_tile = static_cast<uint32_t>(block_bytes / 16);
-  f.total_vecs = tiles * f.vec_per_tile;
-  const uint32_t per_wg = 256 * kFlatUnroll;
-  f.grid = dim3((f.total_vecs + per_wg - 1) / per_wg);
-  f.div_vpt = make_magic(f.vec_per_tile);
-  f.div_nl = make_magic(static_cast<uint32_t>(num_layers));
-  f.lds_bytes = 2 * num_layers * sizeof(uint64_t) +
-                ((num_blocks + 1) & ~1) * sizeof(int32_t);
-  return f;
-}
-
 extern "C" hipError_t kvc_launch_gather(
     const void* const* layer_ptrs_dev, const uint64_t* layer_strides_dev,
     int num_layers, uint64_t block_bytes, const int32_t* block_ids,
@@ -328,15 +170,6 @@ extern "C" hipError_t kvc_launch_gather(
     for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
   }
   const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
-  if (use_flat_path(tiles, block_bytes)) {
-    FlatLaunch f = flat_launch(tiles, block_bytes, num_layers, num_blocks);
-    hipLaunchKernelGGL(kvc_gather_blocks_flat, f.grid, dim3(256), f.lds_bytes,
-                       stream, layer_ptrs_dev, layer_strides_dev, num_layers,
-                       num_blocks, bl, ids_dev, f.vec_per_tile, f.total_vecs,
-                       f.div_vpt.mul, f.div_vpt.shift, f.div_nl.mul,
-                       f.div_nl.shift, dst);
-    return hipGetLastError();
-  }
   dim3 grid = copy_grid(tiles, block_bytes);
   hipLaunchKernelGGL(kvc_gather_blocks, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,
@@ -355,15 +188,6 @@ extern "C" hipError_t kvc_launch_scatter(
     for (int i = 0; i < num_blocks; ++i) bl.ids[i] = block_ids[i];
   }
   const uint32_t tiles = static_cast<uint32_t>(num_blocks) * num_layers;
-  if (use_flat_path(tiles, block_bytes)) {
-    FlatLaunch f = flat_launch(tiles, block_bytes, num_layers, num_blocks);
-    hipLaunchKernelGGL(kvc_scatter_blocks_flat, f.grid, dim3(256), f.lds_bytes,
-                       stream, layer_ptrs_dev, layer_strides_dev, num_layers,
-                       num_blocks, bl, ids_dev, f.vec_per_tile, f.total_vecs,
-                       f.div_vpt.mul, f.div_vpt.shift, f.div_nl.mul,
-                       f.div_nl.shift, src);
-    return hipGetLastError();
-  }
   dim3 grid = copy_grid(tiles, block_bytes);
   hipLaunchKernelGGL(kvc_scatter_blocks, grid, dim3(256), 0, stream,
                      layer_ptrs_dev, layer_strides_dev, num_layers, block_bytes,

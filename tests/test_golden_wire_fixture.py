@@ -16,7 +16,7 @@ the independent CPU oracle in tests/reference_impl.py.)
 import pytest
 
 from llm_d_kv_cache_amd import ensure_native
-from reference_impl import ref_block_keys
+from reference_impl import block_keys
 
 k = ensure_native()
 
@@ -82,4 +82,4 @@ def test_golden_hash_constants():
     assert tp2.tokens_to_block_keys(TOKENS[:16], "m", 0xDEADBEEF) == \
         [GOLDEN_CHAINED_KEY]
     # and the independent CPU oracle agrees with the frozen constants
-    assert ref_block_keys(TOKENS, MODEL, 16, "") == GOLDEN_KEYS
+    assert block_keys(TOKENS, MODEL, 16, "") == GOLDEN_KEYS

@@ -87,6 +87,11 @@ void bench_threads(int nthreads) {
 // orchestration gap or the platform's duplex ceiling (both directions are
 // blit kernels here, not SDMA — see profiles/r01_offload_profile.md).
 void bench_duplex(int streams_per_dir) {
+  // smaller buffers + fewer iters than the one-direction benches: if the
+  // two directions collapse when concurrent, a full-size run would take
+  // minutes — the collapse itself is the finding
+  constexpr size_t ND = 128ull << 20;
+  constexpr int ITD = 5;
   struct Dir {
     double gbps = 0;
   } res[2];
@@ -97,27 +102,27 @@ void bench_duplex(int streams_per_dir) {
       std::vector<void*> d(streams_per_dir), h(streams_per_dir);
       std::vector<hipStream_t> s(streams_per_dir);
       for (int i = 0; i < streams_per_dir; ++i) {
-        if (hipMalloc(&d[i], N) != hipSuccess) return;
-        if (hipHostMalloc(&h[i], N, hipHostMallocPortable) != hipSuccess)
+        if (hipMalloc(&d[i], ND) != hipSuccess) return;
+        if (hipHostMalloc(&h[i], ND, hipHostMallocPortable) != hipSuccess)
           return;
         hipStreamCreateWithFlags(&s[i], hipStreamNonBlocking);
       }
       auto kind = dir == 0 ? hipMemcpyDeviceToHost : hipMemcpyHostToDevice;
       // warmup
       for (int i = 0; i < streams_per_dir; ++i) {
-        hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], N,
+        hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], ND,
                        kind, s[i]);
         hipStreamSynchronize(s[i]);
       }
       double t1 = now();
-      for (int it = 0; it < ITERS; ++it) {
+      for (int it = 0; it < ITD; ++it) {
         for (int i = 0; i < streams_per_dir; ++i)
-          hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], N,
+          hipMemcpyAsync(dir == 0 ? h[i] : d[i], dir == 0 ? d[i] : h[i], ND,
                          kind, s[i]);
         for (int i = 0; i < streams_per_dir; ++i) hipStreamSynchronize(s[i]);
       }
       double dt = now() - t1;
-      res[dir].gbps = N * (double)ITERS * streams_per_dir / dt / 1e9;
+      res[dir].gbps = ND * (double)ITD * streams_per_dir / dt / 1e9;
       for (int i = 0; i < streams_per_dir; ++i) {
         hipStreamDestroy(s[i]);
         hipHostFree(h[i]);

@@ -74,7 +74,7 @@ def build_kvcore(force: bool = False) -> Path:
     if not force and not _needs_build(target, sources):
         return target
     cmd = [
-        "g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fvisibility=hidden",
+        "g++", "-O3", "-std=c++17", "-Werror=return-type", "-shared", "-fPIC", "-fvisibility=hidden",
         "-pthread", "-Wall",
     ]
     for inc in _pybind_includes():
@@ -97,7 +97,7 @@ def build_kvoffload(force: bool = False) -> Path:
     if not force and not _needs_build(target, all_sources):
         return target
     cmd = [
-        HIPCC, f"--offload-arch={GFX_ARCH}", "-O3", "-std=c++17", "-shared",
+        HIPCC, f"--offload-arch={GFX_ARCH}", "-O3", "-std=c++17", "-Werror=return-type", "-shared",
         "-fPIC", "-fvisibility=hidden", "-pthread", "-Wall",
         "-Wno-unused-result",
     ]

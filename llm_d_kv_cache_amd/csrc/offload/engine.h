@@ -288,14 +288,15 @@ class StorageOffloadEngine {
         bool ok = true;
         double t0 = now_s();
         bool wrote = false;
+        bool deferred = false;
         try {
           if (!job->cancelled.load(std::memory_order_acquire)) {
             if (file_exists(ft.path)) {
               touch_atime(ft.path);
               stats_inc([](EngineStats& s) { s.files_deduped++; });
             } else {
-              store_one(ctx, *job, ft);
-              wrote = true;
+              deferred = store_one(ctx, job, ft);
+              wrote = !deferred;
             }
           } else {
             stats_inc([](EngineStats& s) { s.tasks_cancelled++; });
@@ -305,6 +306,7 @@ class StorageOffloadEngine {
           stats_inc([](EngineStats& s) { s.errors++; });
           ok = false;
         }
+        if (deferred) return;  // write continuation owns EMA/pending/done
         if (wrote) update_write_ema(now_s() - t0);
         pending_writes_.fetch_sub(1, std::memory_order_relaxed);
         task_done(job, ok);
@@ -529,7 +531,10 @@ class StorageOffloadEngine {
     return ctx.dev_ids;
   }
 
-  void store_one(WorkerCtx& ctx, Job& job, const FileTransfer& ft) {
+  // Returns true when completion was DEFERRED to a write-continuation task
+  // (two-stage store): the caller must then skip EMA/pending/task_done.
+  bool store_one(WorkerCtx& ctx, const std::shared_ptr<Job>& job,
+                 const FileTransfer& ft) {
     const GroupDesc& g = groups_[ft.group];
     const int nb = static_cast<int>(ft.block_ids.size());
     const int nl = static_cast<int>(g.layer_ptrs.size());
@@ -579,13 +584,13 @@ class StorageOffloadEngine {
           s.bytes_stored += bytes;
           s.host_cache_stores++;
         });
-        return;
+        return false;
       }
     } else {
       // KV-ready fence: the gather must observe the serving engine's
       // completed KV writes for these blocks.
       double t0 = now_s();
-      KVO_HIP_CHECK(hipStreamWaitEvent(ctx.stream, job.kv_ready, 0));
+      KVO_HIP_CHECK(hipStreamWaitEvent(ctx.stream, job->kv_ready, 0));
       uint8_t* kernel_dst = cfg_.copy_path == CopyPath::kStaged
                                 ? ctx.device_staging->ptr()
                                 : ctx.host_staging->device();
@@ -666,7 +671,66 @@ class StorageOffloadEngine {
           s.bytes_stored += bytes;
           s.host_cache_stores++;
         });
-        return;
+        return false;
+      }
+      if (cfg_.copy_path == CopyPath::kStaged && slot != nullptr) {
+        // Two-stage write-through store: this worker completes once the
+        // D2H lands in the DRAM slot; the file write runs as its OWN task
+        // (completion still waits for the rename — durability unchanged).
+        // Per-worker serialization of [D2H -> write] left the D2H lane
+        // idle whenever every worker sat in write(): the duplex wire
+        // measures ~104 GB/s total (profiles/r02) while the bench ran
+        // ~54 — the lane must stay fed from the next task's D2H.
+        hipEvent_t gather_done;
+        KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
+        KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
+        double c0 = now_s();
+        try {
+          mover_->d2h(host_buf, ctx.device_staging->ptr(), bytes, gather_done);
+        } catch (...) {
+          (void)hipEventDestroy(gather_done);
+          throw;
+        }
+        (void)hipEventDestroy(gather_done);
+        double c1 = now_s();
+        guard.ok = true;  // publish the slot; store's reference released
+        cache_->addref(slot);
+        auto* cache = cache_.get();
+        std::string path = ft.path;
+        pool_->enqueue(Priority::kNormal,
+                       [this, job, cache, slot, path, bytes](WorkerCtx&) {
+          bool wok = true;
+          if (!job->cancelled.load(std::memory_order_acquire)) {
+            double w0 = now_s();
+            try {
+              AtomicFileWriter w(path, cfg_.direct_io);
+              w.write_at(0, slot->buf->host(), bytes);
+              w.commit();
+              double dt = now_s() - w0;
+              update_write_ema(dt);
+              stats_inc([&](EngineStats& s) {
+                s.files_written++;
+                s.t_write_ms += dt * 1e3;
+              });
+            } catch (const std::exception& e) {
+              KVO_LOG_ERROR("store write %s failed: %s", path.c_str(),
+                            e.what());
+              stats_inc([](EngineStats& s) { s.errors++; });
+              wok = false;
+            }
+          } else {
+            stats_inc([](EngineStats& s) { s.tasks_cancelled++; });
+          }
+          cache->release(slot);
+          pending_writes_.fetch_sub(1, std::memory_order_relaxed);
+          task_done(job, wok);
+        });
+        stats_inc([&](EngineStats& s) {
+          s.t_d2h_ms += (c1 - c0) * 1e3;
+          s.bytes_stored += bytes;
+          s.host_cache_stores++;
+        });
+        return true;
       }
       if (cfg_.copy_path == CopyPath::kStaged) {
         // Chunked pipeline: D2H of chunk i+1 rides the SDMA mover while
@@ -711,7 +775,7 @@ class StorageOffloadEngine {
           s.bytes_stored += bytes;
           if (slot) s.host_cache_stores++;
         });
-        return;
+        return false;
       }
       KVO_HIP_CHECK(hipStreamSynchronize(ctx.stream));
       double t2 = now_s();
@@ -727,6 +791,7 @@ class StorageOffloadEngine {
       s.t_write_ms += (tw2 - tw) * 1e3;
       if (slot) s.host_cache_stores++;
     });
+    return false;
   }
 
   void load_one(WorkerCtx& ctx, const FileTransfer& ft) {

@@ -521,6 +521,16 @@ class StorageOffloadEngine {
     return n_blocks * g.layer_ptrs.size() * tile_record_bytes(g);
   }
 
+  // A/B switch for the two-stage store (KVC_TWO_STAGE=0 reverts to the
+  // per-worker serial D2H->write flow) — measurement knob, default on.
+  static bool two_stage_stores() {
+    static const bool on = [] {
+      const char* e = std::getenv("KVC_TWO_STAGE");
+      return e == nullptr || e[0] != '0';
+    }();
+    return on;
+  }
+
   // Block ids above the kernarg limit ride the worker's device buffer.
   const int32_t* stage_ids(WorkerCtx& ctx, const FileTransfer& ft) {
     if (ft.block_ids.size() <= static_cast<size_t>(kMaxBlocksPerFileHost))
@@ -673,14 +683,17 @@ class StorageOffloadEngine {
         });
         return false;
       }
-      if (cfg_.copy_path == CopyPath::kStaged && slot != nullptr) {
+      if (cfg_.copy_path == CopyPath::kStaged && slot != nullptr &&
+          two_stage_stores()) {
         // Two-stage write-through store: this worker completes once the
         // D2H lands in the DRAM slot; the file write runs as its OWN task
         // (completion still waits for the rename — durability unchanged).
         // Per-worker serialization of [D2H -> write] left the D2H lane
         // idle whenever every worker sat in write(): the duplex wire
         // measures ~104 GB/s total (profiles/r02) while the bench ran
-        // ~54 — the lane must stay fed from the next task's D2H.
+        // ~54 — the lane must stay fed from the next task's D2H. The
+        // continuation is FRONT-inserted so writes interleave with the
+        // remaining stores' D2H instead of forming a tail.
         hipEvent_t gather_done;
         KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
         KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
@@ -697,8 +710,8 @@ class StorageOffloadEngine {
         cache_->addref(slot);
         auto* cache = cache_.get();
         std::string path = ft.path;
-        pool_->enqueue(Priority::kNormal,
-                       [this, job, cache, slot, path, bytes](WorkerCtx&) {
+        pool_->enqueue_front(Priority::kNormal,
+                             [this, job, cache, slot, path, bytes](WorkerCtx&) {
           bool wok = true;
           if (!job->cancelled.load(std::memory_order_acquire)) {
             double w0 = now_s();

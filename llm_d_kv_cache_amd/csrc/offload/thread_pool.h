@@ -88,6 +88,20 @@ class IoThreadPool {
     cv_.notify_one();
   }
 
+  // Front insertion: continuation tasks (the write half of a two-stage
+  // store) jump ahead of not-yet-started work in their class, so writes
+  // mix with the D2H stream instead of piling into a tail.
+  void enqueue_front(Priority prio, Task task) {
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      if (prio == Priority::kHigh)
+        high_.push_front(std::move(task));
+      else
+        normal_.push_front(std::move(task));
+    }
+    cv_.notify_one();
+  }
+
   size_t queued(Priority prio) {
     std::lock_guard<std::mutex> g(mu_);
     return prio == Priority::kHigh ? high_.size() : normal_.size();

@@ -79,6 +79,27 @@ int main() {
       ops++;
     }
   });
+  // byte-budget index with the TinyLFU admission sketch under contention
+  InMemoryIndexConfig bcfg;
+  bcfg.shards = 4;
+  bcfg.max_bytes = 64 * 1024;
+  auto budget_index = std::make_shared<InMemoryIndex>(bcfg);
+  for (int t = 0; t < 3; ++t) {
+    threads.emplace_back([&, t] {
+      PodEntry e;
+      e.pod = budget_index->strings().intern("bp-" + std::to_string(t));
+      e.tier = budget_index->strings().intern("gpu");
+      uint64_t i = 0;
+      while (!stop) {
+        uint64_t k = (t == 0) ? (i % 64) : (100000 + i);  // hot set + flood
+        budget_index->add({}, {k}, {e});
+        if (t == 0) budget_index->lookup({i % 64}, {});
+        ++i;
+        ops++;
+      }
+    });
+  }
+
   // event pool with live workers processing synthetic batches
   EventPool pool(tp, index, 4);
   pool.start();

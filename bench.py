@@ -734,6 +734,15 @@ def main():
                     "h2d": round(stats.t_h2d_ms, 1),
                     "scatter": round(stats.t_scatter_ms, 1),
                 },
+                # PCIe lane truth: busy wall + bytes per direction — the
+                # direct measure of wire feeding vs the ~104 GB/s duplex
+                # ceiling (profiles/r02_kernels.md §4)
+                "lanes": {
+                    "d2h_busy_ms": round(stats.d2h_lane_busy_ms, 1),
+                    "h2d_busy_ms": round(stats.h2d_lane_busy_ms, 1),
+                    "d2h_GB": round(stats.d2h_lane_bytes / 1e9, 2),
+                    "h2d_GB": round(stats.h2d_lane_bytes / 1e9, 2),
+                },
             },
         }))
     if dist is not None:

@@ -41,7 +41,11 @@ PYBIND11_MODULE(_kvoffload, m) {
       .def_readonly("t_write_ms", &EngineStats::t_write_ms)
       .def_readonly("t_read_ms", &EngineStats::t_read_ms)
       .def_readonly("t_h2d_ms", &EngineStats::t_h2d_ms)
-      .def_readonly("t_scatter_ms", &EngineStats::t_scatter_ms);
+      .def_readonly("t_scatter_ms", &EngineStats::t_scatter_ms)
+      .def_readonly("d2h_lane_busy_ms", &EngineStats::d2h_lane_busy_ms)
+      .def_readonly("h2d_lane_busy_ms", &EngineStats::h2d_lane_busy_ms)
+      .def_readonly("d2h_lane_bytes", &EngineStats::d2h_lane_bytes)
+      .def_readonly("h2d_lane_bytes", &EngineStats::h2d_lane_bytes);
 
   py::class_<StorageOffloadEngine>(m, "StorageOffloadEngine")
       .def(py::init([](std::vector<std::tuple<std::vector<uintptr_t>,

@@ -148,6 +148,9 @@ struct EngineStats {
   // accumulated per-phase wall time across all workers (ms)
   double t_gather_ms = 0, t_d2h_ms = 0, t_write_ms = 0;
   double t_read_ms = 0, t_h2d_ms = 0, t_scatter_ms = 0;
+  // PCIe lane utilization (mover-side truth: busy wall per direction)
+  double d2h_lane_busy_ms = 0, h2d_lane_busy_ms = 0;
+  uint64_t d2h_lane_bytes = 0, h2d_lane_bytes = 0;
 };
 
 class StorageOffloadEngine {
@@ -408,6 +411,13 @@ class StorageOffloadEngine {
       out.t_scatter_ms += s.t_scatter_ms;
     }
     out.avg_write_seconds = avg_write_seconds();
+    if (mover_) {
+      auto d = mover_->lane_stats(0), h = mover_->lane_stats(1);
+      out.d2h_lane_busy_ms = d.busy_s * 1e3;
+      out.h2d_lane_busy_ms = h.busy_s * 1e3;
+      out.d2h_lane_bytes = d.bytes;
+      out.h2d_lane_bytes = h.bytes;
+    }
     return out;
   }
 

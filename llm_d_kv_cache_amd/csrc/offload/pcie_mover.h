@@ -10,6 +10,8 @@
 // against producer kernels rides an event recorded on the worker stream.
 #pragma once
 
+#include <atomic>
+#include <chrono>
 #include <condition_variable>
 #include <deque>
 #include <future>
@@ -68,6 +70,22 @@ class PcieMover {
     h2d_async(dst, src, n).get();
   }
 
+  // Lane utilization (busy = wall spent inside waitEvent+memcpy+sync, i.e.
+  // everything but sitting on an empty queue): the direct measure of
+  // whether the wire is fed. Index 0 = D2H, 1 = H2D.
+  struct LaneStats {
+    double busy_s = 0;
+    uint64_t bytes = 0;
+    uint64_t requests = 0;
+  };
+  LaneStats lane_stats(int dir) const {
+    LaneStats s;
+    s.busy_s = lanes_[dir].busy_ns.load(std::memory_order_relaxed) * 1e-9;
+    s.bytes = lanes_[dir].bytes.load(std::memory_order_relaxed);
+    s.requests = lanes_[dir].requests.load(std::memory_order_relaxed);
+    return s;
+  }
+
  private:
   struct Req {
     void* dst;
@@ -82,6 +100,7 @@ class PcieMover {
     std::deque<std::unique_ptr<Req>> q;
     bool stopping = false;
     std::thread thread;
+    std::atomic<uint64_t> busy_ns{0}, bytes{0}, requests{0};
   };
 
   std::future<void> submit(Lane& lane, void* dst, const void* src, size_t n,
@@ -110,6 +129,7 @@ class PcieMover {
         req = std::move(lane.q.front());
         lane.q.pop_front();
       }
+      auto b0 = std::chrono::steady_clock::now();
       try {
         if (req->pre) KVO_HIP_CHECK(hipStreamWaitEvent(stream, req->pre, 0));
         KVO_HIP_CHECK(hipMemcpyAsync(req->dst, req->src, req->n, kind, stream));
@@ -118,6 +138,13 @@ class PcieMover {
       } catch (...) {
         req->done.set_exception(std::current_exception());
       }
+      lane.busy_ns.fetch_add(
+          std::chrono::duration_cast<std::chrono::nanoseconds>(
+              std::chrono::steady_clock::now() - b0)
+              .count(),
+          std::memory_order_relaxed);
+      lane.bytes.fetch_add(req->n, std::memory_order_relaxed);
+      lane.requests.fetch_add(1, std::memory_order_relaxed);
     }
   }
 

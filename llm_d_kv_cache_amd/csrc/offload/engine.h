@@ -585,7 +585,7 @@ class StorageOffloadEngine {
         cache_->addref(slot);
         auto* cache = cache_.get();
         std::string path = ft.path;
-        pool_->enqueue(Priority::kNormal,
+        pool_->enqueue(Priority::kCont,
                        [this, cache, slot, path, bytes](WorkerCtx&) {
                          try {
                            write_file_atomic(path, slot->buf->host(), bytes);
@@ -667,7 +667,7 @@ class StorageOffloadEngine {
         cache_->addref(slot);
         auto* cache = cache_.get();
         std::string path = ft.path;
-        pool_->enqueue(Priority::kNormal,
+        pool_->enqueue(Priority::kCont,
                        [this, cache, slot, path, bytes](WorkerCtx&) {
                          double w0 = now_s();
                          try {
@@ -702,8 +702,9 @@ class StorageOffloadEngine {
         // idle whenever every worker sat in write(): the duplex wire
         // measures ~104 GB/s total (profiles/r02) while the bench ran
         // ~54 — the lane must stay fed from the next task's D2H. The
-        // continuation is FRONT-inserted so writes interleave with the
-        // remaining stores' D2H instead of forming a tail.
+        // continuation runs in the kCont class: read-preferring workers
+        // pick writes up after their loads drain while write-preferring
+        // workers keep pumping store D2Hs (thread_pool.h preferences).
         hipEvent_t gather_done;
         KVO_HIP_CHECK(hipEventCreateWithFlags(&gather_done, hipEventDisableTiming));
         KVO_HIP_CHECK(hipEventRecord(gather_done, ctx.stream));
@@ -720,8 +721,8 @@ class StorageOffloadEngine {
         cache_->addref(slot);
         auto* cache = cache_.get();
         std::string path = ft.path;
-        pool_->enqueue_front(Priority::kNormal,
-                             [this, job, cache, slot, path, bytes](WorkerCtx&) {
+        pool_->enqueue(Priority::kCont,
+                       [this, job, cache, slot, path, bytes](WorkerCtx&) {
           bool wok = true;
           if (!job->cancelled.load(std::memory_order_acquire)) {
             double w0 = now_s();

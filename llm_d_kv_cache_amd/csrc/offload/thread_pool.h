@@ -33,7 +33,14 @@ struct WorkerCtx {
   int32_t* dev_ids = nullptr;  // device block-id buffer (> kernarg-limit paths)
 };
 
-enum class Priority { kHigh = 0, kNormal = 1 };
+// kHigh = loads, kNormal = store front halves (gather + D2H), kCont =
+// continuation work (the file-write half of a two-stage store, write-back
+// flushes). Worker preference is asymmetric so the PCIe lanes stay fed:
+//   read-preferring:  high -> cont -> normal   (loads, then writes)
+//   write-preferring: normal -> cont -> high   (keep pumping D2Hs)
+// A writer picking cont before normal would drain the D2H queue dry
+// (measured: both lanes <50% busy at full wire speed per byte).
+enum class Priority { kHigh = 0, kNormal = 1, kCont = 2 };
 
 class IoThreadPool {
  public:
@@ -80,31 +87,14 @@ class IoThreadPool {
   void enqueue(Priority prio, Task task) {
     {
       std::lock_guard<std::mutex> g(mu_);
-      if (prio == Priority::kHigh)
-        high_.push_back(std::move(task));
-      else
-        normal_.push_back(std::move(task));
-    }
-    cv_.notify_one();
-  }
-
-  // Front insertion: continuation tasks (the write half of a two-stage
-  // store) jump ahead of not-yet-started work in their class, so writes
-  // mix with the D2H stream instead of piling into a tail.
-  void enqueue_front(Priority prio, Task task) {
-    {
-      std::lock_guard<std::mutex> g(mu_);
-      if (prio == Priority::kHigh)
-        high_.push_front(std::move(task));
-      else
-        normal_.push_front(std::move(task));
+      queue_for(prio).push_back(std::move(task));
     }
     cv_.notify_one();
   }
 
   size_t queued(Priority prio) {
     std::lock_guard<std::mutex> g(mu_);
-    return prio == Priority::kHigh ? high_.size() : normal_.size();
+    return queue_for(prio).size();
   }
 
   void shutdown() {
@@ -160,19 +150,33 @@ class IoThreadPool {
     }
     started_cv_.notify_all();
 
+    std::deque<Task>* order[3];
+    if (read_preferring) {
+      order[0] = &high_;
+      order[1] = &cont_;
+      order[2] = &normal_;
+    } else {
+      order[0] = &normal_;
+      order[1] = &cont_;
+      order[2] = &high_;
+    }
     for (;;) {
       Task task;
       {
         std::unique_lock<std::mutex> g(mu_);
         cv_.wait(g, [this] {
-          return stopping_ || !high_.empty() || !normal_.empty();
+          return stopping_ || !high_.empty() || !normal_.empty() ||
+                 !cont_.empty();
         });
-        if (high_.empty() && normal_.empty()) break;  // stopping and drained
-        auto& primary = read_preferring ? high_ : normal_;
-        auto& secondary = read_preferring ? normal_ : high_;
-        auto& q = !primary.empty() ? primary : secondary;
-        task = std::move(q.front());
-        q.pop_front();
+        std::deque<Task>* q = nullptr;
+        for (auto* cand : order)
+          if (!cand->empty()) {
+            q = cand;
+            break;
+          }
+        if (q == nullptr) break;  // stopping and drained
+        task = std::move(q->front());
+        q->pop_front();
       }
       try {
         task(ctx);
@@ -194,7 +198,12 @@ class IoThreadPool {
   std::mutex mu_;
   std::condition_variable cv_;
   std::condition_variable started_cv_;
-  std::deque<Task> high_, normal_;
+  std::deque<Task>& queue_for(Priority p) {
+    return p == Priority::kHigh ? high_
+                                : (p == Priority::kNormal ? normal_ : cont_);
+  }
+
+  std::deque<Task> high_, normal_, cont_;
   std::vector<std::thread> workers_;
   std::vector<char> ready_;
   std::string init_error_;

@@ -389,6 +389,8 @@ def main():
     ap.add_argument("--serialize", type=str, default="raw",
                     choices=["raw", "fp8_e4m3"])
     ap.add_argument("--io-threads", type=int, default=16)
+    ap.add_argument("--read-ratio", type=float, default=0.75,
+                    help="read-preferring worker fraction")
     ap.add_argument("--device-blocks", type=int, default=2048)
     ap.add_argument("--model", type=str, default="llama-3-8b",
                     choices=sorted(MODEL_PRESETS))
@@ -488,6 +490,7 @@ def main():
                             serialize=args.serialize,
                             host_cache_bytes=int(args.host_cache_gb * 1024**3),
                             write_policy=args.write_policy,
+                            read_preferring_ratio=args.read_ratio,
                             device=local_rank),
     )
     mapper = FileMapper(rank_root, KVCacheLayoutConfig(

@@ -531,12 +531,16 @@ class StorageOffloadEngine {
     return n_blocks * g.layer_ptrs.size() * tile_record_bytes(g);
   }
 
-  // A/B switch for the two-stage store (KVC_TWO_STAGE=0 reverts to the
-  // per-worker serial D2H->write flow) — measurement knob, default on.
+  // Two-stage store A/B switch (KVC_TWO_STAGE=1 opts in). Measured SLOWER
+  // than the serial per-worker D2H->write flow on every box tried (same-
+  // box: 52.0 vs 55.4 GB/s at 32 workers, 42.9 vs 49.1 at 16): splitting
+  // the write off starves the D2H queue worse than the write blocking the
+  // worker does, because submission — not lane service — is the scarce
+  // resource. Kept for future re-measurement; default off.
   static bool two_stage_stores() {
     static const bool on = [] {
       const char* e = std::getenv("KVC_TWO_STAGE");
-      return e == nullptr || e[0] != '0';
+      return e != nullptr && e[0] == '1';
     }();
     return on;
   }

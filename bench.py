@@ -388,7 +388,7 @@ def main():
                     choices=["staged", "zero_copy"])
     ap.add_argument("--serialize", type=str, default="raw",
                     choices=["raw", "fp8_e4m3"])
-    ap.add_argument("--io-threads", type=int, default=16)
+    ap.add_argument("--io-threads", type=int, default=32)
     ap.add_argument("--read-ratio", type=float, default=0.75,
                     help="read-preferring worker fraction")
     ap.add_argument("--device-blocks", type=int, default=2048)

@@ -389,8 +389,11 @@ def main():
     ap.add_argument("--serialize", type=str, default="raw",
                     choices=["raw", "fp8_e4m3"])
     ap.add_argument("--io-threads", type=int, default=32)
-    ap.add_argument("--read-ratio", type=float, default=0.75,
-                    help="read-preferring worker fraction")
+    ap.add_argument("--read-ratio", type=float, default=0.25,
+                    help="read-preferring worker fraction (0.25 measured "
+                         "best for duplex throughput: most workers pump "
+                         "store D2Hs; the ENGINE default stays 0.75 for "
+                         "read-latency QoS — profiles/r02_offload.md)")
     ap.add_argument("--device-blocks", type=int, default=2048)
     ap.add_argument("--model", type=str, default="llama-3-8b",
                     choices=sorted(MODEL_PRESETS))

@@ -53,7 +53,7 @@ BLOCKS_PER_FILE = 16  # 256-token offload chunks
 
 
 def apply_model_preset(name):
-    global MODEL, NUM_LAYERS, KV_HEADS, BLOCK_BYTES
+    global MODEL, NUM_LAYERS, KV_HEADS, BLOCK_BYTES, FILES_PER_STEP
     p = MODEL_PRESETS[name]
     MODEL = p["model"]
     NUM_LAYERS = p["layers"] if not os.environ.get("KVC_BENCH_TINY") else 4
@@ -63,6 +63,13 @@ def apply_model_preset(name):
     else:
         KV_HEADS = p["kv_heads"]
         BLOCK_BYTES = 2 * BLOCK_TOKENS * KV_HEADS * HEAD_SIZE * 2
+    if not os.environ.get("KVC_BENCH_TINY"):
+        # keep ~2 GB of KV per step regardless of shard geometry so small-
+        # file presets (70B-TP8: 10 MB files) don't just measure per-step
+        # fixed overhead; the step is still whole files of the preset's
+        # real size
+        file_bytes = BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES
+        FILES_PER_STEP = max(16, min(256, (2 << 30) // file_bytes))
 
 
 def log(msg):

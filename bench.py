@@ -416,6 +416,12 @@ def main():
                          "disjoint GPU pages)")
     ap.add_argument("--no-overlap", dest="overlap", action="store_false",
                     help="sequential store-then-load phases")
+    ap.add_argument("--pipeline", type=int, default=2,
+                    help="steps kept in flight (overlap mode): 1 = "
+                         "step-synchronous; 2 (default) = true steady "
+                         "state, the next step's stores start while the "
+                         "previous step's tail writes drain. All work "
+                         "completes inside the timed region.")
     args = ap.parse_args()
     apply_model_preset(args.model)
 
@@ -477,10 +483,16 @@ def main():
     shutil.rmtree(rank_root, ignore_errors=True)
     os.makedirs(rank_root, exist_ok=True)
 
-    need = FILES_PER_STEP * BLOCKS_PER_FILE * (2 if args.overlap else 1)
+    page_sets = 1
+    if args.overlap:
+        # store pages + one load window, or two alternating load windows
+        # when steps pipeline (concurrent generations never scatter into
+        # the same pages)
+        page_sets = 2 if args.pipeline <= 1 else 3
+    need = FILES_PER_STEP * BLOCKS_PER_FILE * page_sets
     if args.device_blocks < need:
         log(f"raising --device-blocks {args.device_blocks} -> {need} "
-            f"(overlap mode uses disjoint store/load pages)")
+            f"(disjoint store/load page windows)")
         args.device_blocks = need
 
     # Llama-3-8B canonical KV: one group, 32 layers, (num_blocks, 128 KiB)
@@ -537,37 +549,64 @@ def main():
     for t in deleters:
         t.start()
 
+    pending = {"n": 0}
+    store_jobs = {}        # generation -> store job id
+    done_stores = set()    # completed store job ids
+
+    def poll_done():
+        fin_s = store.get_finished()
+        done_stores.update(r.job_id for r in fin_s)
+        pending["n"] -= len(fin_s) + len(load.get_finished())
+
+    def drain_to(limit):
+        poll_done()
+        while pending["n"] > limit:
+            time.sleep(0.0002)
+            poll_done()
+
+    def await_store(gen):
+        # a generation's load must not race its own store (pipelined steps
+        # drop the old per-step barrier that used to guarantee this)
+        jid = store_jobs.get(gen)
+        while jid is not None and jid not in done_stores:
+            time.sleep(0.0002)
+            poll_done()
+
     def run_step_overlap(step_id):
-        # store gen step_id from pages [0, B); load gen step_id-1 into
-        # pages [B, 2B) — disjoint, so both directions run concurrently
+        # store gen step_id from pages [0, B); load gen step_id-1 into a
+        # load window disjoint from the stores (and, when steps pipeline,
+        # alternating by parity so concurrent generations never scatter
+        # into the same pages)
         base = step_id * FILES_PER_STEP + 1
         hashes = list(range(base, base + FILES_PER_STEP))
         prev_base = (step_id - 1) * FILES_PER_STEP + 1
         prev_hashes = list(range(prev_base, prev_base + FILES_PER_STEP))
         do_load = step_id - 1 >= -args.warmup
-        n_store = n_load = 0
         p0 = time.perf_counter()
         # one submission per direction: tasks are per-FILE inside the
         # engine regardless of job granularity, and 2 python calls beat 16
         # (~100 us each) on the submission path
         all_ids = list(range(blocks_per_step))
-        store.transfer_async(hashes, {0: all_ids})
-        n_store += 1
+        store_jobs[step_id] = store.transfer_async(hashes, {0: all_ids})
+        pending["n"] += 1
         if do_load:
-            lids = [b + blocks_per_step for b in all_ids]
+            await_store(step_id - 1)
+            off = blocks_per_step
+            if args.pipeline > 1:
+                off = blocks_per_step * (1 + (step_id % 2))
+            lids = [b + off for b in all_ids]
             load.transfer_async(prev_hashes, {0: lids})
-            n_load += 1
-        ds = dl = 0
-        while ds < n_store or dl < n_load:
-            ds += len(store.get_finished())
-            dl += len(load.get_finished())
-            if ds < n_store or dl < n_load:
-                time.sleep(0.0002)
+            pending["n"] += 1
+        # steady-state window: at most `pipeline` steps of jobs in flight
+        drain_to(2 * (args.pipeline - 1))
         p1 = time.perf_counter()
         if step_id >= 0:
             moved["bytes"] += step_bytes + (step_bytes if do_load else 0)
-        old = (step_id - 2) * FILES_PER_STEP + 1
-        if step_id - 2 >= -args.warmup:
+        # deletion lags the pipeline window so a generation's files are
+        # never unlinked while its load can still be in flight
+        old_gen = step_id - 2 - max(0, args.pipeline - 1) - 1
+        old = old_gen * FILES_PER_STEP + 1
+        if old_gen >= -args.warmup:
             for h in range(old, old + FILES_PER_STEP):
                 del_q.put(mapper.file_name(h, 0))
         phase_wall["store"] += p1 - p0
@@ -618,6 +657,8 @@ def main():
 
     for sid in range(-args.warmup, 0):  # ascending: overlap loads gen-1
         run_step(sid)
+    if args.overlap:
+        drain_to(0)  # flush before warmup-generation cleanup
     # drop warmup generations BEFORE the timed region starts (pre-existing
     # state, not steady-state work; one generation per step remains inside).
     # Overlap mode keeps the newest warmup generation: timed step 0 loads it.
@@ -636,6 +677,8 @@ def main():
     t0 = time.perf_counter()
     for s in range(args.steps):
         run_step(s)
+    if args.overlap:
+        drain_to(0)  # the pipeline tail is part of the timed work
     # the deletion backlog is part of the steady-state work: drain it
     # inside the timed region
     while not del_q.empty():
@@ -731,6 +774,7 @@ def main():
                 "host_cache_gb": args.host_cache_gb,
                 "write_policy": args.write_policy,
                 "overlap": args.overlap,
+                "pipeline": args.pipeline,
                 "writeback_flushes": stats.writeback_flushes,
                 "host_cache_hits": stats.host_cache_hits,
                 "host_cache_stores": stats.host_cache_stores,

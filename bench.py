@@ -579,9 +579,14 @@ def main():
         # into the same pages)
         base = step_id * FILES_PER_STEP + 1
         hashes = list(range(base, base + FILES_PER_STEP))
-        prev_base = (step_id - 1) * FILES_PER_STEP + 1
+        # pipelined steps load TWO generations back: gen N-1's store still
+        # overlaps into step N, and gating the load on it would serialize
+        # the two directions (measured 3x worse); gen N-2's store is
+        # already done, so await_store is a no-op stall in steady state
+        load_gen = step_id - (1 if args.pipeline <= 1 else 2)
+        prev_base = load_gen * FILES_PER_STEP + 1
         prev_hashes = list(range(prev_base, prev_base + FILES_PER_STEP))
-        do_load = step_id - 1 >= -args.warmup
+        do_load = load_gen >= -args.warmup
         p0 = time.perf_counter()
         # one submission per direction: tasks are per-FILE inside the
         # engine regardless of job granularity, and 2 python calls beat 16
@@ -590,7 +595,7 @@ def main():
         store_jobs[step_id] = store.transfer_async(hashes, {0: all_ids})
         pending["n"] += 1
         if do_load:
-            await_store(step_id - 1)
+            await_store(load_gen)
             off = blocks_per_step
             if args.pipeline > 1:
                 off = blocks_per_step * (1 + (step_id % 2))
@@ -604,7 +609,7 @@ def main():
             moved["bytes"] += step_bytes + (step_bytes if do_load else 0)
         # deletion lags the pipeline window so a generation's files are
         # never unlinked while its load can still be in flight
-        old_gen = step_id - 2 - max(0, args.pipeline - 1) - 1
+        old_gen = step_id - 3 - max(0, args.pipeline - 1) - 1
         old = old_gen * FILES_PER_STEP + 1
         if old_gen >= -args.warmup:
             for h in range(old, old + FILES_PER_STEP):
@@ -662,7 +667,10 @@ def main():
     # drop warmup generations BEFORE the timed region starts (pre-existing
     # state, not steady-state work; one generation per step remains inside).
     # Overlap mode keeps the newest warmup generation: timed step 0 loads it.
-    keep = {-1} if args.overlap else set()
+    keep = set()
+    if args.overlap:
+        # the first timed steps still load these warmup generations
+        keep = {-1} if args.pipeline <= 1 else {-1, -2}
     for gen in range(-args.warmup, 0):
         if gen in keep:
             continue

@@ -418,7 +418,7 @@ def main():
                          "disjoint GPU pages)")
     ap.add_argument("--no-overlap", dest="overlap", action="store_false",
                     help="sequential store-then-load phases")
-    ap.add_argument("--pipeline", type=int, default=2,
+    ap.add_argument("--pipeline", type=int, default=1,
                     help="steps kept in flight (overlap mode): 1 = "
                          "step-synchronous; 2 (default) = true steady "
                          "state, the next step's stores start while the "

@@ -44,23 +44,27 @@ def _run_rank(rank, world, init_file, fn_name, q):
         q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
 
 
-def spawn2(fn_name, tmp_path):
+def spawn_world(fn_name, tmp_path, world):
     init_file = str(tmp_path / "pg_init")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_run_rank, args=(r, 2, init_file, fn_name, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_run_rank, args=(r, world, init_file, fn_name, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(2):
-        rank, status = q.get(timeout=120)
+    for _ in range(world):
+        rank, status = q.get(timeout=240)
         results[rank] = status
     for p in procs:
         p.join(timeout=30)
         if p.is_alive():
             p.terminate()
     assert all(v == "ok" for v in results.values()), results
+
+
+def spawn2(fn_name, tmp_path):
+    spawn_world(fn_name, tmp_path, 2)
 
 
 # ---- scenarios (run inside worker processes) --------------------------------
@@ -402,6 +406,35 @@ def scenario_index_driven_peer_resolution(rank, svc, group):
     dist.barrier()
 
 
+def scenario_mesh_pull_world4(rank, svc, group):
+    """Full-mesh pulls at world 4 (the 8-GPU node's topology in miniature):
+    every rank registers its own chunk and pulls every other rank's,
+    concurrently — no deadlock, every payload bit-exact."""
+    import torch.distributed as dist
+
+    world = dist.get_world_size()
+    CH = 0x4000
+    svc.register_blocks(CH + rank, 0, [0, 1, 2, 3])
+    # publish every rank's golden rows
+    goldens = []
+    for r in range(world):
+        g = (group[0][[0, 1, 2, 3]].clone() if r == rank
+             else torch.zeros(4, 4096, dtype=torch.uint8))
+        dist.broadcast(g, src=r)
+        goldens.append(g)
+    dist.barrier()
+    futs = {}
+    for i, src in enumerate([r for r in range(world) if r != rank]):
+        dst = [8 + 4 * i + j for j in range(4)]
+        futs[src] = (dst, svc.pull(CH + src, 0, dst, src_rank=src,
+                                   timeout=120))
+    for src, (dst, f) in futs.items():
+        assert f.result(timeout=150) is True, f"pull from {src} failed"
+        assert torch.equal(group[0][dst], goldens[src])
+    assert svc.stats().pulls_served >= 1
+    dist.barrier()
+
+
 # ---- tests ------------------------------------------------------------------
 
 @pytest.mark.parametrize("scenario", [
@@ -420,6 +453,11 @@ def scenario_index_driven_peer_resolution(rank, svc, group):
 ])
 def test_peer_migration(scenario, tmp_path):
     spawn2(scenario, tmp_path)
+
+
+@pytest.mark.timeout(300)
+def test_peer_mesh_world4(tmp_path):
+    spawn_world("scenario_mesh_pull_world4", tmp_path, 4)
 
 
 def test_block_copier_roundtrip():

@@ -431,7 +431,7 @@ def main():
         # pipeline window (store gen N while loading N-2, deletions lag
         # further) plus slack — one generation short and every load falls
         # back to file reads (measured: hits 1536 -> 106, 55 -> 24 GB/s)
-        gens = 3 + (2 * max(1, args.pipeline) - 1)
+        gens = 3 + 2 * (max(1, args.pipeline) - 1)
         gen_gb = FILES_PER_STEP * BLOCKS_PER_FILE * NUM_LAYERS * BLOCK_BYTES / 1e9
         args.host_cache_gb = round(gens * gen_gb, 1)
 

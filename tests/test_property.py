@@ -121,3 +121,77 @@ def test_snapshot_loader_rejects_garbage(tmp_path_factory, blob, flip):
         pass
     idx.add([], [9], [k.PodEntry("p", "gpu")])
     assert 9 in idx.lookup([9])
+
+
+# ---- handler file-splitting math (property) --------------------------------
+
+class _RecordingEngine:
+    """Stub engine capturing the exact FileTransfer lists handlers build."""
+
+    def __init__(self, num_layers=2, block_bytes=1024):
+        self.group_geometry = [{"num_layers": num_layers,
+                                "block_bytes": block_bytes}]
+        self.calls = []
+        self._next = 1
+
+    def async_store(self, files, stream=None):
+        self.calls.append(("store", files))
+        self._next += 1
+        return self._next - 1
+
+    def async_load(self, files):
+        self.calls.append(("load", files))
+        self._next += 1
+        return self._next - 1
+
+    def poll_finished(self, ids):
+        return []
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    bpf=st.integers(min_value=1, max_value=16),
+    n_chunks=st.integers(min_value=1, max_value=8),
+    skip=st.integers(min_value=0, max_value=127),
+    tail_drop=st.integers(min_value=0, max_value=15),
+)
+def test_load_split_covers_exactly_the_requested_blocks(
+        bpf, n_chunks, skip, tail_drop):
+    """For any (blocks/file, chunk count, leading skip, short tail): the
+    generated per-file transfers cover exactly the requested block ids in
+    order, the first file is tail-seeked by skip%bpf, later files start at
+    slot 0, and no file exceeds its remaining capacity (reference
+    worker.py head/tail-partial math)."""
+    from llm_d_kv_cache_amd.offload.file_mapper import (
+        FileMapper,
+        KVCacheLayoutConfig,
+    )
+    from llm_d_kv_cache_amd.offload.handlers import StorageToGPUHandler
+
+    total = n_chunks * bpf
+    skip = min(skip, total)
+    n_want = max(0, total - skip - min(tail_drop, total - skip))
+    if n_want == 0:
+        return
+    ids = list(range(1000, 1000 + n_want))
+    eng = _RecordingEngine()
+    import tempfile
+
+    mapper = FileMapper(tempfile.mkdtemp(), KVCacheLayoutConfig(model="prop"))
+    h = StorageToGPUHandler(eng, mapper, [bpf])
+    hashes = list(range(1, n_chunks + 1))
+    h.transfer_async(hashes, {0: ids}, skip_leading_blocks=skip)
+    (kind, files), = eng.calls
+    assert kind == "load"
+    got = []
+    for fi, (group, path, f_ids, slot) in enumerate(files):
+        assert group == 0
+        ci = skip // bpf + fi
+        assert mapper.file_name(hashes[ci], 0) == path
+        expect_slot = skip % bpf if fi == 0 else 0
+        assert slot == expect_slot
+        assert 1 <= len(f_ids) <= bpf - expect_slot
+        if fi not in (0, len(files) - 1):
+            assert len(f_ids) == bpf  # middle files are always full spans
+        got.extend(f_ids)
+    assert got == ids

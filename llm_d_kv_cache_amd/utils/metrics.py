@@ -50,6 +50,10 @@ class KVCacheMetricsCollector:
                 "index lookups with at least one hit", value=s.hits)
             yield GaugeMetricFamily(
                 "kvcache_index_keys", "request keys resident", value=s.keys)
+            yield CounterMetricFamily(
+                "kvcache_index_admission_rejects_total",
+                "adds rejected by the byte-budget admission sketch",
+                value=s.rejections)
         if self.events_pool is not None:
             p = self.events_pool.stats()
             yield CounterMetricFamily(
@@ -65,6 +69,10 @@ class KVCacheMetricsCollector:
                 "kvcache_events_parent_misses_total",
                 "BlockStored dropped on unknown parent chain",
                 value=p.dropped_parent_misses)
+            yield CounterMetricFamily(
+                "kvcache_events_handler_failures_total",
+                "events dropped on backend errors (outages, bad fields)",
+                value=p.handler_failures)
         if self.offload_engine is not None:
             e = self.offload_engine.stats()
             yield CounterMetricFamily(
@@ -85,6 +93,18 @@ class KVCacheMetricsCollector:
             yield CounterMetricFamily(
                 "kv_offload_bytes_loaded_total", "bytes loaded back",
                 value=e.bytes_loaded)
+            yield CounterMetricFamily(
+                "kv_offload_d2h_lane_busy_seconds_total",
+                "PCIe D2H lane busy wall", value=e.d2h_lane_busy_ms / 1e3)
+            yield CounterMetricFamily(
+                "kv_offload_h2d_lane_busy_seconds_total",
+                "PCIe H2D lane busy wall", value=e.h2d_lane_busy_ms / 1e3)
+            yield CounterMetricFamily(
+                "kv_offload_d2h_lane_bytes_total",
+                "bytes through the D2H lane", value=e.d2h_lane_bytes)
+            yield CounterMetricFamily(
+                "kv_offload_h2d_lane_bytes_total",
+                "bytes through the H2D lane", value=e.h2d_lane_bytes)
         if self.peer_service is not None:
             ps = self.peer_service.stats()
             yield CounterMetricFamily(

@@ -261,6 +261,22 @@ def bench_control_plane():
     }
 
 
+def _trace(rank, msg):
+    if os.environ.get("KVC_BENCH_TRACE"):
+        print(f"[trace r{rank}] {msg}", file=sys.stderr, flush=True)
+
+
+def _destroy_peer_groups(dist, *groups):
+    # gloo subgroups left to interpreter-exit GC abort the process
+    # sporadically ("terminate called without an active exception",
+    # ~1/6 world-8 runs): destroy them in order while everything is alive
+    for g in groups:
+        try:
+            dist.destroy_process_group(g)
+        except Exception:
+            pass
+
+
 def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     """Measure cross-GPU block-pull bandwidth over RCCL/xGMI (or gloo on
     CPU). Each rank pulls 16-block chunks from its ring neighbor.
@@ -282,6 +298,11 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
         dist.all_reduce(t, op=dist.ReduceOp.MIN)
         return bool(t.item())
 
+    def bail(result):
+        _destroy_peer_groups(dist, ctrl_pg, data_pg)
+        return result
+
+    _trace(rank, "peer: groups created")
     svc = None
     err = None
     try:
@@ -290,10 +311,11 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
                                    device=local_rank if gpu else 0)
     except Exception as e:
         err = f"service init: {e}"
+    _trace(rank, "peer: service up")
     if not agree(svc is not None):
         if svc is not None:
             svc.close()
-        return {"ok": False, "error": err or "peer rank failed init"}
+        return bail({"ok": False, "error": err or "peer rank failed init"})
 
     bpf = BLOCKS_PER_FILE
     nb = int(group[0].shape[0])
@@ -312,7 +334,8 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
         ok_local, err = False, f"register: {e}"
     if not agree(ok_local):
         svc.close()
-        return {"ok": False, "error": err or "peer rank failed register"}
+        return bail({"ok": False, "error": err or "peer rank failed register"})
+    _trace(rank, "peer: registered")
 
     try:
         svc.pull(0xE000 + src * 1000, 0,
@@ -322,7 +345,8 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
         ok_local, err = False, f"warmup pull: {e}"
     if not agree(ok_local):
         svc.close()
-        return {"ok": False, "error": err or "peer rank failed warmup"}
+        return bail({"ok": False, "error": err or "peer rank failed warmup"})
+    _trace(rank, "peer: warmup pull done")
 
     if gpu:
         torch.cuda.synchronize()
@@ -341,12 +365,13 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     ok = agree(ok_local)
+    _trace(rank, "peer: timed pulls done")
     pulled = svc.stats().bytes_received
     t = torch.tensor([dt], device=dev)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     if not ok:
         svc.close()
-        return {"ok": False, "error": err or "peer pull failed on a rank"}
+        return bail({"ok": False, "error": err or "peer pull failed on a rank"})
 
     # second round: fp8 on the wire (peer quantizes during the gather) —
     # same collective structure, half the data-plane bytes per pull.
@@ -371,7 +396,11 @@ def bench_peer_phase(dist, torch, group, rank, world, local_rank):
     wire8 = svc.stats().bytes_received - pulled
     t8 = torch.tensor([dt8], device=dev)
     dist.all_reduce(t8, op=dist.ReduceOp.MAX)
+    _trace(rank, "peer: fp8 round done")
     svc.close()
+    _trace(rank, "peer: closed")
+    _destroy_peer_groups(dist, ctrl_pg, data_pg)
+    _trace(rank, "peer: groups destroyed")
     out = {
         "ok": True,
         "pull_GBps_per_gpu": round(pulled / dt / 1e9, 2),
@@ -721,6 +750,7 @@ def main():
 
     # xGMI peer-migration phase (aux, outside the headline timed region):
     # each rank pulls its neighbor's cached chunks over RCCL send/recv.
+    _trace(rank, "main: timed steps done")
     peer_aux = None
     if world > 1 and dist is not None:
         try:
@@ -730,7 +760,9 @@ def main():
             log(f"peer phase failed: {e}")
             peer_aux = {"ok": False, "error": str(e)}
 
+    _trace(rank, "main: peer phase done")
     aux = bench_control_plane() if rank == 0 else None
+    _trace(rank, "main: control plane done")
     if aux is not None and peer_aux is not None:
         aux["peer_xgmi"] = peer_aux
 
@@ -824,6 +856,7 @@ def main():
         }))
     if dist is not None:
         dist.destroy_process_group()
+    _trace(rank, "main: exit")
 
 
 if __name__ == "__main__":

@@ -213,6 +213,13 @@ class PeerMigrationService:
         with self._q_mu:
             self._cmd_q.append(("bye",))
         self._thread.join(timeout=30.0)
+        # The LISTENER must be joined too: it exits only after draining
+        # every peer's BYE, and destroying (or GC-ing at interpreter exit)
+        # the control group while it still sits in dist.recv aborts the
+        # whole process inside gloo ("terminate called without an active
+        # exception" — reproduced 10/12 at world 8 with an explicit
+        # destroy_process_group racing the listener).
+        self._listener.join(timeout=30.0)
 
     # ---- control listener (blocking gloo recv, any source) ------------------
 

@@ -448,11 +448,12 @@ def main():
     ap.add_argument("--no-overlap", dest="overlap", action="store_false",
                     help="sequential store-then-load phases")
     ap.add_argument("--pipeline", type=int, default=1,
-                    help="steps kept in flight (overlap mode): 1 = "
-                         "step-synchronous; 2 (default) = true steady "
-                         "state, the next step's stores start while the "
-                         "previous step's tail writes drain. All work "
-                         "completes inside the timed region.")
+                    help="steps kept in flight (overlap mode). 1 (default) "
+                         "= step-synchronous; 2 = the next step's stores "
+                         "start while the previous step's tail writes "
+                         "drain (within noise of 1 when the DRAM tier "
+                         "covers the window — profiles/r02_offload.md). "
+                         "All work completes inside the timed region.")
     args = ap.parse_args()
     apply_model_preset(args.model)
     if args.host_cache_gb is None:

@@ -130,6 +130,7 @@ class PeerMigrationService:
         self._ctrl_q: deque = deque()
         self._q_mu = threading.Lock()
         self._stats = PeerStats()
+        self._abandoned: List[tuple] = []  # timed-out (tracker, buf) pairs
         self._stopping = False
         self._byes_seen = 0
         self._next_req_id = self.rank + 1
@@ -336,6 +337,12 @@ class PeerMigrationService:
                         st["fut"].set_exception(
                             TimeoutError(f"pull {req_id} from rank {st['src']}"))
                         self._stats.pulls_failed += 1
+                        # the posted data irecv cannot be cancelled: keep the
+                        # tracker AND the buffer alive for the service
+                        # lifetime, or a late-arriving send would complete
+                        # into freed memory
+                        if st.get("tracker") is not None:
+                            self._abandoned.append((st["tracker"], st["buf"]))
                         del pending_pulls[req_id]
                 before = len(pending_sends)
                 pending_sends[:] = [(t, b) for (t, b) in pending_sends

@@ -57,7 +57,8 @@ def test_full_control_plane(redis_srv):
                            bind=False)
     # dialing PUBs drop messages until the SUB handshake lands: retry the
     # idempotent stores until the index reflects both pods
-    deadline = time.time() + 15
+    expected = {"pod-a": 8.0, "pod-b-dp2": 8.0}
+    deadline = time.time() + 30
     while time.time() < deadline:
         pub_a.publish_events([
             block_stored_payload(list(range(1, 9)), None, tokens, 16)])
@@ -67,10 +68,10 @@ def test_full_control_plane(redis_srv):
                                   sliding_window=32)],
             dp_rank=2)
         s = ix.score_tokens(tokens, MODEL)
-        if len(s) == 2:
+        if s == expected:  # retry until the EXACT state lands (idempotent)
             break
         time.sleep(0.1)
-    assert s == {"pod-a": 8.0, "pod-b-dp2": 8.0}
+    assert s == expected
 
     # sliding pod evicts its out-of-window leading blocks: score survives
     # (window hints) even though the entries live in Redis
